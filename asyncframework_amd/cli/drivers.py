@@ -1,0 +1,145 @@
+"""The five algorithm drivers — CLI-compatible with the reference.
+
+Positional args (13) exactly as the reference drivers parse them
+(SparkASGDThread.scala:39-51): [path name] [file name] [num columns]
+[num rows] [num partitions] [num iterations] [step size] [taw] [batch rate]
+[bucket ratio] [printer freq] [coeff] [seed]; the MLlib baseline takes 8
+(SparkSGDMLLIB.scala:30-37). Extra engine flags (--device, --dtype,
+--objective, --sparse, --history-placement) come after the positionals.
+
+``file name`` = 'synthetic' generates data of the given shape instead of
+loading a LibSVM file (no network in this environment)."""
+
+from __future__ import annotations
+
+import argparse
+import sys
+from typing import List, Optional
+
+from ..engine.config import EngineConfig
+from ..utils import logfmt
+from .. import run as runner
+
+ARG_NAMES_13 = ["path name", "file name", "num columns", "num rows",
+                "num partitions", "num iterations", "step size", "taw",
+                "batch rate", "bucket ratio", "printer freq", "coeff",
+                "seed"]
+ARG_NAMES_8 = ["path name", "file name", "num columns", "num rows",
+               "num partitions", "num iterations", "step size", "batch rate"]
+
+
+def _engine_flags(p: argparse.ArgumentParser) -> None:
+    p.add_argument("--device", default="cpu",
+                   help="cpu | cuda | cuda:N (default cpu)")
+    p.add_argument("--dtype", default="fp32",
+                   choices=["fp32", "fp64", "bf16", "fp16"])
+    p.add_argument("--objective", default="lsq", choices=["lsq", "logistic"])
+    p.add_argument("--sparse", action="store_true",
+                   help="CSR data path (rcv1-shape)")
+    p.add_argument("--history-placement", default="device",
+                   choices=["device", "host"],
+                   help="SAGA history table in HBM (device) or pinned host "
+                        "DRAM (host spill, BASELINE config 5)")
+    p.add_argument("--max-wall-s", type=float, default=None)
+
+
+def _parse13(argv: List[str], prog: str):
+    p = argparse.ArgumentParser(prog=prog)
+    for name in ["pathname", "fname", "d", "N", "numPart", "numIter",
+                 "gamma", "taw", "b", "bucketRatio", "printerFreq", "coeff",
+                 "seed"]:
+        p.add_argument(name)
+    _engine_flags(p)
+    a = p.parse_args(argv)
+    return a
+
+
+def _parse8(argv: List[str], prog: str):
+    p = argparse.ArgumentParser(prog=prog)
+    for name in ["pathname", "fname", "d", "N", "numPart", "numIter",
+                 "gamma", "b"]:
+        p.add_argument(name)
+    _engine_flags(p)
+    return p.parse_args(argv)
+
+
+def _cfg13(a, algo: str, sync: bool) -> EngineConfig:
+    return EngineConfig(
+        d=int(a.d), N=int(a.N), num_workers=int(a.numPart),
+        num_iterations=int(a.numIter), gamma=float(a.gamma),
+        taw=int(a.taw), batch_rate=float(a.b),
+        bucket_ratio=float(a.bucketRatio), printer_freq=int(a.printerFreq),
+        delay_coeff=float(a.coeff), seed=int(a.seed), algo=algo, sync=sync,
+        objective=a.objective, dtype=a.dtype, device=a.device,
+        history_placement=a.history_placement)
+
+
+def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:
+    logfmt.print_header(app, names, vals)
+    sparse = a.sparse
+    data = runner.load_dataset(cfg, a.pathname, a.fname, sparse=sparse,
+                               device=a.device)
+    if sparse:
+        workers = runner.build_csr_workers(cfg, *data)
+    else:
+        workers = runner.build_dense_workers(cfg, *data)
+    res, _srv = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s)
+    runner.final_report(cfg, res, data, sparse, device=a.device)
+
+
+def _vals13(a):
+    return [a.pathname, a.fname, a.d, a.N, a.numPart, a.numIter, a.gamma,
+            a.taw, a.b, a.bucketRatio, a.printerFreq, a.coeff, a.seed]
+
+
+def asgd_thread(argv: Optional[List[str]] = None) -> None:
+    """Async bounded-staleness SGD (reference SparkASGDThread)."""
+    a = _parse13(argv if argv is not None else sys.argv[1:], "asgd-thread")
+    cfg = _cfg13(a, "asgd", sync=False)
+    _run(cfg, a, "ASGD", ARG_NAMES_13, _vals13(a))
+
+
+def asgd_sync(argv: Optional[List[str]] = None) -> None:
+    """Synchronous SGD with user-space barrier (reference SparkASGDSync)."""
+    a = _parse13(argv if argv is not None else sys.argv[1:], "asgd-sync")
+    cfg = _cfg13(a, "asgd", sync=True)
+    _run(cfg, a, "ASGDSync", ARG_NAMES_13, _vals13(a))
+
+
+def asaga_thread(argv: Optional[List[str]] = None) -> None:
+    """Async SAGA with per-sample gradient history (reference
+    SparkASAGAThread)."""
+    a = _parse13(argv if argv is not None else sys.argv[1:], "asaga-thread")
+    cfg = _cfg13(a, "asaga", sync=False)
+    _run(cfg, a, "ASAGA", ARG_NAMES_13, _vals13(a))
+
+
+def asaga_sync(argv: Optional[List[str]] = None) -> None:
+    """Synchronous SAGA (reference SparkASAGASync)."""
+    a = _parse13(argv if argv is not None else sys.argv[1:], "asaga-sync")
+    cfg = _cfg13(a, "asaga", sync=True)
+    _run(cfg, a, "ASAGASync", ARG_NAMES_13, _vals13(a))
+
+
+def sgd_mllib(argv: Optional[List[str]] = None) -> None:
+    """MLlib mini-batch SGD baseline (reference SparkSGDMLLIB +
+    GradientDescent.runMiniBatchSGD with the 100-iteration weight-history
+    hook, GradientDescent.scala:255-260)."""
+    a = _parse8(argv if argv is not None else sys.argv[1:], "sgd-mllib")
+    cfg = EngineConfig(
+        d=int(a.d), N=int(a.N), num_workers=int(a.numPart),
+        num_iterations=int(a.numIter), gamma=float(a.gamma),
+        batch_rate=float(a.b), algo="mllib", sync=True, printer_freq=100,
+        objective=a.objective, dtype=a.dtype, device=a.device, seed=42,
+        delay_coeff=0.0)
+    logfmt.print_header("MLlib SGD", ARG_NAMES_8,
+                        [a.pathname, a.fname, a.d, a.N, a.numPart, a.numIter,
+                         a.gamma, a.b])
+    data = runner.load_dataset(cfg, a.pathname, a.fname, sparse=a.sparse,
+                               device=a.device)
+    if a.sparse:
+        workers = runner.build_csr_workers(cfg, *data)
+    else:
+        workers = runner.build_dense_workers(cfg, *data)
+    res, _ = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s)
+    runner.final_report(cfg, res, data, a.sparse, device=a.device)

@@ -1,0 +1,56 @@
+"""LibSVM-format loader -> dense or CSR torch tensors.
+
+MI355X-native replacement for the reference's MLUtils.loadLibSVMFile
+(reference mllib/src/main/scala/org/apache/spark/mllib/util/MLUtils.scala:
+71-166): plain file parse into CSR arrays, optional densify, no Spark RDDs.
+LibSVM lines are ``label idx:val idx:val ...`` with 1-based indices.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+
+def load_libsvm(path: str, n_features: Optional[int] = None,
+                dense: bool = False,
+                dtype: torch.dtype = torch.float32,
+                device: str | torch.device = "cpu"):
+    """Returns (indptr, indices, values, y) CSR tensors, or (X, y) if
+    ``dense``. Indices are converted to 0-based as in the reference loader
+    (MLUtils.scala:91: ``indices.map(_ - 1)``)."""
+    labels = []
+    indptr = [0]
+    cols: list[int] = []
+    vals: list[float] = []
+    with open(path, "r") as f:
+        for line in f:
+            line = line.strip()
+            if not line or line.startswith("#"):
+                continue
+            parts = line.split()
+            labels.append(float(parts[0]))
+            for item in parts[1:]:
+                i, v = item.split(":")
+                cols.append(int(i) - 1)
+                vals.append(float(v))
+            indptr.append(len(cols))
+    y = np.asarray(labels, dtype=np.float32)
+    indptr_a = np.asarray(indptr, dtype=np.int32)
+    cols_a = np.asarray(cols, dtype=np.int32)
+    vals_a = np.asarray(vals, dtype=np.float32)
+    d = n_features if n_features is not None else (int(cols_a.max()) + 1 if cols_a.size else 0)
+    dev = torch.device(device)
+    if dense:
+        n = len(labels)
+        X = np.zeros((n, d), dtype=np.float32)
+        for r in range(n):
+            s, t = indptr_a[r], indptr_a[r + 1]
+            X[r, cols_a[s:t]] = vals_a[s:t]
+        return torch.from_numpy(X).to(dev).to(dtype), torch.from_numpy(y).to(dev)
+    return (torch.from_numpy(indptr_a).to(dev),
+            torch.from_numpy(cols_a).to(dev),
+            torch.from_numpy(vals_a).to(dev).to(dtype),
+            torch.from_numpy(y).to(dev))

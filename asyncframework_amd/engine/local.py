@@ -1,0 +1,287 @@
+"""In-process engine: worker threads + updater thread + dispatch loop.
+
+Thread roles mirror the reference exactly:
+
+* main thread      = the driver main loop (quorum gate, weight broadcast,
+                     dispatch; SparkASGDThread.scala:230-345),
+* updater thread   = the parameter server (mailbox drain, tau filter, weight
+                     update, worker requeue; :153-226),
+* worker threads   = executors running the fused gradient kernel on their
+                     own HIP stream (Executor.TaskRunner analog).
+
+On GPU all workers of one process share that process's device, each on its
+own stream; the multi-process engine (engine.dist) wires remote ranks into
+the same Server via channel proxies, so this file is the whole control plane.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .config import EngineConfig
+from .delay import DelayInjector
+from .messages import Dispatch, WorkerResult
+from .server import Server
+from .worker import Worker
+
+
+@dataclass
+class RunResult:
+    k: int
+    elapsed_ms: int
+    opt_vars: List[Tuple[int, torch.Tensor]]
+    waiting_time: Dict[int, int]
+    w: torch.Tensor
+    staleness_seen: List[int] = field(default_factory=list)
+    applied: int = 0
+    rejected: int = 0
+
+
+class _LocalChannel:
+    """Dispatch queue + thread for one in-process worker. Results flow
+    straight into the server's completion path (the JobWaiter.taskSucceeded
+    analog, reference JobWaiter.scala:56-60)."""
+
+    def __init__(self, worker: Worker, server: Server):
+        self.worker = worker
+        self.server = server
+        self.q: "deque[Dispatch]" = deque()
+        self._ev = threading.Event()
+        self.thread = threading.Thread(target=self._loop, daemon=True,
+                                       name=f"worker-{worker.id}")
+
+    def start(self):
+        self.thread.start()
+
+    def dispatch(self, msg: Dispatch) -> None:
+        self.q.append(msg)
+        self._ev.set()
+
+    def _loop(self):
+        while True:
+            if not self.q:
+                self._ev.wait(timeout=0.05)
+                self._ev.clear()
+                continue
+            msg = self.q.popleft()
+            if msg.stop:
+                break
+            res = self.worker.process(msg)
+            self.server.on_completion(res)
+
+    def join(self, timeout=None):
+        self.thread.join(timeout)
+
+
+class AsyncEngine:
+    """Bounded-staleness asynchronous engine (ASGD/ASAGA)."""
+
+    def __init__(self, cfg: EngineConfig, workers: List[Worker],
+                 server: Optional[Server] = None,
+                 delay: Optional[DelayInjector] = None):
+        assert len(workers) == cfg.num_workers
+        self.cfg = cfg
+        self.server = server or Server(cfg, device=workers[0].device)
+        self.delay = delay or DelayInjector(
+            cfg.num_workers, cfg.delay_coeff, cfg.seed,
+            calib_window=cfg.calib_factor * cfg.num_workers)
+        self.channels = [_LocalChannel(w, self.server) for w in workers]
+        self.pending: "deque[int]" = deque(range(cfg.num_workers))
+        self._stop = threading.Event()
+        self.staleness_seen: List[int] = []
+        self.applied = 0
+        self.rejected = 0
+        self.verbose = True
+
+    # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
+    def _updater(self):
+        cfg, srv = self.cfg, self.server
+        while srv.k < cfg.num_iterations and not self._stop.is_set():
+            if not srv.AC.hasNext():
+                time.sleep(0.001)
+                continue
+            bsize = srv.AC.getSize()
+            for _ in range(bsize):
+                try:
+                    pr = srv.AC.ASYNCcollectAll(timeout=0.1)
+                except Exception:
+                    break
+                res: WorkerResult = pr.gettaskResult()
+                wid = pr.getWorkerID()
+                now = time.perf_counter()
+                if self.cfg.algo == "asaga":
+                    self.staleness_seen.append(srv.k - pr.getStaleness())
+                else:
+                    self.staleness_seen.append(pr.getStaleness())
+                if self.accepts_now(pr):
+                    srv.finish_time[wid] = now
+                    sub = srv.submit_time.get(wid)
+                    if sub is not None:
+                        self.delay.record_task(srv.k, (now - sub) * 1000.0)
+                    srv.apply(res)
+                    srv.last_accept[wid] = True
+                    self.pending.append(wid)
+                    if srv.k % cfg.printer_freq == 0:
+                        if self.verbose:
+                            print(f"Iteration {srv.k} is finished")
+                        srv.maybe_log()
+                    srv.k += 1
+                    self.applied += 1
+                else:
+                    srv.last_accept[wid] = False
+                    self.pending.append(wid)
+                    self.rejected += 1
+                if srv.k >= cfg.num_iterations:
+                    break
+
+    def accepts_now(self, pr) -> bool:
+        return self.server.accepts(pr)
+
+    # -- main dispatch loop (reference SparkASGDThread.scala:230-345) --------
+    def run(self, max_wall_s: Optional[float] = None) -> RunResult:
+        cfg, srv = self.cfg, self.server
+        srv.start_time = time.perf_counter()
+        if cfg.snapshot_weights and not srv.opt_vars:
+            srv.opt_vars.append((0, srv.w.detach().cpu().clone()))
+        for ch in self.channels:
+            ch.start()
+        updater = threading.Thread(target=self._updater, daemon=True,
+                                   name="updater")
+        updater.start()
+        t_start = time.perf_counter()
+        first = True
+        while srv.k < cfg.num_iterations:
+            if max_wall_s and time.perf_counter() - t_start > max_wall_s:
+                self._stop.set()
+                break
+            init_workers = (cfg.num_workers if first
+                            else srv.available_workers())
+            if init_workers >= cfg.gate and self.pending:
+                first = False
+                workers_list = []
+                qsize = len(self.pending)
+                for _ in range(qsize):
+                    workers_list.append(self.pending.popleft())
+                self.delay.maybe_activate(srv.k)
+                w_snap = srv.w.detach().clone()
+                now = time.perf_counter()
+                k_now = srv.k
+                for wid in workers_list:
+                    prev_fin = srv.finish_time.get(wid, now)
+                    srv.waiting_time[wid] = (srv.waiting_time.get(wid, 0)
+                                             + int((now - prev_fin) * 1000))
+                    srv.submit_time[wid] = now
+                    srv.AC.STAT[wid].setAvailability(False)
+                    msg = Dispatch(
+                        w=w_snap, ts=srv.AC.getCurrentTime(),
+                        k_submit=k_now,
+                        accept_prev=srv.last_accept.get(wid, True),
+                        delay_s=self.delay.delay_ms(wid, k_now) / 1000.0)
+                    self.channels[wid].dispatch(msg)
+            else:
+                time.sleep(0.001)
+        elapsed = srv.elapsed_ms()
+        self._stop.set()
+        for ch in self.channels:
+            ch.dispatch(Dispatch(w=None, stop=True))
+        updater.join(timeout=10.0)
+        for ch in self.channels:
+            ch.join(timeout=10.0)
+        return RunResult(k=srv.k, elapsed_ms=elapsed, opt_vars=srv.opt_vars,
+                         waiting_time=srv.waiting_time, w=srv.w,
+                         staleness_seen=self.staleness_seen,
+                         applied=self.applied, rejected=self.rejected)
+
+
+class SyncEngine:
+    """Synchronous variants: SparkASGDSync (user-space barrier counting all P
+    results per round, average, step gamma/sqrt(k+1); reference
+    SparkASGDSync.scala:239-277) and SparkASAGASync (full-barrier SAGA;
+    SparkASAGASync.scala:263-304)."""
+
+    def __init__(self, cfg: EngineConfig, workers: List[Worker],
+                 server: Optional[Server] = None,
+                 delay: Optional[DelayInjector] = None):
+        self.cfg = cfg
+        self.server = server or Server(cfg, device=workers[0].device)
+        self.delay = delay or DelayInjector(
+            cfg.num_workers, cfg.delay_coeff, cfg.seed,
+            calib_window=cfg.calib_factor * cfg.num_workers)
+        self.channels = [_LocalChannel(w, self.server) for w in workers]
+        self.verbose = True
+
+    def run(self, max_wall_s: Optional[float] = None) -> RunResult:
+        import math
+        cfg, srv = self.cfg, self.server
+        srv.start_time = time.perf_counter()
+        for ch in self.channels:
+            ch.start()
+        P = cfg.num_workers
+        t_start = time.perf_counter()
+        for k in range(cfg.num_iterations):
+            if max_wall_s and time.perf_counter() - t_start > max_wall_s:
+                break
+            self.delay.maybe_activate(k)
+            w_snap = srv.w.detach().clone()
+            now = time.perf_counter()
+            for wid in range(P):
+                srv.submit_time[wid] = now
+                msg = Dispatch(w=w_snap, ts=srv.AC.getCurrentTime(),
+                               k_submit=k, accept_prev=True,
+                               delay_s=self.delay.delay_ms(wid, k) / 1000.0)
+                self.channels[wid].dispatch(msg)
+            acc = torch.zeros_like(srv.w)
+            got = 0
+            nrows_round = 0
+            while got < P:
+                pr = srv.AC.ASYNCcollectAll(timeout=60.0)
+                res: WorkerResult = pr.gettaskResult()
+                wid = pr.getWorkerID()
+                now2 = time.perf_counter()
+                sub = srv.submit_time.get(wid)
+                if sub is not None:
+                    self.delay.record_task(k * P, (now2 - sub) * 1000.0)
+                g = res.g
+                if g.device != srv.device:
+                    g = g.to(srv.device)
+                acc += g
+                nrows_round += res.nrows
+                got += 1
+            from .. import ops
+            if cfg.algo == "asaga":
+                # SparkASAGASync.scala:300-304: parRecs = b*N
+                ops.saga_update(srv.w, acc, srv.alpha_bar, cfg.gamma,
+                                1.0 / (cfg.batch_rate * cfg.N), 1.0 / cfg.N)
+            elif cfg.algo == "mllib":
+                # MLlib GradientDescent.runMiniBatchSGD semantics: the
+                # gradient sum is divided by the ACTUAL minibatch size and
+                # stepped with stepSize/sqrt(iter), iter from 1 (reference
+                # mllib/.../optimization/GradientDescent.scala:287-290,
+                # SimpleUpdater).
+                gamma_k = cfg.gamma / math.sqrt(k + 1)
+                ops.sgd_update(srv.w, acc, gamma_k,
+                               1.0 / max(nrows_round, 1))
+            else:
+                # SparkASGDSync.scala:273-277
+                gamma_k = cfg.gamma / math.sqrt(k + 1)
+                ops.sgd_update(srv.w, acc, gamma_k,
+                               1.0 / (cfg.batch_rate * cfg.N))
+            if k % cfg.printer_freq == 0:
+                if self.verbose:
+                    print(f"Iteration {k} is finished")
+                srv.k = k
+                srv.maybe_log()
+            srv.k = k + 1
+        elapsed = srv.elapsed_ms()
+        for ch in self.channels:
+            ch.dispatch(Dispatch(w=None, stop=True))
+        for ch in self.channels:
+            ch.join(timeout=10.0)
+        return RunResult(k=srv.k, elapsed_ms=elapsed, opt_vars=srv.opt_vars,
+                         waiting_time=srv.waiting_time, w=srv.w)

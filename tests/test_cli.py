@@ -1,0 +1,95 @@
+"""CLI drivers: 13-positional-arg parsing + the stdout contract
+(reference run template, README.md; log lines SparkASGDThread.scala:29-65,
+196, 349, 355-362, 400-410)."""
+
+import io
+import re
+from contextlib import redirect_stdout
+
+import pytest
+
+from asyncframework_amd.cli import drivers
+
+
+ARGS13 = ["synthetic", "synthetic", "16", "200", "2", "30", "0.5", "1000000",
+          "0.3", "0.5", "10", "0", "42"]
+ARGS8 = ["synthetic", "synthetic", "16", "200", "2", "30", "0.5", "0.3"]
+
+
+def _capture(fn, args):
+    buf = io.StringIO()
+    with redirect_stdout(buf):
+        fn(args)
+    return buf.getvalue()
+
+
+def _check_contract(out, app):
+    lines = out.splitlines()
+    assert lines[0] == f"Spark {app} application started"
+    assert lines[1] == "Input arguments:"
+    assert any(l.startswith("Input format:") for l in lines)
+    assert any(re.match(r"Elapsed time\(ms\): \d+", l) for l in lines)
+    assert "finished" == lines[-1]
+    # final time,objective CSV lines
+    csv = [l for l in lines if re.match(r"^\d+,[0-9.eE+-]+$", l)]
+    assert len(csv) >= 1
+    return lines
+
+
+def test_asgd_thread_contract():
+    out = _capture(drivers.asgd_thread, ARGS13)
+    lines = _check_contract(out, "ASGD")
+    assert any(re.match(r"Iteration \d+ is finished", l) for l in lines)
+    assert "Individual waiting times:" in lines
+    assert any(l.startswith("Average waiting time(ms) per worker and "
+                            "iteration:") for l in lines)
+
+
+def test_asgd_sync_contract():
+    out = _capture(drivers.asgd_sync, ARGS13)
+    _check_contract(out, "ASGDSync")
+
+
+def test_asaga_thread_contract():
+    args = list(ARGS13)
+    args[6] = "0.05"
+    out = _capture(drivers.asaga_thread, args)
+    _check_contract(out, "ASAGA")
+
+
+def test_asaga_sync_contract():
+    args = list(ARGS13)
+    args[6] = "0.05"
+    out = _capture(drivers.asaga_sync, args)
+    _check_contract(out, "ASAGASync")
+
+
+def test_sgd_mllib_contract():
+    out = _capture(drivers.sgd_mllib, ARGS8)
+    _check_contract(out, "MLlib SGD")
+
+
+def test_arg_echo():
+    out = _capture(drivers.asgd_thread, ARGS13)
+    assert "num columns: 16" in out
+    assert "num rows: 200" in out
+    assert "num partitions: 2" in out
+    assert "step size: 0.5" in out
+    assert "taw: 1000000" in out
+    assert "batch rate: 0.3" in out
+    assert "bucket ratio: 0.5" in out
+    assert "printer freq: 10" in out
+    assert "coeff: 0" in out
+    assert "seed: 42" in out
+
+
+def test_objective_decreases_in_output():
+    out = _capture(drivers.asgd_thread, ARGS13)
+    lines = out.splitlines()
+    # objective CSV comes after the LAST ********* separator
+    # (waiting-time lines share the "<int>,<num>" shape — reference format)
+    sep = max(i for i, l in enumerate(lines) if l.startswith("*********"))
+    csv = [l for l in lines[sep + 1:] if re.match(r"^\d+,[0-9.eE+-]+$", l)]
+    objs = [float(l.split(",")[1]) for l in csv]
+    assert len(objs) >= 2
+    assert objs[-1] < objs[0]
