@@ -82,22 +82,32 @@ class _LocalChannel:
 class AsyncEngine:
     """Bounded-staleness asynchronous engine (ASGD/ASAGA)."""
 
-    def __init__(self, cfg: EngineConfig, workers: List[Worker],
+    def __init__(self, cfg: EngineConfig, workers: Optional[List[Worker]] = None,
                  server: Optional[Server] = None,
-                 delay: Optional[DelayInjector] = None):
-        assert len(workers) == cfg.num_workers
+                 delay: Optional[DelayInjector] = None,
+                 channels: Optional[List] = None):
         self.cfg = cfg
-        self.server = server or Server(cfg, device=workers[0].device)
+        if channels is None:
+            assert workers is not None and len(workers) == cfg.num_workers
+            self.server = server or Server(cfg, device=workers[0].device)
+            channels = [_LocalChannel(w, self.server) for w in workers]
+        else:
+            assert server is not None
+            self.server = server
         self.delay = delay or DelayInjector(
             cfg.num_workers, cfg.delay_coeff, cfg.seed,
             calib_window=cfg.calib_factor * cfg.num_workers)
-        self.channels = [_LocalChannel(w, self.server) for w in workers]
+        self.channels = channels
+        assert len(self.channels) == cfg.num_workers
         self.pending: "deque[int]" = deque(range(cfg.num_workers))
         self._stop = threading.Event()
         self.staleness_seen: List[int] = []
         self.applied = 0
         self.rejected = 0
         self.verbose = True
+        # bench hooks: wall-clock stamps taken when k first reaches a mark
+        self.mark_at = set()
+        self.marks = {}
 
     # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
     def _updater(self):
@@ -133,6 +143,8 @@ class AsyncEngine:
                         srv.maybe_log()
                     srv.k += 1
                     self.applied += 1
+                    if srv.k in self.mark_at:
+                        self.marks[srv.k] = time.perf_counter()
                 else:
                     srv.last_accept[wid] = False
                     self.pending.append(wid)
@@ -205,16 +217,25 @@ class SyncEngine:
     SparkASGDSync.scala:239-277) and SparkASAGASync (full-barrier SAGA;
     SparkASAGASync.scala:263-304)."""
 
-    def __init__(self, cfg: EngineConfig, workers: List[Worker],
+    def __init__(self, cfg: EngineConfig, workers: Optional[List[Worker]] = None,
                  server: Optional[Server] = None,
-                 delay: Optional[DelayInjector] = None):
+                 delay: Optional[DelayInjector] = None,
+                 channels: Optional[List] = None):
         self.cfg = cfg
-        self.server = server or Server(cfg, device=workers[0].device)
+        if channels is None:
+            assert workers is not None
+            self.server = server or Server(cfg, device=workers[0].device)
+            channels = [_LocalChannel(w, self.server) for w in workers]
+        else:
+            assert server is not None
+            self.server = server
         self.delay = delay or DelayInjector(
             cfg.num_workers, cfg.delay_coeff, cfg.seed,
             calib_window=cfg.calib_factor * cfg.num_workers)
-        self.channels = [_LocalChannel(w, self.server) for w in workers]
+        self.channels = channels
         self.verbose = True
+        self.mark_at = set()
+        self.marks = {}
 
     def run(self, max_wall_s: Optional[float] = None) -> RunResult:
         import math
@@ -278,6 +299,8 @@ class SyncEngine:
                 srv.k = k
                 srv.maybe_log()
             srv.k = k + 1
+            if srv.k in self.mark_at:
+                self.marks[srv.k] = time.perf_counter()
         elapsed = srv.elapsed_ms()
         for ch in self.channels:
             ch.dispatch(Dispatch(w=None, stop=True))

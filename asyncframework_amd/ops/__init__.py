@@ -27,7 +27,7 @@ def _load_hip():
     if _hip is not None or _hip_err is not None:
         return _hip
     try:
-        from .. import _hip as mod  # in-tree .so
+        from . import hip as mod  # adapter over the in-tree _hip_core .so
         _hip = mod
     except ImportError as e:  # pragma: no cover - exercised on GPU box only
         _hip_err = str(e)

@@ -1,0 +1,129 @@
+"""Tensor-level adapter over the raw CDNA4 kernel module
+(``asyncframework_amd._hip_core``, built in-tree by setup.py with
+``hipcc --offload-arch=gfx950``). Validates shapes/dtypes, extracts device
+pointers and the current HIP stream, and returns torch-native results.
+
+Importing this module raises ImportError when the extension .so is missing —
+ops.__init__ turns that into a loud error on GPU boxes."""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from .. import _hip_core  # in-tree .so — ImportError here is intentional
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _chk(t: torch.Tensor, name: str, dtype=None, contiguous=True):
+    assert t.is_cuda, f"{name} must be a CUDA tensor"
+    if dtype is not None:
+        assert t.dtype == dtype, f"{name} must be {dtype}, got {t.dtype}"
+    if contiguous:
+        assert t.is_contiguous(), f"{name} must be contiguous"
+
+
+def _xcode(X: torch.Tensor) -> int:
+    if X.dtype == torch.bfloat16:
+        return 1
+    if X.dtype == torch.float32:
+        return 0
+    raise TypeError(f"X dtype must be fp32 or bf16, got {X.dtype}")
+
+
+def grad_dense(X, y, w, out, seed, round_k, row_start, rate, obj) -> int:
+    _chk(X, "X")
+    _chk(y, "y", torch.float32)
+    _chk(out, "out", torch.float32)
+    w = w.float().contiguous()
+    n_rows, d = X.shape
+    n_ctr = torch.zeros(1, dtype=torch.int32, device=X.device)
+    _hip_core.grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
+                         out.data_ptr(), n_ctr.data_ptr(), n_rows, d, seed,
+                         round_k & 0xFFFFFFFF, row_start, rate, obj,
+                         _xcode(X), _stream())
+    return int(n_ctr.item())
+
+
+def grad_csr(indptr, indices, values, y, w, out, seed, round_k, row_start,
+             rate, obj) -> int:
+    _chk(indptr, "indptr", torch.int32)
+    _chk(indices, "indices", torch.int32)
+    _chk(values, "values")
+    _chk(y, "y", torch.float32)
+    _chk(out, "out", torch.float32)
+    w = w.float().contiguous()
+    n_rows = indptr.shape[0] - 1
+    n_ctr = torch.zeros(1, dtype=torch.int32, device=w.device)
+    _hip_core.grad_csr(indptr.data_ptr(), indices.data_ptr(),
+                       values.data_ptr(), y.data_ptr(), w.data_ptr(),
+                       out.data_ptr(), n_ctr.data_ptr(), n_rows, seed,
+                       round_k & 0xFFFFFFFF, row_start, rate, obj,
+                       _xcode(values), _stream())
+    return int(n_ctr.item())
+
+
+def saga_grad_dense(X, y, w, alpha, g, seed, round_k, row_start, rate, obj
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
+    _chk(X, "X")
+    _chk(alpha, "alpha", torch.float32)
+    _chk(g, "g", torch.float32)
+    w = w.float().contiguous()
+    n_rows, d = X.shape
+    cap = n_rows
+    idx = torch.empty(cap, dtype=torch.int32, device=X.device)
+    e = torch.empty(cap, dtype=torch.float32, device=X.device)
+    ctr = torch.zeros(2, dtype=torch.int32, device=X.device)  # [n, pos]
+    _hip_core.saga_grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
+                              alpha.data_ptr(), g.data_ptr(), ctr.data_ptr(),
+                              idx.data_ptr(), e.data_ptr(),
+                              ctr.data_ptr() + 4, n_rows, d, seed,
+                              round_k & 0xFFFFFFFF, row_start, rate, obj,
+                              _xcode(X), _stream())
+    n = int(ctr[1].item())
+    return idx[:n].long(), e[:n]
+
+
+def saga_grad_csr(indptr, indices, values, y, w, alpha, g, seed, round_k,
+                  row_start, rate, obj) -> Tuple[torch.Tensor, torch.Tensor]:
+    _chk(indptr, "indptr", torch.int32)
+    _chk(alpha, "alpha", torch.float32)
+    _chk(g, "g", torch.float32)
+    w = w.float().contiguous()
+    n_rows = indptr.shape[0] - 1
+    idx = torch.empty(n_rows, dtype=torch.int32, device=w.device)
+    e = torch.empty(n_rows, dtype=torch.float32, device=w.device)
+    ctr = torch.zeros(2, dtype=torch.int32, device=w.device)
+    _hip_core.saga_grad_csr(indptr.data_ptr(), indices.data_ptr(),
+                            values.data_ptr(), y.data_ptr(), w.data_ptr(),
+                            alpha.data_ptr(), g.data_ptr(), ctr.data_ptr(),
+                            idx.data_ptr(), e.data_ptr(), ctr.data_ptr() + 4,
+                            n_rows, seed, round_k & 0xFFFFFFFF, row_start,
+                            rate, obj, _xcode(values), _stream())
+    n = int(ctr[1].item())
+    return idx[:n].long(), e[:n]
+
+
+def saga_commit(alpha, idx, e) -> None:
+    _chk(alpha, "alpha", torch.float32)
+    idx32 = idx.to(torch.int32) if idx.dtype != torch.int32 else idx
+    _hip_core.saga_commit(alpha.data_ptr(), idx32.data_ptr(), e.data_ptr(),
+                          int(idx32.numel()), _stream())
+
+
+def sgd_update(w, g, gamma_k, inv_batch) -> None:
+    _chk(w, "w", torch.float32)
+    _chk(g, "g", torch.float32)
+    _hip_core.sgd_update(w.data_ptr(), g.data_ptr(), gamma_k, inv_batch,
+                         int(w.numel()), _stream())
+
+
+def saga_update(w, g, alpha_bar, gamma, inv_batch, inv_N) -> None:
+    _chk(w, "w", torch.float32)
+    _chk(alpha_bar, "alpha_bar", torch.float32)
+    _hip_core.saga_update(w.data_ptr(), g.data_ptr(), alpha_bar.data_ptr(),
+                          gamma, inv_batch, inv_N, int(w.numel()), _stream())

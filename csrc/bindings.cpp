@@ -1,0 +1,127 @@
+// pybind11 bindings for the CDNA4 kernels (csrc/kernels.hip).
+//
+// The module is deliberately torch-ABI-free: functions take raw device
+// pointers (as uintptr_t) + shapes + the HIP stream handle, which the thin
+// Python adapter (asyncframework_amd/ops/hip.py) extracts from torch
+// tensors. This keeps the extension a pure hipcc build (no hipify, no torch
+// C++ headers) and lets the upcoming native runtime call the same
+// launchers directly from C++.
+
+#include <pybind11/pybind11.h>
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <stdexcept>
+
+namespace py = pybind11;
+
+extern "C" {
+void launch_grad_dense(const void*, const float*, const float*, float*, int*,
+                       long, int, uint64_t, uint32_t, uint64_t, double, int,
+                       int, hipStream_t);
+void launch_saga_grad_dense(const void*, const float*, const float*,
+                            const float*, float*, int*, int*, float*, int*,
+                            long, int, uint64_t, uint32_t, uint64_t, double,
+                            int, int, hipStream_t);
+void launch_grad_csr(const int*, const int*, const void*, const float*,
+                     const float*, float*, int*, long, uint64_t, uint32_t,
+                     uint64_t, double, int, int, hipStream_t);
+void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
+                          const float*, const float*, float*, int*, int*,
+                          float*, int*, long, uint64_t, uint32_t, uint64_t,
+                          double, int, int, hipStream_t);
+void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
+void launch_saga_update(float*, const float*, float*, float, float, float,
+                        int, hipStream_t);
+void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
+}
+
+static void check(hipError_t err, const char* what) {
+  if (err != hipSuccess)
+    throw std::runtime_error(std::string(what) + ": " +
+                             hipGetErrorString(err));
+}
+
+PYBIND11_MODULE(_hip_core, m) {
+  m.doc() = "MI355X CDNA4 kernels for asyncframework_amd";
+
+  m.def("grad_dense",
+        [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t g, uintptr_t n,
+           long n_rows, int d, uint64_t seed, uint32_t round_k,
+           uint64_t row_start, double rate, int objective, int x_is_bf16,
+           uintptr_t stream) {
+          launch_grad_dense((const void*)X, (const float*)y, (const float*)w,
+                            (float*)g, (int*)n, n_rows, d, seed, round_k,
+                            row_start, rate, objective, x_is_bf16,
+                            (hipStream_t)stream);
+          check(hipGetLastError(), "grad_dense launch");
+        });
+
+  m.def("saga_grad_dense",
+        [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t alpha, uintptr_t g,
+           uintptr_t n, uintptr_t idx, uintptr_t e, uintptr_t pos,
+           long n_rows, int d, uint64_t seed, uint32_t round_k,
+           uint64_t row_start, double rate, int objective, int x_is_bf16,
+           uintptr_t stream) {
+          launch_saga_grad_dense((const void*)X, (const float*)y,
+                                 (const float*)w, (const float*)alpha,
+                                 (float*)g, (int*)n, (int*)idx, (float*)e,
+                                 (int*)pos, n_rows, d, seed, round_k,
+                                 row_start, rate, objective, x_is_bf16,
+                                 (hipStream_t)stream);
+          check(hipGetLastError(), "saga_grad_dense launch");
+        });
+
+  m.def("grad_csr",
+        [](uintptr_t indptr, uintptr_t indices, uintptr_t values, uintptr_t y,
+           uintptr_t w, uintptr_t g, uintptr_t n, long n_rows, uint64_t seed,
+           uint32_t round_k, uint64_t row_start, double rate, int objective,
+           int v_is_bf16, uintptr_t stream) {
+          launch_grad_csr((const int*)indptr, (const int*)indices,
+                          (const void*)values, (const float*)y,
+                          (const float*)w, (float*)g, (int*)n, n_rows, seed,
+                          round_k, row_start, rate, objective, v_is_bf16,
+                          (hipStream_t)stream);
+          check(hipGetLastError(), "grad_csr launch");
+        });
+
+  m.def("saga_grad_csr",
+        [](uintptr_t indptr, uintptr_t indices, uintptr_t values, uintptr_t y,
+           uintptr_t w, uintptr_t alpha, uintptr_t g, uintptr_t n,
+           uintptr_t idx, uintptr_t e, uintptr_t pos, long n_rows,
+           uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
+           int objective, int v_is_bf16, uintptr_t stream) {
+          launch_saga_grad_csr((const int*)indptr, (const int*)indices,
+                               (const void*)values, (const float*)y,
+                               (const float*)w, (const float*)alpha,
+                               (float*)g, (int*)n, (int*)idx, (float*)e,
+                               (int*)pos, n_rows, seed, round_k, row_start,
+                               rate, objective, v_is_bf16,
+                               (hipStream_t)stream);
+          check(hipGetLastError(), "saga_grad_csr launch");
+        });
+
+  m.def("sgd_update", [](uintptr_t w, uintptr_t g, float gamma_k,
+                         float inv_batch, int d, uintptr_t stream) {
+    launch_sgd_update((float*)w, (const float*)g, gamma_k, inv_batch, d,
+                      (hipStream_t)stream);
+    check(hipGetLastError(), "sgd_update launch");
+  });
+
+  m.def("saga_update", [](uintptr_t w, uintptr_t g, uintptr_t ab, float gamma,
+                          float inv_batch, float inv_N, int d,
+                          uintptr_t stream) {
+    launch_saga_update((float*)w, (const float*)g, (float*)ab, gamma,
+                       inv_batch, inv_N, d, (hipStream_t)stream);
+    check(hipGetLastError(), "saga_update launch");
+  });
+
+  m.def("saga_commit", [](uintptr_t alpha, uintptr_t idx, uintptr_t e, int n,
+                          uintptr_t stream) {
+    launch_saga_commit((float*)alpha, (const int*)idx, (const float*)e, n,
+                       (hipStream_t)stream);
+    check(hipGetLastError(), "saga_commit launch");
+  });
+
+  m.attr("__hip__") = true;
+}

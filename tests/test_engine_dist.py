@@ -1,0 +1,82 @@
+"""Multi-process engine over gloo (CPU, world_size=2) — the no-GPU stand-in
+for the 8-GPU RCCL topology (rank 0 = server + worker 0, rank 1 = worker 1).
+Analog of the reference's local-cluster[2,1,1024] DistributedSuite strategy
+(SURVEY §4.3)."""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from asyncframework_amd.data.shard import row_shards
+from asyncframework_amd.data.synthetic import synthetic_dense
+from asyncframework_amd.engine.config import EngineConfig
+from asyncframework_amd.engine.dist import DistEngine
+from asyncframework_amd.engine.worker import Shard, Worker
+
+WORLD = 2
+
+
+def _mk_cfg(sync: bool, algo: str = "asgd") -> EngineConfig:
+    return EngineConfig(d=24, N=400, num_workers=WORLD, num_iterations=40,
+                        gamma=0.5 if algo == "asgd" else 0.05,
+                        taw=2 ** 30, batch_rate=0.3, bucket_ratio=0.5,
+                        printer_freq=10, delay_coeff=0.0, seed=42,
+                        device="cpu", sync=sync, algo=algo)
+
+
+def _rank_main(rank: int, init_file: str, sync: bool, algo: str,
+               out_file: str):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        cfg = _mk_cfg(sync, algo)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+        (s, t) = row_shards(cfg.N, WORLD)[rank]
+        worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X[s:t],
+                                    y=y[s:t]), cfg,
+                        device=torch.device("cpu"))
+        eng = DistEngine(cfg, worker, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=120)
+        if rank == 0:
+            obj0 = float(((X @ torch.zeros(cfg.d) - y) ** 2).mean())
+            obj1 = float(((X @ res.w - y) ** 2).mean())
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{obj0},{obj1}")
+    finally:
+        dist.destroy_process_group()
+
+
+def _run_dist(sync: bool, algo: str = "asgd"):
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_rank_main, args=(init_file, sync, algo, out_file),
+                 nprocs=WORLD, join=True)
+        with open(out_file) as f:
+            k, obj0, obj1 = f.read().split(",")
+        return int(k), float(obj0), float(obj1)
+
+
+@pytest.mark.timeout(300)
+def test_dist_async_asgd_gloo():
+    k, obj0, obj1 = _run_dist(sync=False)
+    assert k >= 40
+    assert obj1 < obj0
+
+
+@pytest.mark.timeout(300)
+def test_dist_sync_asgd_gloo():
+    k, obj0, obj1 = _run_dist(sync=True)
+    assert k == 40
+    assert obj1 < obj0
+
+
+@pytest.mark.timeout(300)
+def test_dist_async_asaga_gloo():
+    k, obj0, obj1 = _run_dist(sync=False, algo="asaga")
+    assert k >= 40
+    assert obj1 < obj0
