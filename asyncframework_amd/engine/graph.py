@@ -1,0 +1,189 @@
+"""GPU-resident round loop: the whole ASGD/ASAGA iteration captured in a
+hipGraph.
+
+On one GPU with one worker, the async engine's round degenerates to the
+strict sequence [sample+gradient, update] with staleness 0 — exactly the
+reference's semantics at P=1 (every result accepted, clock advances once per
+round). This engine keeps ALL round state on the device:
+
+* ``k_dev``  — the round counter (the reference's iteration ``k``); the
+  gradient kernel derives the Philox round key (= k+1, the analog of
+  ``sample(false, b, seed+k+1)``) and the update kernel derives the step
+  size gamma/sqrt(k/P+1) from it and increments it,
+* ``w``/``g``/``alpha``/``alpha_bar`` — weights, gradient accumulator,
+  SAGA history, SAGA average — so an unrolled sequence of
+  [grad, fused_update] kernel pairs is a complete hipGraph with zero host
+  logic. Replaying the graph runs UNROLL updates per ~5 us launch instead of
+  per ~1 ms of Python thread handoffs (measured: 780 -> tens of thousands of
+  updates/s on mnist8m shape).
+
+This is the launch-bound-inner-loop -> hipGraph design the MI355X build
+targets (no analog in the reference — Spark's per-iteration overhead is the
+4x the paper beats; here we remove ours)."""
+
+from __future__ import annotations
+
+import time
+from typing import List, Optional, Tuple
+
+import torch
+
+from .config import EngineConfig
+from .worker import Shard
+
+
+class GraphEngine:
+    """Sequential device-resident ASGD/ASAGA on one GPU (num_workers == 1)."""
+
+    UNROLL = 20
+
+    def __init__(self, cfg: EngineConfig, shard: Shard,
+                 device: torch.device, unroll: Optional[int] = None):
+        from .. import _hip_core  # mandatory native path
+        self._core = _hip_core
+        assert device.type == "cuda"
+        self.cfg = cfg
+        self.shard = shard
+        self.device = device
+        self.unroll = unroll or self.UNROLL
+        d = cfg.d
+        self.w = torch.zeros(d, dtype=torch.float32, device=device)
+        self.g = torch.zeros(d, dtype=torch.float32, device=device)
+        self.k_dev = torch.zeros(1, dtype=torch.int32, device=device)
+        self.n_dummy = torch.zeros(1, dtype=torch.int32, device=device)
+        self.alpha = None
+        self.alpha_bar = None
+        if cfg.algo == "asaga":
+            self.alpha = torch.zeros(shard.n_rows, dtype=torch.float32,
+                                     device=device)
+            self.alpha_bar = torch.zeros(d, dtype=torch.float32,
+                                         device=device)
+        self._graph: Optional[torch.cuda.CUDAGraph] = None
+        self._obj_code = {"lsq": 0, "logistic": 1}[cfg.objective]
+
+    # -- one round = [grad(k_dev), fused_update(k_dev++)] --------------------
+    def _launch_round(self) -> None:
+        cfg, sh = self.cfg, self.shard
+        stream = torch.cuda.current_stream().cuda_stream
+        seed = cfg.seed
+        if cfg.algo == "asaga":
+            if sh.is_sparse:
+                self._core.saga_grad_csr(
+                    sh.indptr.data_ptr(), sh.indices.data_ptr(),
+                    sh.values.data_ptr(), sh.y.data_ptr(), self.w.data_ptr(),
+                    self.alpha.data_ptr(), self.g.data_ptr(),
+                    self.n_dummy.data_ptr(), 0, 0, 0, self.k_dev.data_ptr(),
+                    1, sh.n_rows, seed, 0, sh.row_start, cfg.batch_rate,
+                    self._obj_code, 1 if sh.values.dtype == torch.bfloat16
+                    else 0, stream)
+            else:
+                self._core.saga_grad_dense(
+                    sh.X.data_ptr(), sh.y.data_ptr(), self.w.data_ptr(),
+                    self.alpha.data_ptr(), self.g.data_ptr(),
+                    self.n_dummy.data_ptr(), 0, 0, 0, self.k_dev.data_ptr(),
+                    1, sh.n_rows, cfg.d, seed, 0, sh.row_start,
+                    cfg.batch_rate, self._obj_code,
+                    1 if sh.X.dtype == torch.bfloat16 else 0, stream)
+            self._core.saga_update_fused(
+                self.w.data_ptr(), self.g.data_ptr(),
+                self.alpha_bar.data_ptr(), self.k_dev.data_ptr(), cfg.gamma,
+                1.0 / cfg.par_recs, 1.0 / cfg.N, cfg.d, stream)
+        else:
+            if sh.is_sparse:
+                self._core.grad_csr(
+                    sh.indptr.data_ptr(), sh.indices.data_ptr(),
+                    sh.values.data_ptr(), sh.y.data_ptr(), self.w.data_ptr(),
+                    self.g.data_ptr(), self.n_dummy.data_ptr(),
+                    self.k_dev.data_ptr(), sh.n_rows, seed, 0, sh.row_start,
+                    cfg.batch_rate, self._obj_code,
+                    1 if sh.values.dtype == torch.bfloat16 else 0, stream)
+            else:
+                self._core.grad_dense(
+                    sh.X.data_ptr(), sh.y.data_ptr(), self.w.data_ptr(),
+                    self.g.data_ptr(), self.n_dummy.data_ptr(),
+                    self.k_dev.data_ptr(), sh.n_rows, cfg.d, seed, 0,
+                    sh.row_start, cfg.batch_rate, self._obj_code,
+                    1 if sh.X.dtype == torch.bfloat16 else 0, stream)
+            self._core.sgd_update_fused(
+                self.w.data_ptr(), self.g.data_ptr(), self.k_dev.data_ptr(),
+                cfg.gamma, 1.0 / cfg.par_recs, cfg.num_workers, cfg.d,
+                stream)
+
+    def _capture(self) -> None:
+        # warm the kernels outside capture, then roll state back
+        state = self._save_state()
+        s = torch.cuda.Stream(self.device)
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._launch_round()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self._restore_state(state)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            for _ in range(self.unroll):
+                self._launch_round()
+        self._graph = g
+        self._restore_state(state)
+        torch.cuda.synchronize()
+
+    def _save_state(self):
+        st = [self.w.clone(), self.g.clone(), self.k_dev.clone()]
+        if self.alpha is not None:
+            st += [self.alpha.clone(), self.alpha_bar.clone()]
+        return st
+
+    def _restore_state(self, st):
+        self.w.copy_(st[0]); self.g.copy_(st[1]); self.k_dev.copy_(st[2])
+        if self.alpha is not None:
+            self.alpha.copy_(st[3]); self.alpha_bar.copy_(st[4])
+
+    def step_rounds(self, n: int) -> None:
+        """Advance n rounds (graph replays + per-round tail)."""
+        if self._graph is None:
+            self._capture()
+        full, rem = divmod(n, self.unroll)
+        for _ in range(full):
+            self._graph.replay()
+        for _ in range(rem):
+            self._launch_round()
+
+    def run(self, num_iterations: int,
+            snapshot_every: Optional[int] = None,
+            start_time: Optional[float] = None
+            ) -> List[Tuple[int, torch.Tensor]]:
+        """Run to completion; optionally snapshot (ms, w) every
+        ``snapshot_every`` applied updates (the optVars loss-curve mechanism,
+        reference SparkASGDThread.scala:195-198)."""
+        t0 = start_time or time.perf_counter()
+        opt_vars: List[Tuple[int, torch.Tensor]] = []
+        if snapshot_every:
+            opt_vars.append((0, self.w.detach().cpu().clone()))
+        done = 0
+        while done < num_iterations:
+            chunk = (min(snapshot_every, num_iterations - done)
+                     if snapshot_every else num_iterations - done)
+            self.step_rounds(chunk)
+            done += chunk
+            if snapshot_every:
+                torch.cuda.synchronize()
+                ms = int((time.perf_counter() - t0) * 1000)
+                opt_vars.append((ms, self.w.detach().cpu().clone()))
+        torch.cuda.synchronize()
+        return opt_vars
+
+    def bench(self, warmup: int, steps: int) -> Tuple[float, float]:
+        """Timed exactly-K-steps contract: returns (t0, t1) wall stamps
+        bracketing ``steps`` applied updates after ``warmup`` untimed ones,
+        with a device synchronize on both sides."""
+        self.step_rounds(warmup)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        self.step_rounds(steps)
+        torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        return t0, t1
+
+    @property
+    def k(self) -> int:
+        return int(self.k_dev.item())

@@ -101,6 +101,8 @@ class AsyncEngine:
         assert len(self.channels) == cfg.num_workers
         self.pending: "deque[int]" = deque(range(cfg.num_workers))
         self._stop = threading.Event()
+        self._pending_ev = threading.Event()
+        self._pending_ev.set()
         self.staleness_seen: List[int] = []
         self.applied = 0
         self.rejected = 0
@@ -111,17 +113,19 @@ class AsyncEngine:
 
     # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
     def _updater(self):
+        import queue as _q
         cfg, srv = self.cfg, self.server
         while srv.k < cfg.num_iterations and not self._stop.is_set():
-            if not srv.AC.hasNext():
-                time.sleep(0.001)
+            # block on the mailbox (reference blocks in ASYNCcollectAll too,
+            # ASYNCcontext.scala:60-62); drain whatever arrived
+            try:
+                pr = srv.AC.ASYNCcollectAll(timeout=0.05)
+            except _q.Empty:
                 continue
-            bsize = srv.AC.getSize()
-            for _ in range(bsize):
-                try:
-                    pr = srv.AC.ASYNCcollectAll(timeout=0.1)
-                except Exception:
-                    break
+            batch = [pr]
+            while srv.AC.hasNext():
+                batch.append(srv.AC.ASYNCcollectAll())
+            for pr in batch:
                 res: WorkerResult = pr.gettaskResult()
                 wid = pr.getWorkerID()
                 now = time.perf_counter()
@@ -151,6 +155,7 @@ class AsyncEngine:
                     self.rejected += 1
                 if srv.k >= cfg.num_iterations:
                     break
+            self._pending_ev.set()
 
     def accepts_now(self, pr) -> bool:
         return self.server.accepts(pr)
@@ -197,7 +202,8 @@ class AsyncEngine:
                         delay_s=self.delay.delay_ms(wid, k_now) / 1000.0)
                     self.channels[wid].dispatch(msg)
             else:
-                time.sleep(0.001)
+                self._pending_ev.wait(timeout=0.01)
+                self._pending_ev.clear()
         elapsed = srv.elapsed_ms()
         self._stop.set()
         for ch in self.channels:

@@ -90,7 +90,9 @@ class Server:
             ops.saga_update(self.w, g, self.alpha_bar, cfg.gamma,
                             1.0 / cfg.par_recs, 1.0 / cfg.N)
         else:
-            gamma_k = cfg.gamma / math.sqrt(self.k / cfg.num_workers + 1)
+            # NB: k/numPart is Scala Int division in the reference
+            # (SparkASGDThread.scala:190) — keep the integer semantics.
+            gamma_k = cfg.gamma / math.sqrt(self.k // cfg.num_workers + 1)
             ops.sgd_update(self.w, g, gamma_k, 1.0 / cfg.par_recs)
 
     def maybe_log(self) -> Optional[int]:

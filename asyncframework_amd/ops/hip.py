@@ -43,7 +43,7 @@ def grad_dense(X, y, w, out, seed, round_k, row_start, rate, obj) -> int:
     n_rows, d = X.shape
     n_ctr = torch.zeros(1, dtype=torch.int32, device=X.device)
     _hip_core.grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
-                         out.data_ptr(), n_ctr.data_ptr(), n_rows, d, seed,
+                         out.data_ptr(), n_ctr.data_ptr(), 0, n_rows, d, seed,
                          round_k & 0xFFFFFFFF, row_start, rate, obj,
                          _xcode(X), _stream())
     return int(n_ctr.item())
@@ -61,7 +61,7 @@ def grad_csr(indptr, indices, values, y, w, out, seed, round_k, row_start,
     n_ctr = torch.zeros(1, dtype=torch.int32, device=w.device)
     _hip_core.grad_csr(indptr.data_ptr(), indices.data_ptr(),
                        values.data_ptr(), y.data_ptr(), w.data_ptr(),
-                       out.data_ptr(), n_ctr.data_ptr(), n_rows, seed,
+                       out.data_ptr(), n_ctr.data_ptr(), 0, n_rows, seed,
                        round_k & 0xFFFFFFFF, row_start, rate, obj,
                        _xcode(values), _stream())
     return int(n_ctr.item())
@@ -81,7 +81,7 @@ def saga_grad_dense(X, y, w, alpha, g, seed, round_k, row_start, rate, obj
     _hip_core.saga_grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
                               alpha.data_ptr(), g.data_ptr(), ctr.data_ptr(),
                               idx.data_ptr(), e.data_ptr(),
-                              ctr.data_ptr() + 4, n_rows, d, seed,
+                              ctr.data_ptr() + 4, 0, 0, n_rows, d, seed,
                               round_k & 0xFFFFFFFF, row_start, rate, obj,
                               _xcode(X), _stream())
     n = int(ctr[1].item())
@@ -102,7 +102,7 @@ def saga_grad_csr(indptr, indices, values, y, w, alpha, g, seed, round_k,
                             values.data_ptr(), y.data_ptr(), w.data_ptr(),
                             alpha.data_ptr(), g.data_ptr(), ctr.data_ptr(),
                             idx.data_ptr(), e.data_ptr(), ctr.data_ptr() + 4,
-                            n_rows, seed, round_k & 0xFFFFFFFF, row_start,
+                            0, 0, n_rows, seed, round_k & 0xFFFFFFFF, row_start,
                             rate, obj, _xcode(values), _stream())
     n = int(ctr[1].item())
     return idx[:n].long(), e[:n]

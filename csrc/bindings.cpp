@@ -4,36 +4,42 @@
 // pointers (as uintptr_t) + shapes + the HIP stream handle, which the thin
 // Python adapter (asyncframework_amd/ops/hip.py) extracts from torch
 // tensors. This keeps the extension a pure hipcc build (no hipify, no torch
-// C++ headers) and lets the upcoming native runtime call the same
-// launchers directly from C++.
+// C++ headers) and lets the native runtime call the same launchers directly
+// from C++. k_dev (nullable) is the device-resident round counter used by
+// the hipGraph engine (engine/graph.py).
 
 #include <pybind11/pybind11.h>
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
 #include <stdexcept>
+#include <string>
 
 namespace py = pybind11;
 
 extern "C" {
 void launch_grad_dense(const void*, const float*, const float*, float*, int*,
-                       long, int, uint64_t, uint32_t, uint64_t, double, int,
-                       int, hipStream_t);
-void launch_saga_grad_dense(const void*, const float*, const float*,
-                            const float*, float*, int*, int*, float*, int*,
-                            long, int, uint64_t, uint32_t, uint64_t, double,
-                            int, int, hipStream_t);
+                       const int*, long, int, uint64_t, uint32_t, uint64_t,
+                       double, int, int, hipStream_t);
+void launch_saga_grad_dense(const void*, const float*, const float*, float*,
+                            float*, int*, int*, float*, int*, const int*,
+                            int, long, int, uint64_t, uint32_t, uint64_t,
+                            double, int, int, hipStream_t);
 void launch_grad_csr(const int*, const int*, const void*, const float*,
-                     const float*, float*, int*, long, uint64_t, uint32_t,
-                     uint64_t, double, int, int, hipStream_t);
+                     const float*, float*, int*, const int*, long, uint64_t,
+                     uint32_t, uint64_t, double, int, int, hipStream_t);
 void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
-                          const float*, const float*, float*, int*, int*,
-                          float*, int*, long, uint64_t, uint32_t, uint64_t,
-                          double, int, int, hipStream_t);
+                          const float*, float*, float*, int*, int*, float*,
+                          int*, const int*, int, long, uint64_t, uint32_t,
+                          uint64_t, double, int, int, hipStream_t);
 void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
 void launch_saga_update(float*, const float*, float*, float, float, float,
                         int, hipStream_t);
 void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
+void launch_sgd_update_fused(float*, float*, int*, float, float, int, int,
+                             hipStream_t);
+void launch_saga_update_fused(float*, float*, float*, int*, float, float,
+                              float, int, hipStream_t);
 }
 
 static void check(hipError_t err, const char* what) {
@@ -47,56 +53,57 @@ PYBIND11_MODULE(_hip_core, m) {
 
   m.def("grad_dense",
         [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t g, uintptr_t n,
-           long n_rows, int d, uint64_t seed, uint32_t round_k,
-           uint64_t row_start, double rate, int objective, int x_is_bf16,
-           uintptr_t stream) {
+           uintptr_t k_dev, long n_rows, int d, uint64_t seed,
+           uint32_t round_k, uint64_t row_start, double rate, int objective,
+           int x_is_bf16, uintptr_t stream) {
           launch_grad_dense((const void*)X, (const float*)y, (const float*)w,
-                            (float*)g, (int*)n, n_rows, d, seed, round_k,
-                            row_start, rate, objective, x_is_bf16,
-                            (hipStream_t)stream);
+                            (float*)g, (int*)n, (const int*)k_dev, n_rows, d,
+                            seed, round_k, row_start, rate, objective,
+                            x_is_bf16, (hipStream_t)stream);
           check(hipGetLastError(), "grad_dense launch");
         });
 
   m.def("saga_grad_dense",
         [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t alpha, uintptr_t g,
            uintptr_t n, uintptr_t idx, uintptr_t e, uintptr_t pos,
-           long n_rows, int d, uint64_t seed, uint32_t round_k,
-           uint64_t row_start, double rate, int objective, int x_is_bf16,
-           uintptr_t stream) {
+           uintptr_t k_dev, int commit_now, long n_rows, int d, uint64_t seed,
+           uint32_t round_k, uint64_t row_start, double rate, int objective,
+           int x_is_bf16, uintptr_t stream) {
           launch_saga_grad_dense((const void*)X, (const float*)y,
-                                 (const float*)w, (const float*)alpha,
-                                 (float*)g, (int*)n, (int*)idx, (float*)e,
-                                 (int*)pos, n_rows, d, seed, round_k,
-                                 row_start, rate, objective, x_is_bf16,
-                                 (hipStream_t)stream);
+                                 (const float*)w, (float*)alpha, (float*)g,
+                                 (int*)n, (int*)idx, (float*)e, (int*)pos,
+                                 (const int*)k_dev, commit_now, n_rows, d,
+                                 seed, round_k, row_start, rate, objective,
+                                 x_is_bf16, (hipStream_t)stream);
           check(hipGetLastError(), "saga_grad_dense launch");
         });
 
   m.def("grad_csr",
         [](uintptr_t indptr, uintptr_t indices, uintptr_t values, uintptr_t y,
-           uintptr_t w, uintptr_t g, uintptr_t n, long n_rows, uint64_t seed,
-           uint32_t round_k, uint64_t row_start, double rate, int objective,
-           int v_is_bf16, uintptr_t stream) {
+           uintptr_t w, uintptr_t g, uintptr_t n, uintptr_t k_dev,
+           long n_rows, uint64_t seed, uint32_t round_k, uint64_t row_start,
+           double rate, int objective, int v_is_bf16, uintptr_t stream) {
           launch_grad_csr((const int*)indptr, (const int*)indices,
                           (const void*)values, (const float*)y,
-                          (const float*)w, (float*)g, (int*)n, n_rows, seed,
-                          round_k, row_start, rate, objective, v_is_bf16,
-                          (hipStream_t)stream);
+                          (const float*)w, (float*)g, (int*)n,
+                          (const int*)k_dev, n_rows, seed, round_k, row_start,
+                          rate, objective, v_is_bf16, (hipStream_t)stream);
           check(hipGetLastError(), "grad_csr launch");
         });
 
   m.def("saga_grad_csr",
         [](uintptr_t indptr, uintptr_t indices, uintptr_t values, uintptr_t y,
            uintptr_t w, uintptr_t alpha, uintptr_t g, uintptr_t n,
-           uintptr_t idx, uintptr_t e, uintptr_t pos, long n_rows,
-           uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
-           int objective, int v_is_bf16, uintptr_t stream) {
+           uintptr_t idx, uintptr_t e, uintptr_t pos, uintptr_t k_dev,
+           int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
+           uint64_t row_start, double rate, int objective, int v_is_bf16,
+           uintptr_t stream) {
           launch_saga_grad_csr((const int*)indptr, (const int*)indices,
                                (const void*)values, (const float*)y,
-                               (const float*)w, (const float*)alpha,
-                               (float*)g, (int*)n, (int*)idx, (float*)e,
-                               (int*)pos, n_rows, seed, round_k, row_start,
-                               rate, objective, v_is_bf16,
+                               (const float*)w, (float*)alpha, (float*)g,
+                               (int*)n, (int*)idx, (float*)e, (int*)pos,
+                               (const int*)k_dev, commit_now, n_rows, seed,
+                               round_k, row_start, rate, objective, v_is_bf16,
                                (hipStream_t)stream);
           check(hipGetLastError(), "saga_grad_csr launch");
         });
@@ -122,6 +129,25 @@ PYBIND11_MODULE(_hip_core, m) {
                        (hipStream_t)stream);
     check(hipGetLastError(), "saga_commit launch");
   });
+
+  m.def("sgd_update_fused",
+        [](uintptr_t w, uintptr_t g, uintptr_t k_dev, float gamma,
+           float inv_batch, int num_part, int d, uintptr_t stream) {
+          launch_sgd_update_fused((float*)w, (float*)g, (int*)k_dev, gamma,
+                                  inv_batch, num_part, d,
+                                  (hipStream_t)stream);
+          check(hipGetLastError(), "sgd_update_fused launch");
+        });
+
+  m.def("saga_update_fused",
+        [](uintptr_t w, uintptr_t g, uintptr_t ab, uintptr_t k_dev,
+           float gamma, float inv_batch, float inv_N, int d,
+           uintptr_t stream) {
+          launch_saga_update_fused((float*)w, (float*)g, (float*)ab,
+                                   (int*)k_dev, gamma, inv_batch, inv_N, d,
+                                   (hipStream_t)stream);
+          check(hipGetLastError(), "saga_update_fused launch");
+        });
 
   m.attr("__hip__") = true;
 }

@@ -86,11 +86,15 @@ template <typename XT, bool SAGA>
 __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     const XT* __restrict__ X, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
-    int* __restrict__ n_out, const float* __restrict__ alpha,
+    int* __restrict__ n_out, float* __restrict__ alpha,
     int* __restrict__ idx_out, float* __restrict__ e_out,
-    int* __restrict__ pos_ctr, long n_rows, int d, uint64_t seed,
-    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all,
-    int objective) {
+    int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
+    int commit_now, long n_rows, int d,
+    uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
+    int take_all, int objective) {
+  // device-resident round index (graph mode): round = *k_dev + 1, the
+  // analog of the reference's sample(false, b, seed+k+1)
+  if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   extern __shared__ float smem[];
   float* w_lds = smem;           // [d]
   float* gacc = smem + d;        // [WAVES_PER_BLOCK][d]
@@ -139,11 +143,19 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
       float e = link_residual(z, y[rr], objective);
       float coeff = e;
       if (SAGA) {
-        coeff = e - alpha[rr];
+        const float a_old = alpha[rr];
+        coeff = e - a_old;
         if (lane == 0) {
-          const int pos = atomicAdd(pos_ctr, 1);
-          idx_out[pos] = (int)rr;
-          e_out[pos] = e;
+          if (commit_now) {
+            // sequential graph mode: every round is accepted, commit the
+            // history scalar in place (a_old was read above; each row is
+            // sampled at most once per round, so no cross-row hazard)
+            alpha[rr] = e;
+          } else {
+            const int pos = atomicAdd(pos_ctr, 1);
+            idx_out[pos] = (int)rr;
+            e_out[pos] = e;
+          }
         }
       }
       ++local_count;
@@ -175,10 +187,13 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     const int* __restrict__ indptr, const int* __restrict__ indices,
     const VT* __restrict__ values, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
-    int* __restrict__ n_out, const float* __restrict__ alpha,
+    int* __restrict__ n_out, float* __restrict__ alpha,
     int* __restrict__ idx_out, float* __restrict__ e_out,
-    int* __restrict__ pos_ctr, long n_rows, uint64_t seed, uint32_t round_k,
-    uint64_t row_start, uint32_t threshold, int take_all, int objective) {
+    int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
+    int commit_now, long n_rows, uint64_t seed,
+    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all,
+    int objective) {
+  if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   int local_count = 0;
@@ -204,11 +219,16 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
       float e = link_residual(z, y[rr], objective);
       float coeff = e;
       if (SAGA) {
-        coeff = e - alpha[rr];
+        const float a_old = alpha[rr];
+        coeff = e - a_old;
         if (lane == 0) {
-          const int pos = atomicAdd(pos_ctr, 1);
-          idx_out[pos] = (int)rr;
-          e_out[pos] = e;
+          if (commit_now) {
+            alpha[rr] = e;
+          } else {
+            const int pos = atomicAdd(pos_ctr, 1);
+            idx_out[pos] = (int)rr;
+            e_out[pos] = e;
+          }
         }
       }
       ++local_count;
@@ -250,6 +270,40 @@ __global__ void saga_commit_kernel(float* __restrict__ alpha,
   if (i < n) alpha[idx[i]] = e[i];
 }
 
+// Fused device-loop update kernels (graph mode): ONE workgroup applies the
+// update, zeroes the gradient accumulator for the next round, and advances
+// the device round counter — so an unrolled sequence of
+// [grad, update] node pairs forms a complete hipGraph with no host logic.
+__global__ __launch_bounds__(1024) void sgd_update_fused_kernel(
+    float* __restrict__ w, float* __restrict__ g, int* __restrict__ k_dev,
+    float gamma, float inv_batch, int num_part, int d) {
+  const int k = *k_dev;
+  // integer division k/num_part matches the reference's Scala Int semantics
+  // (SparkASGDThread.scala:190)
+  const float gamma_k =
+      (float)((double)gamma / sqrt((double)(k / num_part + 1)));
+  for (int i = threadIdx.x; i < d; i += blockDim.x) {
+    w[i] -= gamma_k * inv_batch * g[i];
+    g[i] = 0.f;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) *k_dev = k + 1;
+}
+
+__global__ __launch_bounds__(1024) void saga_update_fused_kernel(
+    float* __restrict__ w, float* __restrict__ g,
+    float* __restrict__ alpha_bar, int* __restrict__ k_dev, float gamma,
+    float inv_batch, float inv_N, int d) {
+  for (int i = threadIdx.x; i < d; i += blockDim.x) {
+    const float gi = g[i];
+    w[i] -= gamma * (inv_batch * gi + alpha_bar[i]);
+    alpha_bar[i] += inv_N * gi;
+    g[i] = 0.f;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) *k_dev += 1;
+}
+
 // ---------------------------------------------------------------- launchers
 
 static inline int grad_grid(long n_rows) {
@@ -262,10 +316,10 @@ static inline int grad_grid(long n_rows) {
 extern "C" {
 
 void launch_grad_dense(const void* X, const float* y, const float* w,
-                       float* g_out, int* n_out, long n_rows, int d,
-                       uint64_t seed, uint32_t round_k, uint64_t row_start,
-                       double rate, int objective, int x_is_bf16,
-                       hipStream_t stream) {
+                       float* g_out, int* n_out, const int* k_dev,
+                       long n_rows, int d, uint64_t seed, uint32_t round_k,
+                       uint64_t row_start, double rate, int objective,
+                       int x_is_bf16, hipStream_t stream) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const size_t smem = (size_t)(1 + WAVES_PER_BLOCK) * d * sizeof(float);
@@ -274,24 +328,24 @@ void launch_grad_dense(const void* X, const float* y, const float* w,
     hipLaunchKernelGGL((grad_dense_kernel<__hip_bfloat16, false>), dim3(grid),
                        dim3(BLOCK), smem, stream,
                        (const __hip_bfloat16*)X, y, w, g_out, n_out, nullptr,
-                       nullptr, nullptr, nullptr, n_rows, d, seed, round_k,
-                       row_start, thr, take_all, objective);
+                       nullptr, nullptr, nullptr, k_dev, 0, n_rows,
+                       d, seed, round_k, row_start, thr, take_all, objective);
   } else {
     hipLaunchKernelGGL((grad_dense_kernel<float, false>), dim3(grid),
                        dim3(BLOCK), smem, stream, (const float*)X, y, w,
                        g_out, n_out, nullptr, nullptr, nullptr, nullptr,
-                       n_rows, d, seed, round_k, row_start, thr, take_all,
-                       objective);
+                       k_dev, 0, n_rows, d, seed, round_k,
+                       row_start, thr, take_all, objective);
   }
 }
 
 void launch_saga_grad_dense(const void* X, const float* y, const float* w,
-                            const float* alpha, float* g_out, int* n_out,
+                            float* alpha, float* g_out, int* n_out,
                             int* idx_out, float* e_out, int* pos_ctr,
-                            long n_rows, int d, uint64_t seed,
-                            uint32_t round_k, uint64_t row_start, double rate,
-                            int objective, int x_is_bf16,
-                            hipStream_t stream) {
+                            const int* k_dev, int commit_now, long n_rows,
+                            int d, uint64_t seed, uint32_t round_k,
+                            uint64_t row_start, double rate, int objective,
+                            int x_is_bf16, hipStream_t stream) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const size_t smem = (size_t)(1 + WAVES_PER_BLOCK) * d * sizeof(float);
@@ -300,21 +354,23 @@ void launch_saga_grad_dense(const void* X, const float* y, const float* w,
     hipLaunchKernelGGL((grad_dense_kernel<__hip_bfloat16, true>), dim3(grid),
                        dim3(BLOCK), smem, stream,
                        (const __hip_bfloat16*)X, y, w, g_out, n_out, alpha,
-                       idx_out, e_out, pos_ctr, n_rows, d, seed, round_k,
-                       row_start, thr, take_all, objective);
+                       idx_out, e_out, pos_ctr, k_dev, commit_now,
+                       n_rows, d, seed, round_k, row_start, thr, take_all,
+                       objective);
   } else {
     hipLaunchKernelGGL((grad_dense_kernel<float, true>), dim3(grid),
                        dim3(BLOCK), smem, stream, (const float*)X, y, w,
-                       g_out, n_out, alpha, idx_out, e_out, pos_ctr, n_rows,
-                       d, seed, round_k, row_start, thr, take_all, objective);
+                       g_out, n_out, alpha, idx_out, e_out, pos_ctr,
+                       k_dev, commit_now, n_rows, d, seed, round_k, row_start,
+                       thr, take_all, objective);
   }
 }
 
 void launch_grad_csr(const int* indptr, const int* indices, const void* values,
                      const float* y, const float* w, float* g_out, int* n_out,
-                     long n_rows, uint64_t seed, uint32_t round_k,
-                     uint64_t row_start, double rate, int objective,
-                     int v_is_bf16, hipStream_t stream) {
+                     const int* k_dev, long n_rows, uint64_t seed,
+                     uint32_t round_k, uint64_t row_start, double rate,
+                     int objective, int v_is_bf16, hipStream_t stream) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -322,24 +378,26 @@ void launch_grad_csr(const int* indptr, const int* indices, const void* values,
     hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, false>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       nullptr, nullptr, nullptr, nullptr, n_rows, seed,
-                       round_k, row_start, thr, take_all, objective);
+                       nullptr, nullptr, nullptr, nullptr, k_dev, 0,
+                       n_rows, seed, round_k, row_start, thr, take_all,
+                       objective);
   } else {
     hipLaunchKernelGGL((grad_csr_kernel<float, false>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const float*)values, y, w, g_out, n_out, nullptr,
-                       nullptr, nullptr, nullptr, n_rows, seed, round_k,
-                       row_start, thr, take_all, objective);
+                       nullptr, nullptr, nullptr, k_dev, 0, n_rows,
+                       seed, round_k, row_start, thr, take_all, objective);
   }
 }
 
 void launch_saga_grad_csr(const int* indptr, const int* indices,
                           const void* values, const float* y, const float* w,
-                          const float* alpha, float* g_out, int* n_out,
+                          float* alpha, float* g_out, int* n_out,
                           int* idx_out, float* e_out, int* pos_ctr,
-                          long n_rows, uint64_t seed, uint32_t round_k,
-                          uint64_t row_start, double rate, int objective,
-                          int v_is_bf16, hipStream_t stream) {
+                          const int* k_dev, int commit_now, long n_rows,
+                          uint64_t seed, uint32_t round_k, uint64_t row_start,
+                          double rate, int objective, int v_is_bf16,
+                          hipStream_t stream) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -347,14 +405,16 @@ void launch_saga_grad_csr(const int* indptr, const int* indices,
     hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, true>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       alpha, idx_out, e_out, pos_ctr, n_rows, seed, round_k,
-                       row_start, thr, take_all, objective);
+                       alpha, idx_out, e_out, pos_ctr, k_dev,
+                       commit_now, n_rows, seed, round_k, row_start, thr,
+                       take_all, objective);
   } else {
     hipLaunchKernelGGL((grad_csr_kernel<float, true>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const float*)values, y, w, g_out, n_out, alpha,
-                       idx_out, e_out, pos_ctr, n_rows, seed, round_k,
-                       row_start, thr, take_all, objective);
+                       idx_out, e_out, pos_ctr, k_dev, commit_now,
+                       n_rows, seed, round_k, row_start, thr, take_all,
+                       objective);
   }
 }
 
@@ -379,6 +439,21 @@ void launch_saga_commit(float* alpha, const int* idx, const float* e, int n,
   const int grid = (n + 255) / 256;
   hipLaunchKernelGGL(saga_commit_kernel, dim3(grid), dim3(256), 0, stream,
                      alpha, idx, e, n);
+}
+
+void launch_sgd_update_fused(float* w, float* g, int* k_dev, float gamma,
+                             float inv_batch, int num_part, int d,
+                             hipStream_t stream) {
+  hipLaunchKernelGGL(sgd_update_fused_kernel, dim3(1), dim3(1024), 0, stream,
+                     w, g, k_dev, gamma, inv_batch, num_part, d);
+}
+
+void launch_saga_update_fused(float* w, float* g, float* alpha_bar,
+                              int* k_dev, float gamma, float inv_batch,
+                              float inv_N, int d, hipStream_t stream) {
+  hipLaunchKernelGGL(saga_update_fused_kernel, dim3(1), dim3(1024), 0,
+                     stream, w, g, alpha_bar, k_dev, gamma, inv_batch, inv_N,
+                     d);
 }
 
 }  // extern "C"
