@@ -37,13 +37,15 @@ def _xcode(X: torch.Tensor) -> int:
 
 def grad_dense(X, y, w, out, seed, round_k, row_start, rate, obj) -> int:
     _chk(X, "X")
+    assert row_start % 4 == 0, "shard starts must be 4-aligned (philox block)" 
     _chk(y, "y", torch.float32)
     _chk(out, "out", torch.float32)
     w = w.float().contiguous()
     n_rows, d = X.shape
     n_ctr = torch.zeros(1, dtype=torch.int32, device=X.device)
     _hip_core.grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
-                         out.data_ptr(), n_ctr.data_ptr(), 0, n_rows, d, seed,
+                         out.data_ptr(), 0, n_ctr.data_ptr(), 0, n_rows, d,
+                         seed,
                          round_k & 0xFFFFFFFF, row_start, rate, obj,
                          _xcode(X), _stream())
     return int(n_ctr.item())
@@ -70,6 +72,7 @@ def grad_csr(indptr, indices, values, y, w, out, seed, round_k, row_start,
 def saga_grad_dense(X, y, w, alpha, g, seed, round_k, row_start, rate, obj
                     ) -> Tuple[torch.Tensor, torch.Tensor]:
     _chk(X, "X")
+    assert row_start % 4 == 0, "shard starts must be 4-aligned (philox block)" 
     _chk(alpha, "alpha", torch.float32)
     _chk(g, "g", torch.float32)
     w = w.float().contiguous()
@@ -79,8 +82,8 @@ def saga_grad_dense(X, y, w, alpha, g, seed, round_k, row_start, rate, obj
     e = torch.empty(cap, dtype=torch.float32, device=X.device)
     ctr = torch.zeros(2, dtype=torch.int32, device=X.device)  # [n, pos]
     _hip_core.saga_grad_dense(X.data_ptr(), y.data_ptr(), w.data_ptr(),
-                              alpha.data_ptr(), g.data_ptr(), ctr.data_ptr(),
-                              idx.data_ptr(), e.data_ptr(),
+                              alpha.data_ptr(), g.data_ptr(), 0,
+                              ctr.data_ptr(), idx.data_ptr(), e.data_ptr(),
                               ctr.data_ptr() + 4, 0, 0, n_rows, d, seed,
                               round_k & 0xFFFFFFFF, row_start, rate, obj,
                               _xcode(X), _stream())

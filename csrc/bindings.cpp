@@ -18,13 +18,16 @@
 namespace py = pybind11;
 
 extern "C" {
-void launch_grad_dense(const void*, const float*, const float*, float*, int*,
-                       const int*, long, int, uint64_t, uint32_t, uint64_t,
-                       double, int, int, hipStream_t);
+void launch_grad_dense(const void*, const float*, const float*, float*,
+                       float*, int*, const int*, long, int, uint64_t,
+                       uint32_t, uint64_t, double, int, int, hipStream_t);
 void launch_saga_grad_dense(const void*, const float*, const float*, float*,
-                            float*, int*, int*, float*, int*, const int*,
-                            int, long, int, uint64_t, uint32_t, uint64_t,
-                            double, int, int, hipStream_t);
+                            float*, float*, int*, int*, float*, int*,
+                            const int*, int, long, int, uint64_t, uint32_t,
+                            uint64_t, double, int, int, hipStream_t);
+int query_grad_grid(long);
+void launch_reduce_partials(const float*, float*, int, int, int,
+                            hipStream_t);
 void launch_grad_csr(const int*, const int*, const void*, const float*,
                      const float*, float*, int*, const int*, long, uint64_t,
                      uint32_t, uint64_t, double, int, int, hipStream_t);
@@ -52,29 +55,41 @@ PYBIND11_MODULE(_hip_core, m) {
   m.doc() = "MI355X CDNA4 kernels for asyncframework_amd";
 
   m.def("grad_dense",
-        [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t g, uintptr_t n,
-           uintptr_t k_dev, long n_rows, int d, uint64_t seed,
-           uint32_t round_k, uint64_t row_start, double rate, int objective,
-           int x_is_bf16, uintptr_t stream) {
+        [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t g,
+           uintptr_t g_part, uintptr_t n, uintptr_t k_dev, long n_rows,
+           int d, uint64_t seed, uint32_t round_k, uint64_t row_start,
+           double rate, int objective, int x_is_bf16, uintptr_t stream) {
           launch_grad_dense((const void*)X, (const float*)y, (const float*)w,
-                            (float*)g, (int*)n, (const int*)k_dev, n_rows, d,
-                            seed, round_k, row_start, rate, objective,
-                            x_is_bf16, (hipStream_t)stream);
+                            (float*)g, (float*)g_part, (int*)n,
+                            (const int*)k_dev, n_rows, d, seed, round_k,
+                            row_start, rate, objective, x_is_bf16,
+                            (hipStream_t)stream);
           check(hipGetLastError(), "grad_dense launch");
+        });
+
+  m.def("grad_grid", [](long n_rows) { return query_grad_grid(n_rows); });
+
+  m.def("reduce_partials",
+        [](uintptr_t g_part, uintptr_t g, int d, int G, int splits,
+           uintptr_t stream) {
+          launch_reduce_partials((const float*)g_part, (float*)g, d, G,
+                                 splits, (hipStream_t)stream);
+          check(hipGetLastError(), "reduce_partials launch");
         });
 
   m.def("saga_grad_dense",
         [](uintptr_t X, uintptr_t y, uintptr_t w, uintptr_t alpha, uintptr_t g,
-           uintptr_t n, uintptr_t idx, uintptr_t e, uintptr_t pos,
-           uintptr_t k_dev, int commit_now, long n_rows, int d, uint64_t seed,
-           uint32_t round_k, uint64_t row_start, double rate, int objective,
-           int x_is_bf16, uintptr_t stream) {
+           uintptr_t g_part, uintptr_t n, uintptr_t idx, uintptr_t e,
+           uintptr_t pos, uintptr_t k_dev, int commit_now, long n_rows,
+           int d, uint64_t seed, uint32_t round_k, uint64_t row_start,
+           double rate, int objective, int x_is_bf16, uintptr_t stream) {
           launch_saga_grad_dense((const void*)X, (const float*)y,
                                  (const float*)w, (float*)alpha, (float*)g,
-                                 (int*)n, (int*)idx, (float*)e, (int*)pos,
-                                 (const int*)k_dev, commit_now, n_rows, d,
-                                 seed, round_k, row_start, rate, objective,
-                                 x_is_bf16, (hipStream_t)stream);
+                                 (float*)g_part, (int*)n, (int*)idx,
+                                 (float*)e, (int*)pos, (const int*)k_dev,
+                                 commit_now, n_rows, d, seed, round_k,
+                                 row_start, rate, objective, x_is_bf16,
+                                 (hipStream_t)stream);
           check(hipGetLastError(), "saga_grad_dense launch");
         });
 

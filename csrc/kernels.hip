@@ -12,36 +12,38 @@
 //   K6 saga_update     — fused SAGA triple-axpy (:217-220).
 //   K8 philox.h        — in-kernel counter-based Bernoulli mask.
 //
-// Design notes (MI355X):
-//  * 64-wide wavefronts; each wave evaluates the Philox mask for 64 rows in
-//    one shot (one lane per row) and compresses via __ballot — at the
-//    reference's sampling rates (b = 0.01..0.1) unsampled rows cost ZERO
-//    HBM traffic, so the kernel reads only b*N*d bytes per round.
-//  * dense path: w staged in LDS once per block; per-wave fp32 gradient
-//    slabs in LDS (no atomics inside the block), one global atomicAdd pass
-//    per block at the end. LDS budget 20*d bytes -> d <= 8000 (mnist8m 784,
-//    epsilon 2000 both fit with >=4 blocks/CU).
-//  * sparse path: w stays in L2 (rcv1 d=47236 -> 189 KB, L2 is 4 MiB/XCD);
-//    gradient scatter via global fp32 atomics (~73 nnz/row).
-//  * bf16 rows are loaded as ushort4 (8 B/lane) when d % 4 == 0 — scalar
-//    bf16 loads halve effective bandwidth (CDNA guide, common mistake #2).
+// Design notes (MI355X, measured on hardware):
+//  * 64-wide wavefronts. ONE Philox eval decides FOUR consecutive rows
+//    (counter = row/4, output word = row%4) — the full-dataset mask scan is
+//    the dominant fixed cost per round (8.1M rows; 32-bit integer multiply
+//    is slow on the VALU), so each wave covers 256 rows per eval step.
+//  * row processing is latency-bound, not bandwidth-bound, at the
+//    reference's sampling rates (b=0.01 samples ~81k of 8.1M rows): rows are
+//    processed by SUB-WAVES of LPR lanes (template param 64/32/16), so a
+//    wave keeps 64/LPR rows in flight — raising memory-level parallelism
+//    without extra registers.
+//  * dense path: w staged in LDS; per-sub-wave fp32 gradient slabs in LDS
+//    (LDS = (1 + 4*64/LPR)*d*4 bytes); finalize writes per-block partials
+//    to a transposed slab, summed by reduce_partials (no global atomics:
+//    a d-wide atomicAdd finalize serialized on 784 addresses).
+//  * sparse path: w stays in L2 (rcv1 d=47236 -> 189 KB; L2 is 4 MiB/XCD);
+//    gradient scatter via global fp32 atomics (~73 nnz/row, low contention).
+//  * bf16 rows load as ushort4 (8 B/lane) — scalar bf16 loads halve
+//    effective bandwidth (CDNA guide, common mistake #2).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <cstdint>
+#include <cstdlib>
 #include "philox.h"
 
 #define WAVE 64
 #define BLOCK 256
 #define WAVES_PER_BLOCK (BLOCK / WAVE)
+#define ROWS_PER_WAVE 256  // 64 lanes x 4 rows per philox eval
+#define ROWS_PER_BLOCK_ITER (WAVES_PER_BLOCK * ROWS_PER_WAVE)
 
 // ---------------------------------------------------------------- helpers
-
-__device__ __forceinline__ float wave_reduce_sum(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
-  return v;  // every lane holds the sum
-}
 
 __device__ __forceinline__ float link_residual(float z, float yv, int obj) {
   if (obj == 1) return 1.0f / (1.0f + __expf(-z)) - yv;  // logistic
@@ -80,104 +82,153 @@ __device__ __forceinline__ void load4<__hip_bfloat16>(
 
 // ---------------------------------------------------------------- K1 (+K3)
 
-// One kernel template covers plain ASGD (SAGA=false) and SAGA (alpha gather
-// + staged (idx, e) emit). Grid-stride over 64-row groups per wave.
-template <typename XT, bool SAGA>
+// LPR = lanes per row (64, 32 or 16): rows are processed by aligned
+// sub-waves so 64/LPR rows are in flight per wave.
+template <typename XT, bool SAGA, int LPR>
 __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     const XT* __restrict__ X, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
-    int* __restrict__ n_out, float* __restrict__ alpha,
-    int* __restrict__ idx_out, float* __restrict__ e_out,
-    int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
-    int commit_now, long n_rows, int d,
+    float* __restrict__ g_part, int* __restrict__ n_out,
+    float* __restrict__ alpha, int* __restrict__ idx_out,
+    float* __restrict__ e_out, int* __restrict__ pos_ctr,
+    const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
     int take_all, int objective) {
-  // device-resident round index (graph mode): round = *k_dev + 1, the
-  // analog of the reference's sample(false, b, seed+k+1)
-  if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
+  if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;  // graph mode: round = k+1
+  constexpr int NSUB = WAVE / LPR;
+  constexpr int NSLAB = WAVES_PER_BLOCK * NSUB;
   extern __shared__ float smem[];
   float* w_lds = smem;           // [d]
-  float* gacc = smem + d;        // [WAVES_PER_BLOCK][d]
+  float* gacc = smem + d;        // [NSLAB][d]
   for (int j = threadIdx.x; j < d; j += BLOCK) {
     w_lds[j] = w[j];
-    gacc[j] = 0.f; gacc[d + j] = 0.f; gacc[2 * d + j] = 0.f;
-    gacc[3 * d + j] = 0.f;
+#pragma unroll
+    for (int s2 = 0; s2 < NSLAB; ++s2) gacc[(size_t)s2 * d + j] = 0.f;
   }
   __syncthreads();
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  float* gw = gacc + (size_t)wave * d;
+  const int sub = lane / LPR;
+  const int sl = lane % LPR;
+  float* gw = gacc + (size_t)(wave * NSUB + sub) * d;
   int local_count = 0;
-  const bool vec4 = (d % 4) == 0;
   const int d4 = d >> 2;
 
-  const long group_stride = (long)gridDim.x * BLOCK;
-  for (long base = (long)blockIdx.x * BLOCK + wave * WAVE; base < n_rows;
-       base += group_stride) {
-    const long row = base + lane;
-    bool sampled = false;
-    if (row < n_rows) {
-      sampled = take_all ||
-                philox_x0(seed, round_k, row_start + (uint64_t)row) < threshold;
+  const long gstride = (long)gridDim.x * ROWS_PER_BLOCK_ITER;
+  for (long bb = (long)blockIdx.x * ROWS_PER_BLOCK_ITER; bb < n_rows;
+       bb += gstride) {
+    const long base = bb + (long)wave * ROWS_PER_WAVE;
+    if (base >= n_rows) continue;
+    // scan 256 rows: lane's philox block covers rows base+4*lane..+3
+    // (requires row_start % 4 == 0 — enforced by the launcher/sharder)
+    const uint4 x = philox_block4(
+        seed, round_k, (row_start + (uint64_t)base) / 4 + (uint64_t)lane);
+    const long rem = n_rows - base;
+    unsigned long long m[4];
+    {
+      const long lrow = 4L * lane;
+      m[0] = __ballot(lrow + 0 < rem && (take_all || x.x < threshold));
+      m[1] = __ballot(lrow + 1 < rem && (take_all || x.y < threshold));
+      m[2] = __ballot(lrow + 2 < rem && (take_all || x.z < threshold));
+      m[3] = __ballot(lrow + 3 < rem && (take_all || x.w < threshold));
     }
-    unsigned long long m = __ballot(sampled);
-    while (m) {
-      const int bit = __ffsll((long long)m) - 1;
-      m &= m - 1;
-      const long rr = base + bit;
-      const XT* xrow = X + (size_t)rr * d;
-      float z = 0.f;
-      if (vec4) {
-        for (int j4 = lane; j4 < d4; j4 += WAVE) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      unsigned long long mm = m[i];
+      for (int s2 = 0; s2 < sub && mm; ++s2) mm &= mm - 1;  // my first bit
+      while (mm) {
+        const int bit = __ffsll((long long)mm) - 1;
+        const long rr = base + 4L * bit + i;
+        const XT* xrow = X + (size_t)rr * d;
+        float z = 0.f;
+        for (int j4 = sl; j4 < d4; j4 += LPR) {
           float xv[4];
           load4<XT>(xrow, j4, xv);
           const int j = j4 * 4;
           z += xv[0] * w_lds[j] + xv[1] * w_lds[j + 1] +
                xv[2] * w_lds[j + 2] + xv[3] * w_lds[j + 3];
         }
-      } else {
-        for (int j = lane; j < d; j += WAVE) z += to_f32<XT>(xrow[j]) * w_lds[j];
-      }
-      z = wave_reduce_sum(z);
-      float e = link_residual(z, y[rr], objective);
-      float coeff = e;
-      if (SAGA) {
-        const float a_old = alpha[rr];
-        coeff = e - a_old;
-        if (lane == 0) {
-          if (commit_now) {
-            // sequential graph mode: every round is accepted, commit the
-            // history scalar in place (a_old was read above; each row is
-            // sampled at most once per round, so no cross-row hazard)
-            alpha[rr] = e;
-          } else {
-            const int pos = atomicAdd(pos_ctr, 1);
-            idx_out[pos] = (int)rr;
-            e_out[pos] = e;
+#pragma unroll
+        for (int off = LPR / 2; off > 0; off >>= 1)
+          z += __shfl_xor(z, off, WAVE);
+        float e = link_residual(z, y[rr], objective);
+        float coeff = e;
+        if (SAGA) {
+          const float a_old = alpha[rr];
+          coeff = e - a_old;
+          if (sl == 0) {
+            if (commit_now) {
+              // sequential graph mode: every round accepted -> commit in
+              // place (each row sampled once per round, read-before-write)
+              alpha[rr] = e;
+            } else {
+              const int pos = atomicAdd(pos_ctr, 1);
+              idx_out[pos] = (int)rr;
+              e_out[pos] = e;
+            }
           }
         }
-      }
-      ++local_count;
-      if (vec4) {
-        for (int j4 = lane; j4 < d4; j4 += WAVE) {
+        ++local_count;
+        for (int j4 = sl; j4 < d4; j4 += LPR) {
           float xv[4];
           load4<XT>(xrow, j4, xv);
           const int j = j4 * 4;
           gw[j] += coeff * xv[0]; gw[j + 1] += coeff * xv[1];
           gw[j + 2] += coeff * xv[2]; gw[j + 3] += coeff * xv[3];
         }
-      } else {
-        for (int j = lane; j < d; j += WAVE) gw[j] += coeff * to_f32<XT>(xrow[j]);
+        for (int s2 = 0; s2 < NSUB && mm; ++s2) mm &= mm - 1;  // next mine
       }
     }
   }
   __syncthreads();
-  for (int j = threadIdx.x; j < d; j += BLOCK) {
-    const float s = gacc[j] + gacc[d + j] + gacc[2 * d + j] + gacc[3 * d + j];
-    if (s != 0.f) atomicAdd(&g_out[j], s);
+  if (g_part != nullptr) {
+    // plain-store partials g_part[j*G + b]; reduce_partials sums them
+    const size_t G = gridDim.x;
+    for (int j = threadIdx.x; j < d; j += BLOCK) {
+      float s = 0.f;
+#pragma unroll
+      for (int s2 = 0; s2 < NSLAB; ++s2) s += gacc[(size_t)s2 * d + j];
+      g_part[(size_t)j * G + blockIdx.x] = s;
+    }
+  } else {
+    for (int j = threadIdx.x; j < d; j += BLOCK) {
+      float s = 0.f;
+#pragma unroll
+      for (int s2 = 0; s2 < NSLAB; ++s2) s += gacc[(size_t)s2 * d + j];
+      if (s != 0.f) atomicAdd(&g_out[j], s);
+    }
   }
-  if (lane == 0 && local_count) atomicAdd(n_out, local_count);
+  if (sl == 0 && local_count) atomicAdd(n_out, local_count);
+}
+
+// Sums the per-block partial slabs into g (layout g_part[j][G], contiguous
+// per column). Grid = ceil(d/BLOCK) * SPLITS.
+__global__ __launch_bounds__(BLOCK) void reduce_partials_kernel(
+    const float* __restrict__ g_part, float* __restrict__ g_out, int d,
+    int G, int splits) {
+  const int njc = (d + BLOCK - 1) / BLOCK;
+  const int jc = blockIdx.x % njc;
+  const int sp = blockIdx.x / njc;
+  const int j = jc * BLOCK + threadIdx.x;
+  if (j >= d) return;
+  const int per = (G + splits - 1) / splits;
+  const int b0 = sp * per;
+  const int b1 = min(G, b0 + per);
+  const float* base = g_part + (size_t)j * G;
+  float s = 0.f;
+  if (((b1 - b0) & 3) == 0 && (b0 & 3) == 0) {
+    const float4* v = reinterpret_cast<const float4*>(base + b0);
+    const int n4 = (b1 - b0) >> 2;
+    for (int q = 0; q < n4; ++q) {
+      const float4 x = v[q];
+      s += x.x + x.y + x.z + x.w;
+    }
+  } else {
+    for (int b = b0; b < b1; ++b) s += base[b];
+  }
+  if (splits == 1) g_out[j] += s;
+  else atomicAdd(&g_out[j], s);
 }
 
 // ---------------------------------------------------------------- K2 (+K3)
@@ -190,50 +241,60 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     int* __restrict__ n_out, float* __restrict__ alpha,
     int* __restrict__ idx_out, float* __restrict__ e_out,
     int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
-    int commit_now, long n_rows, uint64_t seed,
-    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all,
-    int objective) {
+    int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, uint32_t threshold, int take_all, int objective) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   int local_count = 0;
-  const long group_stride = (long)gridDim.x * BLOCK;
-  for (long base = (long)blockIdx.x * BLOCK + wave * WAVE; base < n_rows;
-       base += group_stride) {
-    const long row = base + lane;
-    bool sampled = false;
-    if (row < n_rows) {
-      sampled = take_all ||
-                philox_x0(seed, round_k, row_start + (uint64_t)row) < threshold;
+  const long gstride = (long)gridDim.x * ROWS_PER_BLOCK_ITER;
+  for (long bb = (long)blockIdx.x * ROWS_PER_BLOCK_ITER; bb < n_rows;
+       bb += gstride) {
+    const long base = bb + (long)wave * ROWS_PER_WAVE;
+    if (base >= n_rows) continue;
+    const uint4 x = philox_block4(
+        seed, round_k, (row_start + (uint64_t)base) / 4 + (uint64_t)lane);
+    const long rem = n_rows - base;
+    unsigned long long m[4];
+    {
+      const long lrow = 4L * lane;
+      m[0] = __ballot(lrow + 0 < rem && (take_all || x.x < threshold));
+      m[1] = __ballot(lrow + 1 < rem && (take_all || x.y < threshold));
+      m[2] = __ballot(lrow + 2 < rem && (take_all || x.z < threshold));
+      m[3] = __ballot(lrow + 3 < rem && (take_all || x.w < threshold));
     }
-    unsigned long long m = __ballot(sampled);
-    while (m) {
-      const int bit = __ffsll((long long)m) - 1;
-      m &= m - 1;
-      const long rr = base + bit;
-      const int s = indptr[rr], t = indptr[rr + 1];
-      float z = 0.f;
-      for (int p = s + lane; p < t; p += WAVE)
-        z += to_f32<VT>(values[p]) * w[indices[p]];
-      z = wave_reduce_sum(z);
-      float e = link_residual(z, y[rr], objective);
-      float coeff = e;
-      if (SAGA) {
-        const float a_old = alpha[rr];
-        coeff = e - a_old;
-        if (lane == 0) {
-          if (commit_now) {
-            alpha[rr] = e;
-          } else {
-            const int pos = atomicAdd(pos_ctr, 1);
-            idx_out[pos] = (int)rr;
-            e_out[pos] = e;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      unsigned long long mm = m[i];
+      while (mm) {
+        const int bit = __ffsll((long long)mm) - 1;
+        mm &= mm - 1;
+        const long rr = base + 4L * bit + i;
+        const int s = indptr[rr], t = indptr[rr + 1];
+        float z = 0.f;
+        for (int p = s + lane; p < t; p += WAVE)
+          z += to_f32<VT>(values[p]) * w[indices[p]];
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) z += __shfl_xor(z, off, WAVE);
+        float e = link_residual(z, y[rr], objective);
+        float coeff = e;
+        if (SAGA) {
+          const float a_old = alpha[rr];
+          coeff = e - a_old;
+          if (lane == 0) {
+            if (commit_now) {
+              alpha[rr] = e;
+            } else {
+              const int pos = atomicAdd(pos_ctr, 1);
+              idx_out[pos] = (int)rr;
+              e_out[pos] = e;
+            }
           }
         }
+        ++local_count;
+        for (int p = s + lane; p < t; p += WAVE)
+          atomicAdd(&g_out[indices[p]], coeff * to_f32<VT>(values[p]));
       }
-      ++local_count;
-      for (int p = s + lane; p < t; p += WAVE)
-        atomicAdd(&g_out[indices[p]], coeff * to_f32<VT>(values[p]));
     }
   }
   if (lane == 0 && local_count) atomicAdd(n_out, local_count);
@@ -273,7 +334,8 @@ __global__ void saga_commit_kernel(float* __restrict__ alpha,
 // Fused device-loop update kernels (graph mode): ONE workgroup applies the
 // update, zeroes the gradient accumulator for the next round, and advances
 // the device round counter — so an unrolled sequence of
-// [grad, update] node pairs forms a complete hipGraph with no host logic.
+// [grad, reduce, update] node triples forms a complete hipGraph with no
+// host logic.
 __global__ __launch_bounds__(1024) void sgd_update_fused_kernel(
     float* __restrict__ w, float* __restrict__ g, int* __restrict__ k_dev,
     float gamma, float inv_batch, int num_part, int d) {
@@ -307,63 +369,101 @@ __global__ __launch_bounds__(1024) void saga_update_fused_kernel(
 // ---------------------------------------------------------------- launchers
 
 static inline int grad_grid(long n_rows) {
-  long g = (n_rows + BLOCK - 1) / BLOCK;
-  if (g > 4096) g = 4096;   // grid-stride beyond this
+  static int override_grid = [] {
+    const char* s = std::getenv("ASYNCAMD_GRAD_GRID");
+    return s ? std::atoi(s) : 0;
+  }();
+  if (override_grid > 0) return override_grid;
+  long g = (n_rows + ROWS_PER_BLOCK_ITER - 1) / ROWS_PER_BLOCK_ITER;
+  if (g > 2048) g = 2048;
   if (g < 1) g = 1;
   return (int)g;
 }
 
-extern "C" {
+static inline int pick_lpr(int d) {
+  static int override_lpr = [] {
+    const char* s = std::getenv("ASYNCAMD_LPR");
+    return s ? std::atoi(s) : 0;
+  }();
+  if (override_lpr == 64 || override_lpr == 32 || override_lpr == 16)
+    return override_lpr;
+  // LDS = (1 + 4*64/LPR)*d*4 bytes; keep >= 2 blocks/CU (160 KiB LDS)
+  if (d <= 1200) return 16;   // 53 KiB at d=784
+  if (d <= 4000) return 32;   // 72 KiB at d=2000
+  return 64;
+}
 
-void launch_grad_dense(const void* X, const float* y, const float* w,
-                       float* g_out, int* n_out, const int* k_dev,
-                       long n_rows, int d, uint64_t seed, uint32_t round_k,
-                       uint64_t row_start, double rate, int objective,
-                       int x_is_bf16, hipStream_t stream) {
+template <typename XT, bool SAGA>
+static void launch_dense(const XT* X, const float* y, const float* w,
+                         float* g_out, float* g_part, int* n_out,
+                         float* alpha, int* idx_out, float* e_out,
+                         int* pos_ctr, const int* k_dev, int commit_now,
+                         long n_rows, int d, uint64_t seed, uint32_t round_k,
+                         uint64_t row_start, double rate, int objective,
+                         hipStream_t stream) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
-  const size_t smem = (size_t)(1 + WAVES_PER_BLOCK) * d * sizeof(float);
   const int grid = grad_grid(n_rows);
-  if (x_is_bf16) {
-    hipLaunchKernelGGL((grad_dense_kernel<__hip_bfloat16, false>), dim3(grid),
-                       dim3(BLOCK), smem, stream,
-                       (const __hip_bfloat16*)X, y, w, g_out, n_out, nullptr,
-                       nullptr, nullptr, nullptr, k_dev, 0, n_rows,
-                       d, seed, round_k, row_start, thr, take_all, objective);
-  } else {
-    hipLaunchKernelGGL((grad_dense_kernel<float, false>), dim3(grid),
-                       dim3(BLOCK), smem, stream, (const float*)X, y, w,
-                       g_out, n_out, nullptr, nullptr, nullptr, nullptr,
-                       k_dev, 0, n_rows, d, seed, round_k,
-                       row_start, thr, take_all, objective);
-  }
+  const int lpr = pick_lpr(d);
+  const size_t smem = (size_t)(1 + 4 * (WAVE / lpr)) * d * sizeof(float);
+#define DISPATCH_LPR(L)                                                      \
+  hipLaunchKernelGGL((grad_dense_kernel<XT, SAGA, L>), dim3(grid),           \
+                     dim3(BLOCK), smem, stream, X, y, w, g_out, g_part,      \
+                     n_out, alpha, idx_out, e_out, pos_ctr, k_dev,           \
+                     commit_now, n_rows, d, seed, round_k, row_start, thr,   \
+                     take_all, objective)
+  if (lpr == 16) DISPATCH_LPR(16);
+  else if (lpr == 32) DISPATCH_LPR(32);
+  else DISPATCH_LPR(64);
+#undef DISPATCH_LPR
+}
+
+extern "C" {
+
+int query_grad_grid(long n_rows) { return grad_grid(n_rows); }
+
+void launch_grad_dense(const void* X, const float* y, const float* w,
+                       float* g_out, float* g_part, int* n_out,
+                       const int* k_dev, long n_rows, int d, uint64_t seed,
+                       uint32_t round_k, uint64_t row_start, double rate,
+                       int objective, int x_is_bf16, hipStream_t stream) {
+  if (x_is_bf16)
+    launch_dense<__hip_bfloat16, false>(
+        (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, nullptr,
+        nullptr, nullptr, nullptr, k_dev, 0, n_rows, d, seed, round_k,
+        row_start, rate, objective, stream);
+  else
+    launch_dense<float, false>((const float*)X, y, w, g_out, g_part, n_out,
+                               nullptr, nullptr, nullptr, nullptr, k_dev, 0,
+                               n_rows, d, seed, round_k, row_start, rate,
+                               objective, stream);
 }
 
 void launch_saga_grad_dense(const void* X, const float* y, const float* w,
-                            float* alpha, float* g_out, int* n_out,
-                            int* idx_out, float* e_out, int* pos_ctr,
-                            const int* k_dev, int commit_now, long n_rows,
-                            int d, uint64_t seed, uint32_t round_k,
-                            uint64_t row_start, double rate, int objective,
-                            int x_is_bf16, hipStream_t stream) {
-  const uint32_t thr = philox_threshold(rate);
-  const int take_all = rate >= 1.0;
-  const size_t smem = (size_t)(1 + WAVES_PER_BLOCK) * d * sizeof(float);
-  const int grid = grad_grid(n_rows);
-  if (x_is_bf16) {
-    hipLaunchKernelGGL((grad_dense_kernel<__hip_bfloat16, true>), dim3(grid),
-                       dim3(BLOCK), smem, stream,
-                       (const __hip_bfloat16*)X, y, w, g_out, n_out, alpha,
-                       idx_out, e_out, pos_ctr, k_dev, commit_now,
-                       n_rows, d, seed, round_k, row_start, thr, take_all,
-                       objective);
-  } else {
-    hipLaunchKernelGGL((grad_dense_kernel<float, true>), dim3(grid),
-                       dim3(BLOCK), smem, stream, (const float*)X, y, w,
-                       g_out, n_out, alpha, idx_out, e_out, pos_ctr,
-                       k_dev, commit_now, n_rows, d, seed, round_k, row_start,
-                       thr, take_all, objective);
-  }
+                            float* alpha, float* g_out, float* g_part,
+                            int* n_out, int* idx_out, float* e_out,
+                            int* pos_ctr, const int* k_dev, int commit_now,
+                            long n_rows, int d, uint64_t seed,
+                            uint32_t round_k, uint64_t row_start, double rate,
+                            int objective, int x_is_bf16,
+                            hipStream_t stream) {
+  if (x_is_bf16)
+    launch_dense<__hip_bfloat16, true>(
+        (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, alpha, idx_out,
+        e_out, pos_ctr, k_dev, commit_now, n_rows, d, seed, round_k,
+        row_start, rate, objective, stream);
+  else
+    launch_dense<float, true>((const float*)X, y, w, g_out, g_part, n_out,
+                              alpha, idx_out, e_out, pos_ctr, k_dev,
+                              commit_now, n_rows, d, seed, round_k, row_start,
+                              rate, objective, stream);
+}
+
+void launch_reduce_partials(const float* g_part, float* g_out, int d, int G,
+                            int splits, hipStream_t stream) {
+  const int njc = (d + BLOCK - 1) / BLOCK;
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(njc * splits), dim3(BLOCK),
+                     0, stream, g_part, g_out, d, G, splits);
 }
 
 void launch_grad_csr(const int* indptr, const int* indices, const void* values,
@@ -378,15 +478,14 @@ void launch_grad_csr(const int* indptr, const int* indices, const void* values,
     hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, false>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       nullptr, nullptr, nullptr, nullptr, k_dev, 0,
-                       n_rows, seed, round_k, row_start, thr, take_all,
-                       objective);
+                       nullptr, nullptr, nullptr, nullptr, k_dev, 0, n_rows,
+                       seed, round_k, row_start, thr, take_all, objective);
   } else {
     hipLaunchKernelGGL((grad_csr_kernel<float, false>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const float*)values, y, w, g_out, n_out, nullptr,
-                       nullptr, nullptr, nullptr, k_dev, 0, n_rows,
-                       seed, round_k, row_start, thr, take_all, objective);
+                       nullptr, nullptr, nullptr, k_dev, 0, n_rows, seed,
+                       round_k, row_start, thr, take_all, objective);
   }
 }
 
@@ -405,16 +504,15 @@ void launch_saga_grad_csr(const int* indptr, const int* indices,
     hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, true>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       alpha, idx_out, e_out, pos_ctr, k_dev,
-                       commit_now, n_rows, seed, round_k, row_start, thr,
-                       take_all, objective);
+                       alpha, idx_out, e_out, pos_ctr, k_dev, commit_now,
+                       n_rows, seed, round_k, row_start, thr, take_all,
+                       objective);
   } else {
     hipLaunchKernelGGL((grad_csr_kernel<float, true>), dim3(grid),
                        dim3(BLOCK), 0, stream, indptr, indices,
                        (const float*)values, y, w, g_out, n_out, alpha,
-                       idx_out, e_out, pos_ctr, k_dev, commit_now,
-                       n_rows, seed, round_k, row_start, thr, take_all,
-                       objective);
+                       idx_out, e_out, pos_ctr, k_dev, commit_now, n_rows,
+                       seed, round_k, row_start, thr, take_all, objective);
   }
 }
 

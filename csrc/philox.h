@@ -8,6 +8,33 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 
+// Full 4-word output for counter = row BLOCK (row/4): one eval decides four
+// consecutive rows (word = row%4). Must match utils/philox.py bernoulli_mask.
+__device__ __forceinline__ uint4 philox_block4(uint64_t seed, uint32_t round_k,
+                                               uint64_t row_block) {
+  uint32_t c0 = (uint32_t)(row_block & 0xFFFFFFFFull);
+  uint32_t c1 = (uint32_t)(row_block >> 32);
+  uint32_t c2 = round_k;
+  uint32_t c3 = 0u;
+  uint32_t k0 = (uint32_t)(seed & 0xFFFFFFFFull);
+  uint32_t k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    uint64_t p0 = 0xD2511F53ull * (uint64_t)c0;
+    uint64_t p1 = 0xCD9E8D57ull * (uint64_t)c2;
+    uint32_t hi0 = (uint32_t)(p0 >> 32), lo0 = (uint32_t)p0;
+    uint32_t hi1 = (uint32_t)(p1 >> 32), lo1 = (uint32_t)p1;
+    uint32_t n0 = hi1 ^ c1 ^ k0;
+    uint32_t n1 = lo1;
+    uint32_t n2 = hi0 ^ c3 ^ k1;
+    uint32_t n3 = lo0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  return make_uint4(c0, c1, c2, c3);
+}
+
 __device__ __forceinline__ uint32_t philox_x0(uint64_t seed, uint32_t round_k,
                                               uint64_t row) {
   uint32_t c0 = (uint32_t)(row & 0xFFFFFFFFull);
