@@ -149,6 +149,8 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
           z += xv[0] * w_lds[j] + xv[1] * w_lds[j + 1] +
                xv[2] * w_lds[j + 2] + xv[3] * w_lds[j + 3];
         }
+        for (int j = d4 * 4 + sl; j < d; j += LPR)  // d % 4 tail
+          z += to_f32<XT>(xrow[j]) * w_lds[j];
 #pragma unroll
         for (int off = LPR / 2; off > 0; off >>= 1)
           z += __shfl_xor(z, off, WAVE);
@@ -177,6 +179,8 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
           gw[j] += coeff * xv[0]; gw[j + 1] += coeff * xv[1];
           gw[j + 2] += coeff * xv[2]; gw[j + 3] += coeff * xv[3];
         }
+        for (int j = d4 * 4 + sl; j < d; j += LPR)  // d % 4 tail
+          gw[j] += coeff * to_f32<XT>(xrow[j]);
         for (int s2 = 0; s2 < NSUB && mm; ++s2) mm &= mm - 1;  // next mine
       }
     }
@@ -200,6 +204,201 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
+}
+
+// ------------------------------------------------- K1 pipelined (queue)
+//
+// Two-phase dense gradient: (1) Philox scan appends sampled local row ids to
+// an LDS queue (block-shared, atomic append; the near-full guard makes
+// overflow impossible), (2) waves drain the queue round-robin with a
+// DEPTH-deep software pipeline (statically unrolled — runtime-indexed
+// register arrays go to scratch, CDNA guide rule 20): while row q computes,
+// DEPTH-1 later rows' loads are in flight, and the accumulate pass reuses
+// the registers instead of re-reading X. This is the latency fix: at b=0.01
+// a wave sees ~1 sampled row per 256-row scan group, so the non-queued
+// kernel paid a full HBM latency per row (measured ~7 us serial per row).
+
+// queue capacity must exceed rows-per-scan-iteration (the near-full guard
+// is `qn >= QCAP - RPB`; QCAP == RPB would never scan and livelock)
+#define QCAP_FOR(PB) ((PB) == 512 ? 4096 : 2048)
+#define PIPE_MAXIT 8  // supports d <= 2048, d % 4 == 0
+
+template <typename XT> struct RowVec;
+template <> struct RowVec<float> { using T = float4; };
+template <> struct RowVec<__hip_bfloat16> { using T = ushort4; };
+
+__device__ __forceinline__ void cvt4(const float4& r, float o[4]) {
+  o[0] = r.x; o[1] = r.y; o[2] = r.z; o[3] = r.w;
+}
+__device__ __forceinline__ void cvt4(const ushort4& r, float o[4]) {
+  union { unsigned short u; __hip_bfloat16 b; } c0{r.x}, c1{r.y}, c2{r.z},
+      c3{r.w};
+  o[0] = __bfloat162float(c0.b); o[1] = __bfloat162float(c1.b);
+  o[2] = __bfloat162float(c2.b); o[3] = __bfloat162float(c3.b);
+}
+
+template <typename XT>
+__device__ __forceinline__ void load_row_regs(
+    const XT* __restrict__ xrow, int lane, int d4,
+    typename RowVec<XT>::T raw[PIPE_MAXIT]) {
+#pragma unroll
+  for (int it = 0; it < PIPE_MAXIT; ++it) {
+    const int j4 = lane + it * WAVE;
+    if (j4 < d4)
+      raw[it] = reinterpret_cast<const typename RowVec<XT>::T*>(xrow)[j4];
+  }
+}
+
+template <typename XT, bool SAGA, int PBLOCK, int DEPTH>
+__global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
+    const XT* __restrict__ X, const float* __restrict__ y,
+    const float* __restrict__ w, float* __restrict__ g_out,
+    float* __restrict__ g_part, int* __restrict__ n_out,
+    float* __restrict__ alpha, int* __restrict__ idx_out,
+    float* __restrict__ e_out, int* __restrict__ pos_ctr,
+    const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
+    uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
+    int take_all, int objective) {
+  if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
+  constexpr int NW = PBLOCK / WAVE;            // waves per block
+  constexpr int RPB = NW * ROWS_PER_WAVE;      // rows scanned per block-iter
+  extern __shared__ float smem[];
+  float* w_lds = smem;                          // [d]
+  float* gacc = smem + d;                       // [NW][d]
+  constexpr int QCAP = QCAP_FOR(PBLOCK);
+  static_assert(QCAP >= RPB + 1024, "queue must out-size one scan iter");
+  int* rowq = (int*)(smem + (size_t)(1 + NW) * d);  // [QCAP]
+  int* qn = rowq + QCAP;
+  for (int j = threadIdx.x; j < d; j += PBLOCK) {
+    w_lds[j] = w[j];
+#pragma unroll
+    for (int s2 = 0; s2 < NW; ++s2) gacc[(size_t)s2 * d + j] = 0.f;
+  }
+  if (threadIdx.x == 0) *qn = 0;
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float* gw = gacc + (size_t)wave * d;
+  int local_count = 0;
+  const int d4 = d >> 2;
+  using RV = typename RowVec<XT>::T;
+
+  const long ngroups = (n_rows + RPB - 1) / RPB;
+  long gi = blockIdx.x;
+  bool done = false;
+  while (!done) {
+    // ---- scan phase: fill the queue until near-full or rows exhausted
+    while (true) {
+      __syncthreads();
+      if (gi >= ngroups || *qn >= QCAP - RPB) break;
+      const long base = gi * (long)RPB + (long)wave * ROWS_PER_WAVE;
+      if (base < n_rows) {
+        const uint4 x = philox_block4(
+            seed, round_k, (row_start + (uint64_t)base) / 4 + (uint64_t)lane);
+        const long rem = n_rows - base;
+        const long lrow = base + 4L * lane;
+        const uint32_t xs[4] = {x.x, x.y, x.z, x.w};
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          if (4L * lane + i < rem && (take_all || xs[i] < threshold)) {
+            const int pos = atomicAdd(qn, 1);
+            rowq[pos] = (int)(lrow + i);
+          }
+        }
+      }
+      gi += gridDim.x;
+    }
+    __syncthreads();
+    const int nq = min(*qn, QCAP);
+    // ---- process phase: wave-strided, DEPTH-deep pipeline (static bufs)
+    RV buf[DEPTH][PIPE_MAXIT];
+#pragma unroll
+    for (int p = 0; p < DEPTH; ++p) {
+      const int q = wave + p * NW;
+      if (q < nq) load_row_regs<XT>(X + (size_t)rowq[q] * d, lane, d4, buf[p]);
+    }
+    int q_base = wave;
+    while (q_base < nq) {
+#pragma unroll
+      for (int p = 0; p < DEPTH; ++p) {
+        const int q = q_base + p * NW;
+        if (q < nq) {
+          const int rr = rowq[q];
+          float z = 0.f;
+#pragma unroll
+          for (int it = 0; it < PIPE_MAXIT; ++it) {
+            const int j4 = lane + it * WAVE;
+            if (j4 < d4) {
+              float o[4];
+              cvt4(buf[p][it], o);
+              // b128 LDS read: 16-lane groups hit distinct banks (b32 had
+              // lanes l and l+8 colliding -> 26% LDSBankConflict measured)
+              const float4 wv = reinterpret_cast<const float4*>(w_lds)[j4];
+              z += o[0] * wv.x + o[1] * wv.y + o[2] * wv.z + o[3] * wv.w;
+            }
+          }
+#pragma unroll
+          for (int off = 32; off > 0; off >>= 1)
+            z += __shfl_xor(z, off, WAVE);
+          float e = link_residual(z, y[rr], objective);
+          float coeff = e;
+          if (SAGA) {
+            const float a_old = alpha[rr];
+            coeff = e - a_old;
+            if (lane == 0) {
+              if (commit_now) {
+                alpha[rr] = e;
+              } else {
+                const int pos = atomicAdd(pos_ctr, 1);
+                idx_out[pos] = rr;
+                e_out[pos] = e;
+              }
+            }
+          }
+          ++local_count;
+#pragma unroll
+          for (int it = 0; it < PIPE_MAXIT; ++it) {
+            const int j4 = lane + it * WAVE;
+            if (j4 < d4) {
+              float o[4];
+              cvt4(buf[p][it], o);
+              float4* gw4 = reinterpret_cast<float4*>(gw);
+              float4 cur = gw4[j4];
+              cur.x += coeff * o[0]; cur.y += coeff * o[1];
+              cur.z += coeff * o[2]; cur.w += coeff * o[3];
+              gw4[j4] = cur;
+            }
+          }
+          // refill this buffer DEPTH rows ahead
+          const int qf = q + DEPTH * NW;
+          if (qf < nq)
+            load_row_regs<XT>(X + (size_t)rowq[qf] * d, lane, d4, buf[p]);
+        }
+      }
+      q_base += DEPTH * NW;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) *qn = 0;
+    done = gi >= ngroups;
+  }
+  __syncthreads();
+  if (g_part != nullptr) {
+    const size_t G = gridDim.x;
+    for (int j = threadIdx.x; j < d; j += PBLOCK) {
+      float s = 0.f;
+#pragma unroll
+      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * d + j];
+      g_part[(size_t)j * G + blockIdx.x] = s;
+    }
+  } else {
+    for (int j = threadIdx.x; j < d; j += PBLOCK) {
+      float s = 0.f;
+#pragma unroll
+      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * d + j];
+      if (s != 0.f) atomicAdd(&g_out[j], s);
+    }
+  }
+  if (lane == 0 && local_count) atomicAdd(n_out, local_count);
 }
 
 // Sums the per-block partial slabs into g (layout g_part[j][G], contiguous
@@ -369,27 +568,22 @@ __global__ __launch_bounds__(1024) void saga_update_fused_kernel(
 // ---------------------------------------------------------------- launchers
 
 static inline int grad_grid(long n_rows) {
-  static int override_grid = [] {
-    const char* s = std::getenv("ASYNCAMD_GRAD_GRID");
-    return s ? std::atoi(s) : 0;
-  }();
+  const char* s = std::getenv("ASYNCAMD_GRAD_GRID");  // re-read: sweeps
+  const int override_grid = s ? std::atoi(s) : 0;
   if (override_grid > 0) return override_grid;
   long g = (n_rows + ROWS_PER_BLOCK_ITER - 1) / ROWS_PER_BLOCK_ITER;
-  if (g > 2048) g = 2048;
+  if (g > 512) g = 512;  // measured optimum on 8.1M rows (pipe kernel)
   if (g < 1) g = 1;
   return (int)g;
 }
 
 static inline int pick_lpr(int d) {
-  static int override_lpr = [] {
-    const char* s = std::getenv("ASYNCAMD_LPR");
-    return s ? std::atoi(s) : 0;
-  }();
+  const char* s = std::getenv("ASYNCAMD_LPR");
+  const int override_lpr = s ? std::atoi(s) : 0;
   if (override_lpr == 64 || override_lpr == 32 || override_lpr == 16)
     return override_lpr;
-  // LDS = (1 + 4*64/LPR)*d*4 bytes; keep >= 2 blocks/CU (160 KiB LDS)
-  if (d <= 1200) return 16;   // 53 KiB at d=784
-  if (d <= 4000) return 32;   // 72 KiB at d=2000
+  // measured on the mnist8m shape: LPR 64 > 32 > 16 (sub-wave rows cost LDS
+  // occupancy more than the extra MLP buys) — keep whole-wave rows
   return 64;
 }
 
@@ -404,6 +598,27 @@ static void launch_dense(const XT* X, const float* y, const float* w,
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
+  const char* np = std::getenv("ASYNCAMD_NO_PIPE");
+  const bool pipe_ok = (d % 4 == 0) && (d <= 4 * WAVE * PIPE_MAXIT) &&
+                       !(np && np[0] == '1');
+  if (pipe_ok) {
+    const char* pb = std::getenv("ASYNCAMD_PBLOCK");
+    const int pblock = pb ? std::atoi(pb) : 256;  // measured best
+#define LAUNCH_PIPE(PB, DP)                                                  \
+    do {                                                                     \
+      const size_t smem = (size_t)(1 + PB / WAVE) * d * sizeof(float) +      \
+                          (QCAP_FOR(PB) + 1) * sizeof(int);                  \
+      hipLaunchKernelGGL((grad_dense_pipe_kernel<XT, SAGA, PB, DP>),         \
+                         dim3(grid), dim3(PB), smem, stream, X, y, w, g_out, \
+                         g_part, n_out, alpha, idx_out, e_out, pos_ctr,      \
+                         k_dev, commit_now, n_rows, d, seed, round_k,        \
+                         row_start, thr, take_all, objective);               \
+    } while (0)
+    if (pblock == 256) LAUNCH_PIPE(256, 4);
+    else LAUNCH_PIPE(512, 4);
+#undef LAUNCH_PIPE
+    return;
+  }
   const int lpr = pick_lpr(d);
   const size_t smem = (size_t)(1 + 4 * (WAVE / lpr)) * d * sizeof(float);
 #define DISPATCH_LPR(L)                                                      \
