@@ -1,0 +1,72 @@
+"""Checkpoint / resume of a run's optimizer state.
+
+The reference has NO framework-level checkpointing (optimizer
+state lives in driver memory only; its pitch is that SAGA's history replaces
+lineage-based recovery). The MI355X rebuild adds a cheap periodic snapshot of
+``(w, alpha_bar, per-worker history tables, k, clock)`` — off the hot path:
+tensors are cloned to CPU on the updater thread's interval, serialized with
+torch.save."""
+
+from __future__ import annotations
+
+import os
+import tempfile
+from typing import Dict, List, Optional
+
+import torch
+
+from .config import EngineConfig
+from .server import Server
+from .worker import Worker
+
+
+def capture_state(server: Server, workers: Optional[List[Worker]] = None
+                  ) -> Dict:
+    """Snapshot everything needed to resume (host-resident tensors)."""
+    state = {
+        "k": server.k,
+        "current_time": server.AC.getCurrentTime(),
+        "w": server.w.detach().cpu().clone(),
+        "alpha_bar": (server.alpha_bar.detach().cpu().clone()
+                      if server.alpha_bar is not None else None),
+        "cfg": server.cfg.__dict__.copy(),
+        "alpha": {},
+    }
+    for wk in workers or []:
+        if wk.alpha is not None:
+            state["alpha"][wk.id] = wk.alpha.detach().cpu().clone()
+    return state
+
+
+def save_checkpoint(path: str, server: Server,
+                    workers: Optional[List[Worker]] = None) -> None:
+    """Atomic write (tmp + rename) so a crash mid-save keeps the previous
+    checkpoint valid."""
+    state = capture_state(server, workers)
+    d = os.path.dirname(os.path.abspath(path))
+    os.makedirs(d, exist_ok=True)
+    fd, tmp = tempfile.mkstemp(dir=d, suffix=".ckpt.tmp")
+    os.close(fd)
+    try:
+        torch.save(state, tmp)
+        os.replace(tmp, path)
+    finally:
+        if os.path.exists(tmp):
+            os.unlink(tmp)
+
+
+def load_checkpoint(path: str) -> Dict:
+    return torch.load(path, map_location="cpu", weights_only=False)
+
+
+def restore(server: Server, workers: Optional[List[Worker]], state: Dict
+            ) -> None:
+    """Load a snapshot back into a server + workers (devices preserved)."""
+    server.k = int(state["k"])
+    server.AC.setCurrentTime(int(state["current_time"]))
+    server.w.copy_(state["w"].to(server.w.device))
+    if state.get("alpha_bar") is not None and server.alpha_bar is not None:
+        server.alpha_bar.copy_(state["alpha_bar"].to(server.alpha_bar.device))
+    for wk in workers or []:
+        if wk.alpha is not None and wk.id in state["alpha"]:
+            wk.alpha.copy_(state["alpha"][wk.id].to(wk.alpha.device))

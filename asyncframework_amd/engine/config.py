@@ -39,6 +39,8 @@ class EngineConfig:
     history_placement: str = "device"  # 'device' (HBM) | 'host' (pinned DRAM)
     calib_factor: int = 100         # delay calibration window = calib_factor*P
     snapshot_weights: bool = True   # record optVars (time, w) for loss curves
+    checkpoint_path: str = ""       # periodic optimizer-state snapshots
+    checkpoint_every: int = 0       # every N applied updates (0 = off)
 
     def torch_dtype(self) -> torch.dtype:
         return _DTYPES[self.dtype]

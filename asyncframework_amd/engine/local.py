@@ -104,12 +104,16 @@ class AsyncEngine:
         self._pending_ev = threading.Event()
         self._pending_ev.set()
         self.staleness_seen: List[int] = []
+        self.accepted_staleness: List[int] = []
         self.applied = 0
         self.rejected = 0
         self.verbose = True
         # bench hooks: wall-clock stamps taken when k first reaches a mark
         self.mark_at = set()
         self.marks = {}
+
+    def _local_workers(self):
+        return [ch.worker for ch in self.channels if hasattr(ch, "worker")]
 
     # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
     def _updater(self):
@@ -134,6 +138,7 @@ class AsyncEngine:
                 else:
                     self.staleness_seen.append(pr.getStaleness())
                 if self.accepts_now(pr):
+                    self.accepted_staleness.append(self.staleness_seen[-1])
                     srv.finish_time[wid] = now
                     sub = srv.submit_time.get(wid)
                     if sub is not None:
@@ -149,6 +154,11 @@ class AsyncEngine:
                     self.applied += 1
                     if srv.k in self.mark_at:
                         self.marks[srv.k] = time.perf_counter()
+                    if (cfg.checkpoint_every > 0 and cfg.checkpoint_path
+                            and srv.k % cfg.checkpoint_every == 0):
+                        from .checkpoint import save_checkpoint
+                        save_checkpoint(cfg.checkpoint_path, srv,
+                                        self._local_workers())
                 else:
                     srv.last_accept[wid] = False
                     self.pending.append(wid)
