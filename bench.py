@@ -1,21 +1,20 @@
 #!/usr/bin/env python3
-"""bench.py — flagship benchmark: gradient updates/sec (whole node),
-mnist8m-shape ASGD (BASELINE.json metric/config).
+"""bench.py — driver-contract benchmark: gradient updates/sec (whole node).
 
-Contract (driver): ``python bench.py --gpus N --steps K --warmup W``.
-For N>1 the driver launches one rank per GPU via torch.distributed.run; each
-rank reads RANK/LOCAL_RANK/WORLD_SIZE from the env. A *step* is one applied
-gradient update at the parameter server (the reference's iteration ``k``,
-SparkASGDThread.scala:199). W warmup updates run untimed; then EXACTLY K
-updates are timed between wall-clock marks taken by the single-writer server
-thread, bracketed by a global barrier + torch.cuda.synchronize on both
-sides. value = K / elapsed = whole-node applied updates/sec (all N GPUs feed
-the same server). Rank 0 prints one JSON line.
+``python bench.py --gpus N --steps K --warmup W [--model M]``
 
-Dataset: synthetic mnist8m-shape (8,100,000 x 784) dense bf16, random-init
-weights (no network for real datasets — BASELINE.md). The dataset is FIXED
-as N grows (sharded over workers) => strong scaling, matching the
-reference's fixed-dataset/more-partitions semantics.
+A *step* is one applied gradient update at the parameter server (the
+reference's iteration ``k``, SparkASGDThread.scala:199). W warmup updates run
+untimed, then EXACTLY K updates are timed (wall stamps taken by the
+single-writer server), bracketed by barrier + torch.cuda.synchronize on both
+sides; rank 0 prints ONE JSON line; value = whole-node updates/sec.
+
+Default model = BASELINE config 2: ASGD on synthetic mnist8m-shape
+(8.1M x 784 dense bf16, b=0.01, tau=20M, random-init weights). The dataset
+stays FIXED as N grows (sharded over ranks) => strong scaling, matching the
+reference's fixed-dataset/more-partitions semantics. The other BASELINE
+configs (1,3,4,5) are selectable via --model (single-node; N>1 supports the
+flagship dense ASGD/ASAGA models).
 """
 
 from __future__ import annotations
@@ -28,141 +27,84 @@ import time
 
 import torch
 
+from asyncframework_amd import run as runner
 from asyncframework_amd.data.shard import row_shards
-from asyncframework_amd.data.synthetic import synthetic_dense
+from asyncframework_amd.data.synthetic import synthetic_csr, synthetic_dense
 from asyncframework_amd.engine.config import EngineConfig
-from asyncframework_amd.engine.local import AsyncEngine
+from asyncframework_amd.engine.local import AsyncEngine, SyncEngine
 from asyncframework_amd.engine.server import Server
 from asyncframework_amd.engine.worker import Shard, Worker
 
-BASE = dict(rows=8_100_000, cols=784, rate=0.01, taw=20_000_000,
-            gamma=1.5625e-3, bucket_ratio=0.7, seed=42)
+BASE = dict(taw=20_000_000, gamma=1.5625e-3, bucket_ratio=0.7, seed=42)
+
+MODELS = {
+    # BASELINE.json configs. config 1 is the CPU plumbing check.
+    "sync-tiny-cpu": dict(rows=1_000, cols=784, rate=0.3, algo="asgd",
+                          sync=True, dtype="fp32", sparse=False,
+                          engine="threads", device="cpu", workers=2),
+    "asgd-mnist8m": dict(rows=8_100_000, cols=784, rate=0.01, algo="asgd",
+                         sync=False, dtype="bf16", sparse=False,
+                         engine="graph"),
+    "asaga-rcv1": dict(rows=697_641, cols=47_236, rate=0.02, algo="asaga",
+                       sync=False, dtype="fp32", sparse=True,
+                       engine="graph"),
+    "asgd-epsilon-delay": dict(rows=400_000, cols=2_000, rate=0.01,
+                               algo="asgd", sync=False, dtype="fp32",
+                               sparse=False, engine="threads",
+                               delay_coeff=1.0, workers=8),
+    "asaga-mnist8m-hostspill": dict(rows=8_100_000, cols=784, rate=0.01,
+                                    algo="asaga", sync=False, dtype="bf16",
+                                    sparse=False, engine="threads",
+                                    history="host"),
+}
 
 
 def parse_args():
     p = argparse.ArgumentParser()
+    p.add_argument("--model", default="asgd-mnist8m", choices=list(MODELS))
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=500)
-    p.add_argument("--rows", type=int, default=BASE["rows"])
-    p.add_argument("--cols", type=int, default=BASE["cols"])
-    p.add_argument("--rate", type=float, default=BASE["rate"])
-    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--rows", type=int, default=0)
+    p.add_argument("--cols", type=int, default=0)
+    p.add_argument("--rate", type=float, default=0.0)
+    p.add_argument("--dtype", default="", choices=["", "bf16", "fp32"])
     p.add_argument("--objective", default="lsq", choices=["lsq", "logistic"])
-    p.add_argument("--algo", default="asgd", choices=["asgd", "asaga"])
+    p.add_argument("--algo", default="", choices=["", "asgd", "asaga"])
     p.add_argument("--device", default=None, help="override (cpu for debug)")
-    p.add_argument("--engine", default="graph", choices=["graph", "threads"],
-                   help="N=1 GPU path: hipGraph device loop (default) or the "
-                        "threaded mailbox engine")
-    return p.parse_args()
+    p.add_argument("--engine", default="", choices=["", "graph", "threads"])
+    args = p.parse_args()
+    preset = MODELS[args.model]
+    args.rows = args.rows or preset["rows"]
+    args.cols = args.cols or preset["cols"]
+    args.rate = args.rate or preset["rate"]
+    args.dtype = args.dtype or preset["dtype"]
+    args.algo = args.algo or preset["algo"]
+    args.engine = args.engine or preset["engine"]
+    if args.device is None and "device" in preset:
+        args.device = preset["device"]
+    args.sparse = preset.get("sparse", False)
+    args.sync = preset.get("sync", False)
+    args.delay_coeff = preset.get("delay_coeff", 0.0)
+    args.history = preset.get("history", "device")
+    args.preset_workers = preset.get("workers", 0)
+    return args
 
 
-def main():
-    args = parse_args()
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    rank = int(os.environ.get("RANK", "0"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    n_gpus = max(args.gpus, world)
-
-    if args.device:
-        device = torch.device(args.device)
-    elif torch.cuda.is_available():
-        device = torch.device(f"cuda:{local_rank}")
-        torch.cuda.set_device(device)
-    else:
-        device = torch.device("cpu")
-
-    if device.type == "cuda":
-        # the native kernels are mandatory on GPU (ops dispatch enforces it)
-        import asyncframework_amd.ops as ops
-        assert ops.hip_available(), "HIP extension missing on a GPU box"
-
-    cfg = EngineConfig(
-        d=args.cols, N=args.rows, num_workers=world,
+def make_cfg(args, num_workers, device):
+    return EngineConfig(
+        d=args.cols, N=args.rows, num_workers=num_workers,
         num_iterations=args.warmup + args.steps + 1,
         gamma=BASE["gamma"], taw=BASE["taw"], batch_rate=args.rate,
         bucket_ratio=BASE["bucket_ratio"], printer_freq=1 << 30,
-        delay_coeff=0.0, seed=BASE["seed"], algo=args.algo, sync=False,
-        objective=args.objective, dtype=args.dtype,
-        device=str(device), snapshot_weights=False)
-
-    dtype = cfg.torch_dtype()
-    shards = row_shards(args.rows, world)
-    s, t = shards[rank]
-    # every rank generates only ITS shard (seeded per-rank so ranks differ)
-    X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + rank,
-                           dtype=dtype, device=device,
-                           objective=args.objective)
-    worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X, y=y), cfg,
-                    device=device)
-
-    mark_lo, mark_hi = args.warmup, args.warmup + args.steps
-
-    if world > 1:
-        import torch.distributed as dist
-        from asyncframework_amd.engine.dist import DistEngine
-        dist.init_process_group(
-            "nccl" if device.type == "cuda" else "gloo",
-            rank=rank, world_size=world)
-        dist.barrier()
-        if device.type == "cuda":
-            torch.cuda.synchronize()
-        deng = DistEngine(cfg, worker, device)
-        if rank == 0:
-            server = Server(cfg, device=device)
-            from asyncframework_amd.engine.local import _LocalChannel
-            channels = [_LocalChannel(worker, server)]
-            from asyncframework_amd.engine.dist import _RemoteChannel
-            for i in range(1, world):
-                channels.append(_RemoteChannel(i, deng.pair_groups[i],
-                                               server, cfg, device))
-            eng = AsyncEngine(cfg, server=server, channels=channels)
-            eng.verbose = False
-            eng.mark_at = {mark_lo, mark_hi}
-            res = eng.run(max_wall_s=1800)
-            dist.barrier()
-            if device.type == "cuda":
-                torch.cuda.synchronize()
-            emit(args, cfg, eng, n_gpus=world)
-        else:
-            from asyncframework_amd.engine.dist import remote_worker_loop
-            remote_worker_loop(worker, cfg, deng.pair_groups[rank], device)
-            dist.barrier()
-            if device.type == "cuda":
-                torch.cuda.synchronize()
-        dist.destroy_process_group()
-    elif device.type == "cuda" and args.engine == "graph":
-        # GPU-resident hipGraph round loop (engine/graph.py): the whole
-        # [sample+grad, fused update] round replayed from a captured graph.
-        from asyncframework_amd.engine.graph import GraphEngine
-        geng = GraphEngine(cfg, worker.shard, device)
-        torch.cuda.synchronize()
-        t0, t1 = geng.bench(args.warmup, args.steps)
-        emit_elapsed(args, cfg, t1 - t0, n_gpus=1)
-    else:
-        if device.type == "cuda":
-            torch.cuda.synchronize()
-        eng = AsyncEngine(cfg, workers=[worker])
-        eng.verbose = False
-        eng.mark_at = {mark_lo, mark_hi}
-        res = eng.run(max_wall_s=1800)
-        if device.type == "cuda":
-            torch.cuda.synchronize()
-        emit(args, cfg, eng, n_gpus=1)
+        delay_coeff=args.delay_coeff, seed=BASE["seed"], algo=args.algo,
+        sync=args.sync, objective=args.objective, dtype=args.dtype,
+        device=str(device), history_placement=args.history,
+        calib_factor=10 if args.delay_coeff else 100,
+        snapshot_weights=False)
 
 
-def emit(args, cfg: EngineConfig, eng: AsyncEngine, n_gpus: int):
-    lo, hi = args.warmup, args.warmup + args.steps
-    t0 = eng.marks.get(lo)
-    t1 = eng.marks.get(hi)
-    if t0 is None or t1 is None or t1 <= t0:
-        print(json.dumps({"error": "marks missing", "marks":
-                          {str(k): v for k, v in eng.marks.items()}}))
-        sys.exit(1)
-    emit_elapsed(args, cfg, t1 - t0, n_gpus)
-
-
-def emit_elapsed(args, cfg: EngineConfig, elapsed: float, n_gpus: int):
+def emit(args, cfg, elapsed: float, n_gpus: int):
     ups = args.steps / elapsed
     out = {
         "metric": "gradient updates/sec (whole node)",
@@ -178,16 +120,139 @@ def emit_elapsed(args, cfg: EngineConfig, elapsed: float, n_gpus: int):
         "dtype": args.dtype,
         "data": "synthetic",
         "config": {
-            "model": f"{args.algo}-{args.objective}-mnist8m",
+            "model": args.model,
+            "algo": args.algo, "objective": args.objective,
             "rows": args.rows, "cols": args.cols,
             "batch_rate": args.rate, "taw": cfg.taw,
             "gamma": cfg.gamma, "bucket_ratio": cfg.bucket_ratio,
             "global_batch": int(args.rate * args.rows),
             "seq_len": args.cols,
-            "parallelism": f"async-ps-dp{n_gpus}",
+            "parallelism": ("sync" if args.sync else "async-ps") +
+                           f"-dp{n_gpus}-w{cfg.num_workers}",
         },
     }
     print(json.dumps(out))
+
+
+def run_single(args, device):
+    """One process (N=1): graph engine (device-resident loop) or the
+    threaded mailbox engine (delay injection / host-spill / sync / CPU)."""
+    n_workers = args.preset_workers or 1
+    if args.engine == "graph":
+        n_workers = 1
+    cfg = make_cfg(args, n_workers, device)
+    dt = cfg.torch_dtype()
+    if args.sparse:
+        data = synthetic_csr(args.rows, args.cols, seed=BASE["seed"],
+                             device=device, dtype=dt)
+    else:
+        data = synthetic_dense(args.rows, args.cols, seed=BASE["seed"],
+                               dtype=dt, device=device,
+                               objective=args.objective)
+
+    if args.engine == "graph" and device.type == "cuda":
+        from asyncframework_amd.engine.graph import GraphEngine
+        if args.sparse:
+            sh = Shard(row_start=0, n_rows=args.rows, indptr=data[0],
+                       indices=data[1], values=data[2], y=data[3])
+        else:
+            sh = Shard(row_start=0, n_rows=args.rows, X=data[0], y=data[1])
+        eng = GraphEngine(cfg, sh, device)
+        torch.cuda.synchronize()
+        t0, t1 = eng.bench(args.warmup, args.steps)
+        emit(args, cfg, t1 - t0, n_gpus=1)
+        return
+
+    if args.sparse:
+        workers = runner.build_csr_workers(cfg, *data)
+    else:
+        workers = runner.build_dense_workers(cfg, *data)
+    server = Server(cfg, device=device)
+    eng_cls = SyncEngine if cfg.sync else AsyncEngine
+    eng = eng_cls(cfg, workers=workers, server=server)
+    eng.verbose = False
+    eng.mark_at = {args.warmup, args.warmup + args.steps}
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    eng.run(max_wall_s=1800)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = eng.marks.get(args.warmup)
+    t1 = eng.marks.get(args.warmup + args.steps)
+    if t0 is None or t1 is None or t1 <= t0:
+        print(json.dumps({"error": "marks missing"}))
+        sys.exit(1)
+    emit(args, cfg, t1 - t0, n_gpus=1)
+
+
+def run_dist(args, device, rank, world):
+    import torch.distributed as dist
+    from asyncframework_amd.engine.dist import (DistEngine, _RemoteChannel,
+                                                remote_worker_loop)
+    from asyncframework_amd.engine.local import _LocalChannel
+    cfg = make_cfg(args, world, device)
+    dt = cfg.torch_dtype()
+    s, t = row_shards(args.rows, world)[rank]
+    X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + rank,
+                           dtype=dt, device=device,
+                           objective=args.objective)
+    worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X, y=y), cfg,
+                    device=device)
+    dist.init_process_group("nccl" if device.type == "cuda" else "gloo",
+                            rank=rank, world_size=world)
+    dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    deng = DistEngine(cfg, worker, device)
+    if rank == 0:
+        server = Server(cfg, device=device)
+        channels = [_LocalChannel(worker, server)]
+        for i in range(1, world):
+            channels.append(_RemoteChannel(i, deng.pair_groups[i], server,
+                                           cfg, device))
+        eng = AsyncEngine(cfg, server=server, channels=channels)
+        eng.verbose = False
+        eng.mark_at = {args.warmup, args.warmup + args.steps}
+        eng.run(max_wall_s=1800)
+        dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        t0 = eng.marks.get(args.warmup)
+        t1 = eng.marks.get(args.warmup + args.steps)
+        if t0 is None or t1 is None or t1 <= t0:
+            print(json.dumps({"error": "marks missing"}))
+        else:
+            emit(args, cfg, t1 - t0, n_gpus=world)
+    else:
+        remote_worker_loop(worker, cfg, deng.pair_groups[rank], device)
+        dist.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+    dist.destroy_process_group()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device(f"cuda:{local_rank}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    if device.type == "cuda":
+        import asyncframework_amd.ops as ops
+        assert ops.hip_available(), "HIP extension missing on a GPU box"
+
+    if world > 1:
+        run_dist(args, device, rank, world)
+    else:
+        run_single(args, device)
 
 
 if __name__ == "__main__":
