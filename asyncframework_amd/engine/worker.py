@@ -67,6 +67,9 @@ class Worker:
         if self.cfg.history_placement == "host" and self.is_cuda:
             self.alpha = torch.zeros(n, dtype=torch.float32,
                                      device="cpu").pin_memory()
+            # staging buffer for the per-round pinned-host -> HBM gather
+            self._alpha_dev_buf = torch.zeros(n, dtype=torch.float32,
+                                              device=self.device)
         else:
             self.alpha = torch.zeros(n, dtype=torch.float32, device=self.device)
 
@@ -124,10 +127,12 @@ class Worker:
         alpha = self._alpha_device()
         host_spill = alpha.device.type == "cpu" and self.is_cuda
         if host_spill:
-            # Spill path: gather the shard's history to device for the round
-            # via pinned-host async copy (BASELINE config 5). The kernel then
-            # sees a device-resident table; commits scatter back.
-            alpha_dev = alpha.to(self.device, non_blocking=True)
+            # Spill path: stage the shard's history to HBM for the round via
+            # pinned-host hipMemcpyAsync into a preallocated buffer
+            # (BASELINE config 5). The kernel sees a device-resident table;
+            # accepted commits scatter back to the pinned table.
+            self._alpha_dev_buf.copy_(alpha, non_blocking=True)
+            alpha_dev = self._alpha_dev_buf
         else:
             alpha_dev = alpha
         if sh.is_sparse:

@@ -77,7 +77,8 @@ def saga_grad_dense(X, y, w, alpha, g, seed, round_k, row_start, rate, obj
     _chk(g, "g", torch.float32)
     w = w.float().contiguous()
     n_rows, d = X.shape
-    cap = n_rows
+    cap = n_rows if rate >= 1.0 else min(n_rows,
+                                         int(rate * n_rows * 2) + 4096)
     idx = torch.empty(cap, dtype=torch.int32, device=X.device)
     e = torch.empty(cap, dtype=torch.float32, device=X.device)
     ctr = torch.zeros(2, dtype=torch.int32, device=X.device)  # [n, pos]
@@ -98,8 +99,10 @@ def saga_grad_csr(indptr, indices, values, y, w, alpha, g, seed, round_k,
     _chk(g, "g", torch.float32)
     w = w.float().contiguous()
     n_rows = indptr.shape[0] - 1
-    idx = torch.empty(n_rows, dtype=torch.int32, device=w.device)
-    e = torch.empty(n_rows, dtype=torch.float32, device=w.device)
+    cap = n_rows if rate >= 1.0 else min(n_rows,
+                                         int(rate * n_rows * 2) + 4096)
+    idx = torch.empty(cap, dtype=torch.int32, device=w.device)
+    e = torch.empty(cap, dtype=torch.float32, device=w.device)
     ctr = torch.zeros(2, dtype=torch.int32, device=w.device)
     _hip_core.saga_grad_csr(indptr.data_ptr(), indices.data_ptr(),
                             values.data_ptr(), y.data_ptr(), w.data_ptr(),

@@ -55,7 +55,7 @@ MODELS = {
     "asaga-mnist8m-hostspill": dict(rows=8_100_000, cols=784, rate=0.01,
                                     algo="asaga", sync=False, dtype="bf16",
                                     sparse=False, engine="threads",
-                                    history="host"),
+                                    history="host", workers=8),
 }
 
 
