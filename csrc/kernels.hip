@@ -208,24 +208,80 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
 
 // ------------------------------------------------- K1 pipelined (queue)
 //
-// Two-phase dense gradient: (1) Philox scan appends sampled local row ids to
-// an LDS queue (block-shared, atomic append; the near-full guard makes
-// overflow impossible), (2) waves drain the queue round-robin with a
-// DEPTH-deep software pipeline (statically unrolled — runtime-indexed
-// register arrays go to scratch, CDNA guide rule 20): while row q computes,
-// DEPTH-1 later rows' loads are in flight, and the accumulate pass reuses
-// the registers instead of re-reading X. This is the latency fix: at b=0.01
-// a wave sees ~1 sampled row per 256-row scan group, so the non-queued
-// kernel paid a full HBM latency per row (measured ~7 us serial per row).
+// Two-phase dense gradient. Phase 1 (scan): Philox decides 4 rows per eval;
+// sampled rows are appended to an LDS queue TOGETHER with their y (and SAGA
+// alpha) values — prefetching them here keeps ordinary vmem loads out of
+// the pipelined phase (hipcc drains vmcnt(0) at any ordinary-load use,
+// CDNA guide §5.5 trap (b): measured one full drain per row from y[rr]).
+// Phase 2 (process): waves drain the queue round-robin with a DEPTH-deep
+// statically-unrolled register pipeline; row loads are UNCONDITIONAL
+// bounds-checked buffer loads (per-row scalar descriptor) with a
+// compile-time ITERS count — a per-element `if (j4 < d4)` made hipcc branch
+// around every load with a vmcnt(0) (guide trap 4(c)). The accumulate pass
+// reuses the pipeline registers; X is read exactly once per sampled row.
 
-// queue capacity must exceed rows-per-scan-iteration (the near-full guard
-// is `qn >= QCAP - RPB`; QCAP == RPB would never scan and livelock)
-#define QCAP_FOR(PB) ((PB) == 512 ? 4096 : 2048)
-#define PIPE_MAXIT 8  // supports d <= 2048, d % 4 == 0
+#define QCAP 2048
+
+// Inline-asm row loads + hand-counted waits. hipcc's own scheduler defeats
+// the software pipeline (it emitted a vmcnt(0) drain before each dot and a
+// wait immediately after each refill — measured); per the CDNA guide the
+// fix is asm loads whose destinations are unprotected until OUR counted
+// s_waitcnt. Consumers are pinned below the wait by zero-cost empty-asm
+// register ties on each 32-bit component (tied vector operands are not
+// supported by the backend).
+struct B16x4 { union { ushort4 s; uint2 u; }; };
+struct F32x4 { union { float4 f; uint4 u; }; };
 
 template <typename XT> struct RowVec;
-template <> struct RowVec<float> { using T = float4; };
-template <> struct RowVec<__hip_bfloat16> { using T = ushort4; };
+template <> struct RowVec<float> {
+  using T = F32x4;
+  static __device__ __forceinline__ T load(__amdgpu_buffer_rsrc_t rsrc,
+                                           int voff_bytes) {
+    T out;
+    auto r = __builtin_amdgcn_raw_buffer_load_b128(rsrc, voff_bytes, 0, 0);
+    union { decltype(r) u; float4 f; } c{r};
+    out.f = c.f;
+    return out;
+  }
+  static constexpr int VOFF_SHIFT = 4;  // 16 B per lane-element
+  static __device__ __forceinline__ void tie(T& x) {
+    asm volatile("" : "+v"(x.u.x));
+    asm volatile("" : "+v"(x.u.y));
+    asm volatile("" : "+v"(x.u.z));
+    asm volatile("" : "+v"(x.u.w));
+  }
+};
+template <> struct RowVec<__hip_bfloat16> {
+  using T = B16x4;
+  static __device__ __forceinline__ T load(__amdgpu_buffer_rsrc_t rsrc,
+                                           int voff_bytes) {
+    T out;
+    auto r = __builtin_amdgcn_raw_buffer_load_b64(rsrc, voff_bytes, 0, 0);
+    union { decltype(r) u; ushort4 s; } c{r};
+    out.s = c.s;
+    return out;
+  }
+  static constexpr int VOFF_SHIFT = 3;  // 8 B per lane-element
+  static __device__ __forceinline__ void tie(T& x) {
+    asm volatile("" : "+v"(x.u.x));
+    asm volatile("" : "+v"(x.u.y));
+  }
+};
+
+template <int V>
+__device__ __forceinline__ void wait_vmcnt_imm() {
+  asm volatile("s_waitcnt vmcnt(%[v])" ::[v] "i"(V));
+}
+
+// Counted wait + pin this buffer's consumers below it. V is conservative-
+// safe at queue tails (fewer outstanding loads than the bound always
+// passes; FIFO vmcnt semantics drain everything older than the bound).
+template <int V, typename RV, typename T, int N>
+__device__ __forceinline__ void wait_and_tie(T (&b)[N]) {
+  wait_vmcnt_imm<V>();
+#pragma unroll
+  for (int i = 0; i < N; ++i) RV::tie(b[i]);
+}
 
 __device__ __forceinline__ void cvt4(const float4& r, float o[4]) {
   o[0] = r.x; o[1] = r.y; o[2] = r.z; o[3] = r.w;
@@ -236,20 +292,29 @@ __device__ __forceinline__ void cvt4(const ushort4& r, float o[4]) {
   o[0] = __bfloat162float(c0.b); o[1] = __bfloat162float(c1.b);
   o[2] = __bfloat162float(c2.b); o[3] = __bfloat162float(c3.b);
 }
-
-template <typename XT>
-__device__ __forceinline__ void load_row_regs(
-    const XT* __restrict__ xrow, int lane, int d4,
-    typename RowVec<XT>::T raw[PIPE_MAXIT]) {
-#pragma unroll
-  for (int it = 0; it < PIPE_MAXIT; ++it) {
-    const int j4 = lane + it * WAVE;
-    if (j4 < d4)
-      raw[it] = reinterpret_cast<const typename RowVec<XT>::T*>(xrow)[j4];
-  }
+__device__ __forceinline__ void cvt4(const F32x4& r, float o[4]) {
+  cvt4(r.f, o);
+}
+__device__ __forceinline__ void cvt4(const B16x4& r, float o[4]) {
+  cvt4(r.s, o);
 }
 
-template <typename XT, bool SAGA, int PBLOCK, int DEPTH>
+// Per-row buffer descriptor: base = row start (scalar via readfirstlane),
+// num_records = bytes to the end of X -> hardware bounds check returns 0 for
+// the padded over-read of the last row (w_lds is zero-padded, so padded
+// elements contribute nothing).
+template <typename XT>
+__device__ __forceinline__ __amdgpu_buffer_rsrc_t row_rsrc(
+    const XT* X, long total_elems, int rr, int d) {
+  const size_t off = (size_t)rr * d;
+  const uint64_t rem_bytes = (uint64_t)(total_elems - off) * sizeof(XT);
+  const uint32_t nrec =
+      rem_bytes > 0xFFFFFFF0ull ? 0xFFFFFFF0u : (uint32_t)rem_bytes;
+  return __builtin_amdgcn_make_buffer_rsrc((void*)(X + off), 0, nrec,
+                                           0x00020000);
+}
+
+template <typename XT, bool SAGA, int PBLOCK, int DEPTH, int ITERS>
 __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     const XT* __restrict__ X, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
@@ -260,34 +325,37 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
     int take_all, int objective) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
-  constexpr int NW = PBLOCK / WAVE;            // waves per block
-  constexpr int RPB = NW * ROWS_PER_WAVE;      // rows scanned per block-iter
-  extern __shared__ float smem[];
-  float* w_lds = smem;                          // [d]
-  float* gacc = smem + d;                       // [NW][d]
-  constexpr int QCAP = QCAP_FOR(PBLOCK);
+  constexpr int NW = PBLOCK / WAVE;
+  constexpr int RPB = NW * ROWS_PER_WAVE;
+  constexpr int DPAD = ITERS * 256;  // padded feature dim
   static_assert(QCAP >= RPB + 1024, "queue must out-size one scan iter");
-  int* rowq = (int*)(smem + (size_t)(1 + NW) * d);  // [QCAP]
+  extern __shared__ float smem[];
+  float* w_lds = smem;                               // [DPAD]
+  float* gacc = smem + DPAD;                         // [NW][DPAD]
+  float* yq = smem + (size_t)(1 + NW) * DPAD;        // [QCAP]
+  float* aq = yq + QCAP;                             // [QCAP] (SAGA)
+  float* eq = aq + (SAGA ? QCAP : 0);                // [QCAP] (SAGA)
+  int* rowq = (int*)(eq + (SAGA ? QCAP : 0));        // [QCAP]
   int* qn = rowq + QCAP;
-  for (int j = threadIdx.x; j < d; j += PBLOCK) {
-    w_lds[j] = w[j];
+  for (int j = threadIdx.x; j < DPAD; j += PBLOCK) {
+    w_lds[j] = j < d ? w[j] : 0.f;
 #pragma unroll
-    for (int s2 = 0; s2 < NW; ++s2) gacc[(size_t)s2 * d + j] = 0.f;
+    for (int s2 = 0; s2 < NW; ++s2) gacc[(size_t)s2 * DPAD + j] = 0.f;
   }
   if (threadIdx.x == 0) *qn = 0;
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  float* gw = gacc + (size_t)wave * d;
+  float* gw = gacc + (size_t)wave * DPAD;
   int local_count = 0;
-  const int d4 = d >> 2;
-  using RV = typename RowVec<XT>::T;
+  const long total_elems = n_rows * (long)d;
+  using RV = RowVec<XT>;
 
   const long ngroups = (n_rows + RPB - 1) / RPB;
   long gi = blockIdx.x;
   bool done = false;
   while (!done) {
-    // ---- scan phase: fill the queue until near-full or rows exhausted
+    // ---- scan phase (also prefetches y / alpha into the queue)
     while (true) {
       __syncthreads();
       if (gi >= ngroups || *qn >= QCAP - RPB) break;
@@ -302,7 +370,10 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
         for (int i = 0; i < 4; ++i) {
           if (4L * lane + i < rem && (take_all || xs[i] < threshold)) {
             const int pos = atomicAdd(qn, 1);
-            rowq[pos] = (int)(lrow + i);
+            const int row = (int)(lrow + i);
+            rowq[pos] = row;
+            yq[pos] = y[row];
+            if (SAGA) aq[pos] = alpha[row];
           }
         }
       }
@@ -310,12 +381,22 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     }
     __syncthreads();
     const int nq = min(*qn, QCAP);
-    // ---- process phase: wave-strided, DEPTH-deep pipeline (static bufs)
-    RV buf[DEPTH][PIPE_MAXIT];
+    // ---- process phase: DEPTH-deep static pipeline, buffer loads only
+    typename RV::T buf[DEPTH][ITERS];
+    // the counted-wait protocol needs EXACTLY ITERS loads per slot: a
+    // skipped load would let vmcnt(N) pass while this buffer's own loads
+    // are still in flight (vmcnt waits "<= N outstanding", nothing more).
+    // Tail slots load a clamped row instead of skipping.
+    if (nq > 0) {
 #pragma unroll
-    for (int p = 0; p < DEPTH; ++p) {
-      const int q = wave + p * NW;
-      if (q < nq) load_row_regs<XT>(X + (size_t)rowq[q] * d, lane, d4, buf[p]);
+      for (int p = 0; p < DEPTH; ++p) {
+        const int q = min(wave + p * NW, nq - 1);
+        const int rr = __builtin_amdgcn_readfirstlane(rowq[q]);
+        const auto rs = row_rsrc<XT>(X, total_elems, rr, d);
+#pragma unroll
+        for (int it = 0; it < ITERS; ++it)
+          buf[p][it] = RV::load(rs, (lane + it * WAVE) << RV::VOFF_SHIFT);
+      }
     }
     int q_base = wave;
     while (q_base < nq) {
@@ -323,61 +404,72 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
       for (int p = 0; p < DEPTH; ++p) {
         const int q = q_base + p * NW;
         if (q < nq) {
-          const int rr = rowq[q];
           float z = 0.f;
 #pragma unroll
-          for (int it = 0; it < PIPE_MAXIT; ++it) {
+          for (int it = 0; it < ITERS; ++it) {
+            float o[4];
+            cvt4(buf[p][it], o);
             const int j4 = lane + it * WAVE;
-            if (j4 < d4) {
-              float o[4];
-              cvt4(buf[p][it], o);
-              // b128 LDS read: 16-lane groups hit distinct banks (b32 had
-              // lanes l and l+8 colliding -> 26% LDSBankConflict measured)
-              const float4 wv = reinterpret_cast<const float4*>(w_lds)[j4];
-              z += o[0] * wv.x + o[1] * wv.y + o[2] * wv.z + o[3] * wv.w;
-            }
+            const float4 wv = reinterpret_cast<const float4*>(w_lds)[j4];
+            z += o[0] * wv.x + o[1] * wv.y + o[2] * wv.z + o[3] * wv.w;
           }
 #pragma unroll
           for (int off = 32; off > 0; off >>= 1)
             z += __shfl_xor(z, off, WAVE);
-          float e = link_residual(z, y[rr], objective);
+          float e = link_residual(z, yq[q], objective);
           float coeff = e;
           if (SAGA) {
-            const float a_old = alpha[rr];
-            coeff = e - a_old;
-            if (lane == 0) {
-              if (commit_now) {
-                alpha[rr] = e;
-              } else {
-                const int pos = atomicAdd(pos_ctr, 1);
-                idx_out[pos] = rr;
-                e_out[pos] = e;
-              }
-            }
+            coeff = e - aq[q];
+            if (lane == 0) eq[q] = e;  // committed after the barrier
           }
           ++local_count;
 #pragma unroll
-          for (int it = 0; it < PIPE_MAXIT; ++it) {
+          for (int it = 0; it < ITERS; ++it) {
+            float o[4];
+            cvt4(buf[p][it], o);
             const int j4 = lane + it * WAVE;
-            if (j4 < d4) {
-              float o[4];
-              cvt4(buf[p][it], o);
-              float4* gw4 = reinterpret_cast<float4*>(gw);
-              float4 cur = gw4[j4];
-              cur.x += coeff * o[0]; cur.y += coeff * o[1];
-              cur.z += coeff * o[2]; cur.w += coeff * o[3];
-              gw4[j4] = cur;
-            }
+            float4* gw4 = reinterpret_cast<float4*>(gw);
+            float4 cur = gw4[j4];
+            cur.x += coeff * o[0]; cur.y += coeff * o[1];
+            cur.z += coeff * o[2]; cur.w += coeff * o[3];
+            gw4[j4] = cur;
           }
-          // refill this buffer DEPTH rows ahead
-          const int qf = q + DEPTH * NW;
-          if (qf < nq)
-            load_row_regs<XT>(X + (size_t)rowq[qf] * d, lane, d4, buf[p]);
+          // refill this buffer DEPTH rows ahead (clamped: see prologue).
+          // sched_barrier pins every read of buf[p] (the accumulate above)
+          // BEFORE the async refill overwrites it — the compiler treats an
+          // asm load's register write as instantaneous and would otherwise
+          // rotate reads past the issue (observed: regalloc preservation
+          // copies racing the in-flight load).
+          __builtin_amdgcn_sched_barrier(0);
+          const int qf = min(q + DEPTH * NW, nq - 1);
+          {
+            const int rr2 = __builtin_amdgcn_readfirstlane(rowq[qf]);
+            const auto rs2 = row_rsrc<XT>(X, total_elems, rr2, d);
+#pragma unroll
+            for (int it = 0; it < ITERS; ++it)
+              buf[p][it] = RV::load(rs2, (lane + it * WAVE) << RV::VOFF_SHIFT);
+          }
+          __builtin_amdgcn_sched_barrier(0);
         }
       }
       q_base += DEPTH * NW;
     }
     __syncthreads();
+    if (SAGA) {
+      // history scalar emit/commit, batched outside the pipeline
+      for (int pos = threadIdx.x; pos < nq; pos += PBLOCK) {
+        const int row = rowq[pos];
+        const float e = eq[pos];
+        if (commit_now) {
+          alpha[row] = e;
+        } else {
+          const int gpos = atomicAdd(pos_ctr, 1);
+          idx_out[gpos] = row;
+          e_out[gpos] = e;
+        }
+      }
+      __syncthreads();
+    }
     if (threadIdx.x == 0) *qn = 0;
     done = gi >= ngroups;
   }
@@ -387,14 +479,14 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     for (int j = threadIdx.x; j < d; j += PBLOCK) {
       float s = 0.f;
 #pragma unroll
-      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * d + j];
+      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * DPAD + j];
       g_part[(size_t)j * G + blockIdx.x] = s;
     }
   } else {
     for (int j = threadIdx.x; j < d; j += PBLOCK) {
       float s = 0.f;
 #pragma unroll
-      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * d + j];
+      for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * DPAD + j];
       if (s != 0.f) atomicAdd(&g_out[j], s);
     }
   }
@@ -599,24 +691,38 @@ static void launch_dense(const XT* X, const float* y, const float* w,
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
   const char* np = std::getenv("ASYNCAMD_NO_PIPE");
-  const bool pipe_ok = (d % 4 == 0) && (d <= 4 * WAVE * PIPE_MAXIT) &&
-                       !(np && np[0] == '1');
+  const bool pipe_ok = (d % 4 == 0) && (d <= 2048) && !(np && np[0] == '1');
   if (pipe_ok) {
-    const char* pb = std::getenv("ASYNCAMD_PBLOCK");
-    const int pblock = pb ? std::atoi(pb) : 256;  // measured best
-#define LAUNCH_PIPE(PB, DP)                                                  \
+    const int iters = (d + 255) / 256;
+    const size_t smem = (size_t)(1 + 4) * (iters * 256) * sizeof(float) +
+                        (size_t)QCAP * sizeof(float) * (SAGA ? 3 : 1) +
+                        (QCAP + 1) * sizeof(int);
+    const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
+    const int depth = dps ? std::atoi(dps) : 4;
+#define LAUNCH_PIPE_D(IT, DP)                                                \
+    hipLaunchKernelGGL((grad_dense_pipe_kernel<XT, SAGA, 256, DP, IT>),      \
+                       dim3(grid), dim3(256), smem, stream, X, y, w, g_out,  \
+                       g_part, n_out, alpha, idx_out, e_out, pos_ctr,        \
+                       k_dev, commit_now, n_rows, d, seed, round_k,          \
+                       row_start, thr, take_all, objective)
+#define LAUNCH_PIPE(IT)                                                      \
     do {                                                                     \
-      const size_t smem = (size_t)(1 + PB / WAVE) * d * sizeof(float) +      \
-                          (QCAP_FOR(PB) + 1) * sizeof(int);                  \
-      hipLaunchKernelGGL((grad_dense_pipe_kernel<XT, SAGA, PB, DP>),         \
-                         dim3(grid), dim3(PB), smem, stream, X, y, w, g_out, \
-                         g_part, n_out, alpha, idx_out, e_out, pos_ctr,      \
-                         k_dev, commit_now, n_rows, d, seed, round_k,        \
-                         row_start, thr, take_all, objective);               \
+      if (depth == 1) LAUNCH_PIPE_D(IT, 1);                                  \
+      else if (depth == 2) LAUNCH_PIPE_D(IT, 2);                             \
+      else LAUNCH_PIPE_D(IT, 4);                                             \
     } while (0)
-    if (pblock == 256) LAUNCH_PIPE(256, 4);
-    else LAUNCH_PIPE(512, 4);
+    switch (iters) {
+      case 1: LAUNCH_PIPE(1); break;
+      case 2: LAUNCH_PIPE(2); break;
+      case 3: LAUNCH_PIPE(3); break;
+      case 4: LAUNCH_PIPE(4); break;
+      case 5: LAUNCH_PIPE(5); break;
+      case 6: LAUNCH_PIPE(6); break;
+      case 7: LAUNCH_PIPE(7); break;
+      default: LAUNCH_PIPE(8); break;
+    }
 #undef LAUNCH_PIPE
+#undef LAUNCH_PIPE_D
     return;
   }
   const int lpr = pick_lpr(d);
