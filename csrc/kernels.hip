@@ -709,6 +709,8 @@ static void launch_dense(const XT* X, const float* y, const float* w,
     do {                                                                     \
       if (depth == 1) LAUNCH_PIPE_D(IT, 1);                                  \
       else if (depth == 2) LAUNCH_PIPE_D(IT, 2);                             \
+      else if (depth == 6) LAUNCH_PIPE_D(IT, 6);                             \
+      else if (depth == 8) LAUNCH_PIPE_D(IT, 8);                             \
       else LAUNCH_PIPE_D(IT, 4);                                             \
     } while (0)
     switch (iters) {
