@@ -167,6 +167,23 @@ class ASYNCcontext(Generic[T]):
     def hasNext(self) -> bool:
         return not self.ResultList.empty()
 
+    # -- future-returning collect (analog of Spark's AsyncRDDActions /
+    #    FutureAction pattern, reference rdd/AsyncRDDActions.scala:33-137) ---
+    def ASYNCcollectAsync(self) -> "futures.Future[RDDPartialRes[T]]":
+        """Non-blocking collect: returns a Future fulfilled with the next
+        mailbox entry."""
+        from concurrent import futures
+        fut: "futures.Future[RDDPartialRes[T]]" = futures.Future()
+
+        def _wait():
+            try:
+                fut.set_result(self.ResultList.get())
+            except Exception as e:  # pragma: no cover
+                fut.set_exception(e)
+
+        threading.Thread(target=_wait, daemon=True).start()
+        return fut
+
     # -- engine-side producer (analog of the mergeResult trampolines,
     #    reference ASYNCcontext.scala:77-80) ---------------------------------
     def put(self, res: RDDPartialRes[T]) -> None:

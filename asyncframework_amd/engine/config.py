@@ -41,6 +41,9 @@ class EngineConfig:
     snapshot_weights: bool = True   # record optVars (time, w) for loss curves
     checkpoint_path: str = ""       # periodic optimizer-state snapshots
     checkpoint_every: int = 0       # every N applied updates (0 = off)
+    worker_timeout_s: float = 0.0   # busy-worker timeout -> declared dead
+                                    # (reference has none: a lost task left a
+                                    # worker busy forever)
 
     def torch_dtype(self) -> torch.dtype:
         return _DTYPES[self.dtype]

@@ -100,6 +100,7 @@ class AsyncEngine:
         self.channels = channels
         assert len(self.channels) == cfg.num_workers
         self.pending: "deque[int]" = deque(range(cfg.num_workers))
+        self.dead: set = set()   # workers declared lost (worker_timeout_s)
         self._stop = threading.Event()
         self._pending_ev = threading.Event()
         self._pending_ev.set()
@@ -114,6 +115,24 @@ class AsyncEngine:
 
     def _local_workers(self):
         return [ch.worker for ch in self.channels if hasattr(ch, "worker")]
+
+    def _reap_dead_workers(self):
+        """Failure detection the reference lacks (a lost task
+        leaves a worker permanently busy): a worker whose round exceeds
+        worker_timeout_s is declared dead and excluded from the quorum gate;
+        if its result arrives later it is resurrected."""
+        srv, cfg = self.server, self.cfg
+        now = time.perf_counter()
+        for wid in range(cfg.num_workers):
+            if wid in self.dead:
+                if srv.AC.STAT[wid].getAvailability():
+                    self.dead.discard(wid)  # came back after all
+                continue
+            st = srv.AC.STAT[wid]
+            sub = srv.submit_time.get(wid)
+            if (not st.getAvailability() and sub is not None
+                    and now - sub > cfg.worker_timeout_s):
+                self.dead.add(wid)
 
     # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
     def _updater(self):
@@ -187,9 +206,13 @@ class AsyncEngine:
             if max_wall_s and time.perf_counter() - t_start > max_wall_s:
                 self._stop.set()
                 break
+            if cfg.worker_timeout_s > 0:
+                self._reap_dead_workers()
+            alive = cfg.num_workers - len(self.dead)
+            gate = min(cfg.gate, max(1, int(alive * cfg.bucket_ratio)))
             init_workers = (cfg.num_workers if first
                             else srv.available_workers())
-            if init_workers >= cfg.gate and self.pending:
+            if init_workers >= gate and self.pending:
                 first = False
                 workers_list = []
                 qsize = len(self.pending)
