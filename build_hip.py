@@ -18,7 +18,8 @@ ROOT = Path(__file__).resolve().parent
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "hipcc")
 
-SOURCES = [ROOT / "csrc" / "kernels.hip", ROOT / "csrc" / "bindings.cpp"]
+SOURCES = [ROOT / "csrc" / "kernels.hip", ROOT / "csrc" / "bindings.cpp",
+           ROOT / "csrc" / "libsvm_parser.cpp"]
 HEADERS = [ROOT / "csrc" / "philox.h"]
 
 

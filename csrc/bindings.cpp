@@ -17,6 +17,8 @@
 
 namespace py = pybind11;
 
+void register_libsvm(py::module_& m);  // csrc/libsvm_parser.cpp
+
 extern "C" {
 void launch_grad_dense(const void*, const float*, const float*, float*,
                        float*, int*, const int*, long, int, uint64_t,
@@ -163,6 +165,8 @@ PYBIND11_MODULE(_hip_core, m) {
                                    (hipStream_t)stream);
           check(hipGetLastError(), "saga_update_fused launch");
         });
+
+  register_libsvm(m);
 
   m.attr("__hip__") = true;
 }

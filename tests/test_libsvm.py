@@ -36,3 +36,14 @@ def test_n_features_override(tmp_path):
     path = _write(tmp_path)
     indptr, indices, values, y = load_libsvm(path, n_features=10)
     assert int(indices.max()) < 10
+
+
+def test_native_matches_python_fallback(tmp_path, monkeypatch):
+    path = _write(tmp_path)
+    from asyncframework_amd.data import libsvm as L
+    nat = L.load_libsvm(path)
+    # force the pure-python path
+    monkeypatch.setattr(L, "_parse_native", lambda p: None)
+    pyr = L.load_libsvm(path)
+    for a, b in zip(nat, pyr):
+        assert torch.equal(a, b)
