@@ -48,6 +48,8 @@ class _LocalChannel:
     straight into the server's completion path (the JobWaiter.taskSucceeded
     analog, reference JobWaiter.scala:56-60)."""
 
+    spin_budget = 0  # >0: GIL-yielding hot-window poll after each round
+
     def __init__(self, worker: Worker, server: Server):
         self.worker = worker
         self.server = server
@@ -64,8 +66,13 @@ class _LocalChannel:
         self._ev.set()
 
     def _loop(self):
+        spin = 0
         while True:
             if not self.q:
+                if spin > 0:  # hot window right after a round: the server
+                    spin -= 1  # usually redispatches within ~100 us
+                    time.sleep(0)
+                    continue
                 self._ev.wait(timeout=0.05)
                 self._ev.clear()
                 continue
@@ -74,6 +81,7 @@ class _LocalChannel:
                 break
             res = self.worker.process(msg)
             self.server.on_completion(res)
+            spin = self.spin_budget
 
     def join(self, timeout=None):
         self.thread.join(timeout)
@@ -134,114 +142,121 @@ class AsyncEngine:
                     and now - sub > cfg.worker_timeout_s):
                 self.dead.add(wid)
 
-    # -- updater thread (reference SparkASGDThread.scala:153-226) ------------
-    def _updater(self):
-        import queue as _q
+    # -- server loop: mailbox drain + tau filter + update + requeue +
+    #    quorum-gated dispatch. The reference splits this across its driver
+    #    main loop and updater thread (SparkASGDThread.scala:153-226,
+    #    230-345); folding them into ONE thread removes two cross-thread
+    #    handoffs per update (measured: the threaded engine was handoff-
+    #    bound at ~770 us/update with null workers).
+    def _process_result(self, pr) -> None:
         cfg, srv = self.cfg, self.server
-        while srv.k < cfg.num_iterations and not self._stop.is_set():
-            # block on the mailbox (reference blocks in ASYNCcollectAll too,
-            # ASYNCcontext.scala:60-62); drain whatever arrived
-            try:
-                pr = srv.AC.ASYNCcollectAll(timeout=0.05)
-            except _q.Empty:
-                continue
-            batch = [pr]
-            while srv.AC.hasNext():
-                batch.append(srv.AC.ASYNCcollectAll())
-            for pr in batch:
-                res: WorkerResult = pr.gettaskResult()
-                wid = pr.getWorkerID()
-                now = time.perf_counter()
-                if self.cfg.algo == "asaga":
-                    self.staleness_seen.append(srv.k - pr.getStaleness())
-                else:
-                    self.staleness_seen.append(pr.getStaleness())
-                if self.accepts_now(pr):
-                    self.accepted_staleness.append(self.staleness_seen[-1])
-                    srv.finish_time[wid] = now
-                    sub = srv.submit_time.get(wid)
-                    if sub is not None:
-                        self.delay.record_task(srv.k, (now - sub) * 1000.0)
-                    srv.apply(res)
-                    srv.last_accept[wid] = True
-                    self.pending.append(wid)
-                    if srv.k % cfg.printer_freq == 0:
-                        if self.verbose:
-                            print(f"Iteration {srv.k} is finished")
-                        srv.maybe_log()
-                    srv.k += 1
-                    self.applied += 1
-                    if srv.k in self.mark_at:
-                        self.marks[srv.k] = time.perf_counter()
-                    if (cfg.checkpoint_every > 0 and cfg.checkpoint_path
-                            and srv.k % cfg.checkpoint_every == 0):
-                        from .checkpoint import save_checkpoint
-                        save_checkpoint(cfg.checkpoint_path, srv,
-                                        self._local_workers())
-                else:
-                    srv.last_accept[wid] = False
-                    self.pending.append(wid)
-                    self.rejected += 1
-                if srv.k >= cfg.num_iterations:
-                    break
-            self._pending_ev.set()
+        res: WorkerResult = pr.gettaskResult()
+        wid = pr.getWorkerID()
+        now = time.perf_counter()
+        if cfg.algo == "asaga":
+            self.staleness_seen.append(srv.k - pr.getStaleness())
+        else:
+            self.staleness_seen.append(pr.getStaleness())
+        if self.accepts_now(pr):
+            self.accepted_staleness.append(self.staleness_seen[-1])
+            srv.finish_time[wid] = now
+            sub = srv.submit_time.get(wid)
+            if sub is not None:
+                self.delay.record_task(srv.k, (now - sub) * 1000.0)
+            srv.apply(res)
+            srv.last_accept[wid] = True
+            self.pending.append(wid)
+            if srv.k % cfg.printer_freq == 0:
+                if self.verbose:
+                    print(f"Iteration {srv.k} is finished")
+                srv.maybe_log()
+            srv.k += 1
+            self.applied += 1
+            if srv.k in self.mark_at:
+                self.marks[srv.k] = time.perf_counter()
+            if (cfg.checkpoint_every > 0 and cfg.checkpoint_path
+                    and srv.k % cfg.checkpoint_every == 0):
+                from .checkpoint import save_checkpoint
+                save_checkpoint(cfg.checkpoint_path, srv,
+                                self._local_workers())
+        else:
+            srv.last_accept[wid] = False
+            self.pending.append(wid)
+            self.rejected += 1
+
+    def _dispatch_pending(self, first: bool = False) -> None:
+        """Quorum gate + dispatch (reference main loop,
+        SparkASGDThread.scala:230-345)."""
+        cfg, srv = self.cfg, self.server
+        if not self.pending:
+            return
+        alive = cfg.num_workers - len(self.dead)
+        gate = min(cfg.gate, max(1, int(alive * cfg.bucket_ratio)))
+        init_workers = (cfg.num_workers if first
+                        else srv.available_workers())
+        if init_workers < gate:
+            return
+        workers_list = []
+        qsize = len(self.pending)
+        for _ in range(qsize):
+            workers_list.append(self.pending.popleft())
+        self.delay.maybe_activate(srv.k)
+        w_snap = srv.w.detach().clone()
+        now = time.perf_counter()
+        k_now = srv.k
+        for wid in workers_list:
+            prev_fin = srv.finish_time.get(wid, now)
+            srv.waiting_time[wid] = (srv.waiting_time.get(wid, 0)
+                                     + int((now - prev_fin) * 1000))
+            srv.submit_time[wid] = now
+            srv.AC.STAT[wid].setAvailability(False)
+            msg = Dispatch(
+                w=w_snap, ts=srv.AC.getCurrentTime(), k_submit=k_now,
+                accept_prev=srv.last_accept.get(wid, True),
+                delay_s=self.delay.delay_ms(wid, k_now) / 1000.0)
+            self.channels[wid].dispatch(msg)
 
     def accepts_now(self, pr) -> bool:
         return self.server.accepts(pr)
 
-    # -- main dispatch loop (reference SparkASGDThread.scala:230-345) --------
     def run(self, max_wall_s: Optional[float] = None) -> RunResult:
+        import queue as _q
         cfg, srv = self.cfg, self.server
         srv.start_time = time.perf_counter()
         if cfg.snapshot_weights and not srv.opt_vars:
             srv.opt_vars.append((0, srv.w.detach().cpu().clone()))
         for ch in self.channels:
             ch.start()
-        updater = threading.Thread(target=self._updater, daemon=True,
-                                   name="updater")
-        updater.start()
         t_start = time.perf_counter()
-        first = True
-        while srv.k < cfg.num_iterations:
+        self._dispatch_pending(first=True)
+        while srv.k < cfg.num_iterations and not self._stop.is_set():
             if max_wall_s and time.perf_counter() - t_start > max_wall_s:
                 self._stop.set()
                 break
+            # brief GIL-yielding spin before the blocking (Condition-based)
+            # wait — saves a ~100 us thread wakeup per result when hot
+            spin = 0
+            while not srv.AC.hasNext() and spin < 100:
+                time.sleep(0)
+                spin += 1
+            try:
+                pr = srv.AC.ASYNCcollectAll(timeout=0.05)
+            except _q.Empty:
+                if cfg.worker_timeout_s > 0:
+                    self._reap_dead_workers()
+                    self._dispatch_pending()
+                continue
+            self._process_result(pr)
+            while srv.AC.hasNext() and srv.k < cfg.num_iterations:
+                self._process_result(srv.AC.ASYNCcollectAll())
             if cfg.worker_timeout_s > 0:
                 self._reap_dead_workers()
-            alive = cfg.num_workers - len(self.dead)
-            gate = min(cfg.gate, max(1, int(alive * cfg.bucket_ratio)))
-            init_workers = (cfg.num_workers if first
-                            else srv.available_workers())
-            if init_workers >= gate and self.pending:
-                first = False
-                workers_list = []
-                qsize = len(self.pending)
-                for _ in range(qsize):
-                    workers_list.append(self.pending.popleft())
-                self.delay.maybe_activate(srv.k)
-                w_snap = srv.w.detach().clone()
-                now = time.perf_counter()
-                k_now = srv.k
-                for wid in workers_list:
-                    prev_fin = srv.finish_time.get(wid, now)
-                    srv.waiting_time[wid] = (srv.waiting_time.get(wid, 0)
-                                             + int((now - prev_fin) * 1000))
-                    srv.submit_time[wid] = now
-                    srv.AC.STAT[wid].setAvailability(False)
-                    msg = Dispatch(
-                        w=w_snap, ts=srv.AC.getCurrentTime(),
-                        k_submit=k_now,
-                        accept_prev=srv.last_accept.get(wid, True),
-                        delay_s=self.delay.delay_ms(wid, k_now) / 1000.0)
-                    self.channels[wid].dispatch(msg)
-            else:
-                self._pending_ev.wait(timeout=0.01)
-                self._pending_ev.clear()
+            if srv.k < cfg.num_iterations:
+                self._dispatch_pending()
         elapsed = srv.elapsed_ms()
         self._stop.set()
         for ch in self.channels:
             ch.dispatch(Dispatch(w=None, stop=True))
-        updater.join(timeout=10.0)
         for ch in self.channels:
             ch.join(timeout=10.0)
         return RunResult(k=srv.k, elapsed_ms=elapsed, opt_vars=srv.opt_vars,
