@@ -524,7 +524,11 @@ __global__ __launch_bounds__(BLOCK) void reduce_partials_kernel(
 
 // ---------------------------------------------------------------- K2 (+K3)
 
-template <typename VT, bool SAGA>
+// LPR = lanes per row: rcv1-class rows have ~73 nnz, so a 64-lane row
+// wastes most of the wave; 16-lane sub-waves keep 4 rows in flight per
+// wave (the gradient scatter uses global atomics — no LDS slabs — so
+// unlike the dense kernel, sub-waves here cost no occupancy).
+template <typename VT, bool SAGA, int LPR>
 __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     const int* __restrict__ indptr, const int* __restrict__ indices,
     const VT* __restrict__ values, const float* __restrict__ y,
@@ -535,8 +539,11 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, uint32_t threshold, int take_all, int objective) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
+  constexpr int NSUB = WAVE / LPR;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
+  const int sub = lane / LPR;
+  const int sl = lane % LPR;
   int local_count = 0;
   const long gstride = (long)gridDim.x * ROWS_PER_BLOCK_ITER;
   for (long bb = (long)blockIdx.x * ROWS_PER_BLOCK_ITER; bb < n_rows;
@@ -557,22 +564,23 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       unsigned long long mm = m[i];
+      for (int s2 = 0; s2 < sub && mm; ++s2) mm &= mm - 1;  // my first bit
       while (mm) {
         const int bit = __ffsll((long long)mm) - 1;
-        mm &= mm - 1;
         const long rr = base + 4L * bit + i;
         const int s = indptr[rr], t = indptr[rr + 1];
         float z = 0.f;
-        for (int p = s + lane; p < t; p += WAVE)
+        for (int p = s + sl; p < t; p += LPR)
           z += to_f32<VT>(values[p]) * w[indices[p]];
 #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) z += __shfl_xor(z, off, WAVE);
+        for (int off = LPR / 2; off > 0; off >>= 1)
+          z += __shfl_xor(z, off, WAVE);
         float e = link_residual(z, y[rr], objective);
         float coeff = e;
         if (SAGA) {
           const float a_old = alpha[rr];
           coeff = e - a_old;
-          if (lane == 0) {
+          if (sl == 0) {
             if (commit_now) {
               alpha[rr] = e;
             } else {
@@ -583,12 +591,13 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
           }
         }
         ++local_count;
-        for (int p = s + lane; p < t; p += WAVE)
+        for (int p = s + sl; p < t; p += LPR)
           atomicAdd(&g_out[indices[p]], coeff * to_f32<VT>(values[p]));
+        for (int s2 = 0; s2 < NSUB && mm; ++s2) mm &= mm - 1;  // next mine
       }
     }
   }
-  if (lane == 0 && local_count) atomicAdd(n_out, local_count);
+  if (sl == 0 && local_count) atomicAdd(n_out, local_count);
 }
 
 // ---------------------------------------------------------------- K5/K6
@@ -789,6 +798,12 @@ void launch_reduce_partials(const float* g_part, float* g_out, int d, int G,
                      0, stream, g_part, g_out, d, G, splits);
 }
 
+static inline int csr_lpr() {
+  const char* s = std::getenv("ASYNCAMD_CSR_LPR");
+  const int v = s ? std::atoi(s) : 32;  // measured best on rcv1 shape
+  return (v == 8 || v == 16 || v == 32 || v == 64) ? v : 32;
+}
+
 void launch_grad_csr(const int* indptr, const int* indices, const void* values,
                      const float* y, const float* w, float* g_out, int* n_out,
                      const int* k_dev, long n_rows, uint64_t seed,
@@ -797,19 +812,24 @@ void launch_grad_csr(const int* indptr, const int* indices, const void* values,
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
-  if (v_is_bf16) {
-    hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, false>), dim3(grid),
-                       dim3(BLOCK), 0, stream, indptr, indices,
-                       (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       nullptr, nullptr, nullptr, nullptr, k_dev, 0, n_rows,
-                       seed, round_k, row_start, thr, take_all, objective);
-  } else {
-    hipLaunchKernelGGL((grad_csr_kernel<float, false>), dim3(grid),
-                       dim3(BLOCK), 0, stream, indptr, indices,
-                       (const float*)values, y, w, g_out, n_out, nullptr,
-                       nullptr, nullptr, nullptr, k_dev, 0, n_rows, seed,
-                       round_k, row_start, thr, take_all, objective);
-  }
+  const int lpr = csr_lpr();
+#define CSR_LAUNCH(VT, CAST, L)                                              \
+  hipLaunchKernelGGL((grad_csr_kernel<VT, false, L>), dim3(grid),            \
+                     dim3(BLOCK), 0, stream, indptr, indices, (CAST)values,  \
+                     y, w, g_out, n_out, nullptr, nullptr, nullptr, nullptr, \
+                     k_dev, 0, n_rows, seed, round_k, row_start, thr,        \
+                     take_all, objective)
+#define CSR_DISPATCH(VT, CAST)                                               \
+  do {                                                                       \
+    if (lpr == 8) CSR_LAUNCH(VT, CAST, 8);                                   \
+    else if (lpr == 32) CSR_LAUNCH(VT, CAST, 32);                            \
+    else if (lpr == 64) CSR_LAUNCH(VT, CAST, 64);                            \
+    else CSR_LAUNCH(VT, CAST, 16);                                           \
+  } while (0)
+  if (v_is_bf16) CSR_DISPATCH(__hip_bfloat16, const __hip_bfloat16*);
+  else CSR_DISPATCH(float, const float*);
+#undef CSR_DISPATCH
+#undef CSR_LAUNCH
 }
 
 void launch_saga_grad_csr(const int* indptr, const int* indices,
@@ -823,20 +843,24 @@ void launch_saga_grad_csr(const int* indptr, const int* indices,
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
-  if (v_is_bf16) {
-    hipLaunchKernelGGL((grad_csr_kernel<__hip_bfloat16, true>), dim3(grid),
-                       dim3(BLOCK), 0, stream, indptr, indices,
-                       (const __hip_bfloat16*)values, y, w, g_out, n_out,
-                       alpha, idx_out, e_out, pos_ctr, k_dev, commit_now,
-                       n_rows, seed, round_k, row_start, thr, take_all,
-                       objective);
-  } else {
-    hipLaunchKernelGGL((grad_csr_kernel<float, true>), dim3(grid),
-                       dim3(BLOCK), 0, stream, indptr, indices,
-                       (const float*)values, y, w, g_out, n_out, alpha,
-                       idx_out, e_out, pos_ctr, k_dev, commit_now, n_rows,
-                       seed, round_k, row_start, thr, take_all, objective);
-  }
+  const int lpr = csr_lpr();
+#define CSR_SLAUNCH(VT, CAST, L)                                             \
+  hipLaunchKernelGGL((grad_csr_kernel<VT, true, L>), dim3(grid),             \
+                     dim3(BLOCK), 0, stream, indptr, indices, (CAST)values,  \
+                     y, w, g_out, n_out, alpha, idx_out, e_out, pos_ctr,     \
+                     k_dev, commit_now, n_rows, seed, round_k, row_start,    \
+                     thr, take_all, objective)
+#define CSR_SDISPATCH(VT, CAST)                                              \
+  do {                                                                       \
+    if (lpr == 8) CSR_SLAUNCH(VT, CAST, 8);                                  \
+    else if (lpr == 32) CSR_SLAUNCH(VT, CAST, 32);                           \
+    else if (lpr == 64) CSR_SLAUNCH(VT, CAST, 64);                           \
+    else CSR_SLAUNCH(VT, CAST, 16);                                          \
+  } while (0)
+  if (v_is_bf16) CSR_SDISPATCH(__hip_bfloat16, const __hip_bfloat16*);
+  else CSR_SDISPATCH(float, const float*);
+#undef CSR_SDISPATCH
+#undef CSR_SLAUNCH
 }
 
 void launch_sgd_update(float* w, const float* g, float gamma_k,
