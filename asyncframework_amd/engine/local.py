@@ -234,9 +234,12 @@ class AsyncEngine:
                 self._stop.set()
                 break
             # brief GIL-yielding spin before the blocking (Condition-based)
-            # wait — saves a ~100 us thread wakeup per result when hot
+            # wait — saves a ~100 us thread wakeup per result when hot.
+            # Peek the underlying deque directly: Queue.empty() takes the
+            # mutex and contends with the producing worker threads.
+            _dq = srv.AC.ResultList.queue
             spin = 0
-            while not srv.AC.hasNext() and spin < 100:
+            while not _dq and spin < 100:
                 time.sleep(0)
                 spin += 1
             try:
@@ -247,7 +250,7 @@ class AsyncEngine:
                     self._dispatch_pending()
                 continue
             self._process_result(pr)
-            while srv.AC.hasNext() and srv.k < cfg.num_iterations:
+            while _dq and srv.k < cfg.num_iterations:
                 self._process_result(srv.AC.ASYNCcollectAll())
             if cfg.worker_timeout_s > 0:
                 self._reap_dead_workers()
