@@ -67,7 +67,10 @@ class GraphEngine:
         self._splits = 1
         if not shard.is_sparse:
             self._G = int(self._core.grad_grid(shard.n_rows))
-            self._splits = max(1, min(32, self._G // 128))
+            # reduce kernel parallelism: njc*splits blocks; G//8 keeps >=8
+            # partials per thread (measured: splits=4 -> 16 blocks -> 7 us
+            # latency-bound reduce; more blocks cut it to ~2-3 us)
+            self._splits = max(1, min(64, self._G // 8))
             self.g_part = torch.zeros(d * self._G, dtype=torch.float32,
                                       device=device)
 
