@@ -73,6 +73,29 @@ class GraphEngine:
             self._splits = max(1, min(64, self._G // 8))
             self.g_part = torch.zeros(d * self._G, dtype=torch.float32,
                                       device=device)
+        # scan/compute overlap (ASGD dense): round r+1's Philox scan is
+        # w-independent, so it runs on a side stream concurrent with round
+        # r's gradient+update. Rows land in a double-buffered global list
+        # with prefetched y values; a dedicated scan_round counter (bumped
+        # stream-order on the side stream) avoids racing the k++ update.
+        self._overlap = (cfg.algo == "asgd" and not shard.is_sparse
+                         and d % 4 == 0 and d <= 2048)
+        if self._overlap:
+            n = shard.n_rows
+            self._rowlist = [torch.zeros(n, dtype=torch.int32, device=device)
+                             for _ in range(2)]
+            self._ylist = [torch.zeros(n, dtype=torch.float32, device=device)
+                           for _ in range(2)]
+            self._count = [torch.zeros(1, dtype=torch.int32, device=device)
+                           for _ in range(2)]
+            self._scan_round = torch.zeros(1, dtype=torch.int32,
+                                           device=device)
+            self._sideB = torch.cuda.Stream(device)
+            self._ev_scan = [torch.cuda.Event(), torch.cuda.Event()]
+            self._ev_grad = [torch.cuda.Event(), torch.cuda.Event()]
+            self._cur = 0
+            if self.unroll % 2:
+                self.unroll += 1  # buffer parity must round-trip
 
     # -- one round = [grad(k_dev), fused_update(k_dev++)] --------------------
     def _launch_round(self) -> None:
@@ -129,6 +152,101 @@ class GraphEngine:
                 cfg.gamma, 1.0 / cfg.par_recs, cfg.num_workers, cfg.d,
                 stream)
 
+    # ---- overlap-mode machinery (ASGD dense) ------------------------------
+    def _prime_lists(self) -> None:
+        """Establish the invariant: list[cur] holds rows for the CURRENT
+        k_dev (key k+1); scan_round == k+2."""
+        cfg, sh = self.cfg, self.shard
+        stream = torch.cuda.current_stream().cuda_stream
+        kh = int(self.k_dev.item())
+        self._count[self._cur].zero_()
+        self._core.scan_rows(sh.y.data_ptr(),
+                             self._rowlist[self._cur].data_ptr(),
+                             self._ylist[self._cur].data_ptr(),
+                             self._count[self._cur].data_ptr(), 0,
+                             sh.n_rows, cfg.seed, kh + 1, sh.row_start,
+                             cfg.batch_rate, stream)
+        self._scan_round.fill_(kh + 2)
+        torch.cuda.synchronize()
+
+    def _overlap_compute_part(self) -> None:
+        """grad(list[cur]) -> reduce -> fused update (k++), current stream."""
+        cfg, sh = self.cfg, self.shard
+        stream = torch.cuda.current_stream().cuda_stream
+        cur = self._cur
+        self._core.grad_dense_list(
+            sh.X.data_ptr(), self.w.data_ptr(), self.g_part.data_ptr(),
+            self._rowlist[cur].data_ptr(), self._ylist[cur].data_ptr(),
+            self._count[cur].data_ptr(), sh.n_rows, cfg.d, self._obj_code,
+            1 if sh.X.dtype == torch.bfloat16 else 0, stream)
+        self._core.reduce_partials(self.g_part.data_ptr(),
+                                   self.g.data_ptr(), cfg.d, self._G,
+                                   self._splits, stream)
+        self._core.sgd_update_fused(
+            self.w.data_ptr(), self.g.data_ptr(), self.k_dev.data_ptr(),
+            cfg.gamma, 1.0 / cfg.par_recs, cfg.num_workers, cfg.d, stream)
+
+    def _overlap_scan_part(self, nxt: int) -> None:
+        """memset count[nxt] -> scan(list[nxt]) -> bump scan_round, on the
+        CURRENT stream (caller picks main or side stream)."""
+        cfg, sh = self.cfg, self.shard
+        stream = torch.cuda.current_stream().cuda_stream
+        self._count[nxt].zero_()
+        self._core.scan_rows(sh.y.data_ptr(), self._rowlist[nxt].data_ptr(),
+                             self._ylist[nxt].data_ptr(),
+                             self._count[nxt].data_ptr(),
+                             self._scan_round.data_ptr(), sh.n_rows,
+                             cfg.seed, 0, sh.row_start, cfg.batch_rate,
+                             stream)
+        self._core.bump_counter(self._scan_round.data_ptr(), stream)
+
+    def _tail_round(self) -> None:
+        """One sequential round preserving the list invariant."""
+        nxt = self._cur ^ 1
+        self._overlap_compute_part()
+        self._overlap_scan_part(nxt)
+        self._cur = nxt
+
+    def _capture_overlap(self) -> None:
+        state = self._save_state()
+        s = torch.cuda.Stream(self.device)
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._prime_lists()
+            self._tail_round()   # warm every kernel
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self._restore_state(state)
+        self._cur = 0
+        self._prime_lists()
+        g = torch.cuda.CUDAGraph()
+        main = torch.cuda.current_stream()
+        with torch.cuda.graph(g):
+            main = torch.cuda.current_stream()
+            cur = 0
+            for it in range(self.unroll):
+                nxt = cur ^ 1
+                # side stream: scan round r+1 into list[nxt]
+                self._sideB.wait_stream(main) if it == 0 else None
+                if it > 0:
+                    self._sideB.wait_event(self._ev_grad[nxt])
+                with torch.cuda.stream(self._sideB):
+                    self._overlap_scan_part(nxt)
+                    self._ev_scan[nxt].record(self._sideB)
+                # main stream: compute round r from list[cur]
+                if it > 0:
+                    main.wait_event(self._ev_scan[cur])
+                saved = self._cur
+                self._cur = cur
+                self._overlap_compute_part()
+                self._cur = saved
+                self._ev_grad[cur].record(main)
+                cur = nxt
+            main.wait_stream(self._sideB)
+        self._graph = g
+        # capture executes nothing: state still = primed state for cur=0
+        torch.cuda.synchronize()
+
     def _capture(self) -> None:
         # warm the kernels outside capture, then roll state back
         state = self._save_state()
@@ -161,7 +279,21 @@ class GraphEngine:
     def step_rounds(self, n: int) -> None:
         """Advance n rounds (graph replays + per-round tail)."""
         if self._graph is None:
-            self._capture()
+            if self._overlap:
+                self._capture_overlap()
+            else:
+                self._capture()
+        if self._overlap:
+            # the graph is captured for buffer parity cur==0 and an even
+            # unroll (round-trips the parity); odd tails re-align first
+            while n > 0:
+                if self._cur != 0 or n < self.unroll:
+                    self._tail_round()
+                    n -= 1
+                else:
+                    self._graph.replay()
+                    n -= self.unroll
+            return
         full, rem = divmod(n, self.unroll)
         for _ in range(full):
             self._graph.replay()

@@ -30,6 +30,12 @@ void launch_saga_grad_dense(const void*, const float*, const float*, float*,
 int query_grad_grid(long);
 void launch_reduce_partials(const float*, float*, int, int, int,
                             hipStream_t);
+void launch_scan_rows(const float*, int*, float*, int*, const int*, long,
+                      uint64_t, uint32_t, uint64_t, double, hipStream_t);
+void launch_bump_counter(int*, hipStream_t);
+void launch_grad_dense_list(const void*, const float*, float*, const int*,
+                            const float*, const int*, long, int, int, int,
+                            hipStream_t);
 void launch_grad_csr(const int*, const int*, const void*, const float*,
                      const float*, float*, int*, const int*, long, uint64_t,
                      uint32_t, uint64_t, double, int, int, hipStream_t);
@@ -70,6 +76,33 @@ PYBIND11_MODULE(_hip_core, m) {
         });
 
   m.def("grad_grid", [](long n_rows) { return query_grad_grid(n_rows); });
+
+  m.def("scan_rows",
+        [](uintptr_t y, uintptr_t rowlist, uintptr_t ylist, uintptr_t count,
+           uintptr_t scan_round, long n_rows, uint64_t seed, uint32_t round_k,
+           uint64_t row_start, double rate, uintptr_t stream) {
+          launch_scan_rows((const float*)y, (int*)rowlist, (float*)ylist,
+                           (int*)count, (const int*)scan_round, n_rows, seed,
+                           round_k, row_start, rate, (hipStream_t)stream);
+          check(hipGetLastError(), "scan_rows launch");
+        });
+
+  m.def("bump_counter", [](uintptr_t p, uintptr_t stream) {
+    launch_bump_counter((int*)p, (hipStream_t)stream);
+    check(hipGetLastError(), "bump_counter launch");
+  });
+
+  m.def("grad_dense_list",
+        [](uintptr_t X, uintptr_t w, uintptr_t g_part, uintptr_t rowlist,
+           uintptr_t ylist, uintptr_t count, long n_rows, int d,
+           int objective, int x_is_bf16, uintptr_t stream) {
+          launch_grad_dense_list((const void*)X, (const float*)w,
+                                 (float*)g_part, (const int*)rowlist,
+                                 (const float*)ylist, (const int*)count,
+                                 n_rows, d, objective, x_is_bf16,
+                                 (hipStream_t)stream);
+          check(hipGetLastError(), "grad_dense_list launch");
+        });
 
   m.def("reduce_partials",
         [](uintptr_t g_part, uintptr_t g, int d, int G, int splits,

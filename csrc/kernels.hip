@@ -314,6 +314,173 @@ __device__ __forceinline__ __amdgpu_buffer_rsrc_t row_rsrc(
                                            0x00020000);
 }
 
+// Standalone mask scan: fills a GLOBAL compacted row list (+ prefetched y
+// values) for a FUTURE round. The scan depends only on (seed, round), not on
+// w — so the graph engine runs round r+1's scan on a side stream overlapped
+// with round r's gradient/update. The round key comes from a dedicated
+// scan_round counter bumped stream-order by bump_counter_kernel (reading
+// k_dev here would race the concurrent update's k++).
+__global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
+    const float* __restrict__ y, int* __restrict__ rowlist,
+    float* __restrict__ ylist, int* __restrict__ count_dev,
+    const int* __restrict__ scan_round_dev, long n_rows, uint64_t seed,
+    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all) {
+  if (scan_round_dev) round_k = (uint32_t)(*scan_round_dev);
+  constexpr int SCAP = 2048;
+  __shared__ int rows_s[SCAP];
+  __shared__ int qn_s, base_s;
+  if (threadIdx.x == 0) qn_s = 0;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const long ngroups = (n_rows + ROWS_PER_BLOCK_ITER - 1) / ROWS_PER_BLOCK_ITER;
+  long gi = blockIdx.x;
+  bool done = false;
+  while (!done) {
+    while (true) {
+      __syncthreads();
+      if (gi >= ngroups || qn_s >= SCAP - ROWS_PER_BLOCK_ITER) break;
+      const long base = gi * (long)ROWS_PER_BLOCK_ITER +
+                        (long)wave * ROWS_PER_WAVE;
+      if (base < n_rows) {
+        const uint4 x = philox_block4(
+            seed, round_k, (row_start + (uint64_t)base) / 4 + (uint64_t)lane);
+        const long rem = n_rows - base;
+        const long lrow = base + 4L * lane;
+        const uint32_t xs[4] = {x.x, x.y, x.z, x.w};
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          if (4L * lane + i < rem && (take_all || xs[i] < threshold)) {
+            const int pos = atomicAdd(&qn_s, 1);
+            rows_s[pos] = (int)(lrow + i);
+          }
+        }
+      }
+      gi += gridDim.x;
+    }
+    __syncthreads();
+    const int nq = min(qn_s, SCAP);
+    if (threadIdx.x == 0 && nq > 0) base_s = atomicAdd(count_dev, nq);
+    __syncthreads();
+    for (int i = threadIdx.x; i < nq; i += BLOCK) {
+      const int r = rows_s[i];
+      rowlist[base_s + i] = r;
+      ylist[base_s + i] = y[r];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) qn_s = 0;
+    done = gi >= ngroups;
+  }
+}
+
+__global__ void bump_counter_kernel(int* __restrict__ p) { *p += 1; }
+
+// List-fed gradient: the process phase of the pipe kernel, consuming a
+// globally compacted (rowlist, ylist) produced by scan_rows_kernel. Each
+// block stages its contiguous slice into LDS first (per-row ordinary
+// global loads inside the pipeline would reintroduce the vmcnt(0) drains).
+template <typename XT, int PBLOCK, int DEPTH, int ITERS>
+__global__ __launch_bounds__(PBLOCK) void grad_dense_list_kernel(
+    const XT* __restrict__ X, const float* __restrict__ w,
+    float* __restrict__ g_part, const int* __restrict__ rowlist,
+    const float* __restrict__ ylist, const int* __restrict__ count_dev,
+    long n_rows, int d, int objective) {
+  constexpr int NW = PBLOCK / WAVE;
+  constexpr int DPAD = ITERS * 256;
+  constexpr int LCAP = 2048;
+  extern __shared__ float smem[];
+  float* w_lds = smem;                                // [DPAD]
+  float* gacc = smem + DPAD;                          // [NW][DPAD]
+  float* yq = smem + (size_t)(1 + NW) * DPAD;         // [LCAP]
+  int* rowq = (int*)(yq + LCAP);                      // [LCAP]
+  for (int j = threadIdx.x; j < DPAD; j += PBLOCK) {
+    w_lds[j] = j < d ? w[j] : 0.f;
+#pragma unroll
+    for (int s2 = 0; s2 < NW; ++s2) gacc[(size_t)s2 * DPAD + j] = 0.f;
+  }
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float* gw = gacc + (size_t)wave * DPAD;
+  const long total_elems = n_rows * (long)d;
+  using RV = RowVec<XT>;
+  const int total = *count_dev;
+  const int per_block = (total + gridDim.x - 1) / gridDim.x;
+  const int my0 = blockIdx.x * per_block;
+  const int my1 = min(total, my0 + per_block);
+  for (int off = my0; off < my1; off += LCAP) {
+    const int nq = min(my1 - off, LCAP);
+    __syncthreads();
+    for (int i = threadIdx.x; i < nq; i += PBLOCK) {
+      rowq[i] = rowlist[off + i];
+      yq[i] = ylist[off + i];
+    }
+    __syncthreads();
+    typename RV::T buf[DEPTH][ITERS];
+    if (nq > 0) {
+#pragma unroll
+      for (int p = 0; p < DEPTH; ++p) {
+        const int q = min(wave + p * NW, nq - 1);
+        const int rr = __builtin_amdgcn_readfirstlane(rowq[q]);
+        const auto rs = row_rsrc<XT>(X, total_elems, rr, d);
+#pragma unroll
+        for (int it = 0; it < ITERS; ++it)
+          buf[p][it] = RV::load(rs, (lane + it * WAVE) << RV::VOFF_SHIFT);
+      }
+    }
+    int q_base = wave;
+    while (q_base < nq) {
+#pragma unroll
+      for (int p = 0; p < DEPTH; ++p) {
+        const int q = q_base + p * NW;
+        if (q < nq) {
+          float z = 0.f;
+#pragma unroll
+          for (int it = 0; it < ITERS; ++it) {
+            float o[4];
+            cvt4(buf[p][it], o);
+            const int j4 = lane + it * WAVE;
+            const float4 wv = reinterpret_cast<const float4*>(w_lds)[j4];
+            z += o[0] * wv.x + o[1] * wv.y + o[2] * wv.z + o[3] * wv.w;
+          }
+#pragma unroll
+          for (int offx = 32; offx > 0; offx >>= 1)
+            z += __shfl_xor(z, offx, WAVE);
+          const float coeff = link_residual(z, yq[q], objective);
+#pragma unroll
+          for (int it = 0; it < ITERS; ++it) {
+            float o[4];
+            cvt4(buf[p][it], o);
+            const int j4 = lane + it * WAVE;
+            float4* gw4 = reinterpret_cast<float4*>(gw);
+            float4 cur = gw4[j4];
+            cur.x += coeff * o[0]; cur.y += coeff * o[1];
+            cur.z += coeff * o[2]; cur.w += coeff * o[3];
+            gw4[j4] = cur;
+          }
+          __builtin_amdgcn_sched_barrier(0);
+          const int qf = min(q + DEPTH * NW, nq - 1);
+          {
+            const int rr2 = __builtin_amdgcn_readfirstlane(rowq[qf]);
+            const auto rs2 = row_rsrc<XT>(X, total_elems, rr2, d);
+#pragma unroll
+            for (int it = 0; it < ITERS; ++it)
+              buf[p][it] = RV::load(rs2, (lane + it * WAVE) << RV::VOFF_SHIFT);
+          }
+          __builtin_amdgcn_sched_barrier(0);
+        }
+      }
+      q_base += DEPTH * NW;
+    }
+  }
+  __syncthreads();
+  const size_t G = gridDim.x;
+  for (int j = threadIdx.x; j < d; j += PBLOCK) {
+    float s = 0.f;
+#pragma unroll
+    for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * DPAD + j];
+    g_part[(size_t)j * G + blockIdx.x] = s;
+  }
+}
+
 template <typename XT, bool SAGA, int PBLOCK, int DEPTH, int ITERS>
 __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     const XT* __restrict__ X, const float* __restrict__ y,
@@ -753,6 +920,54 @@ static void launch_dense(const XT* X, const float* y, const float* w,
 extern "C" {
 
 int query_grad_grid(long n_rows) { return grad_grid(n_rows); }
+
+void launch_scan_rows(const float* y, int* rowlist, float* ylist,
+                      int* count_dev, const int* scan_round_dev, long n_rows,
+                      uint64_t seed, uint32_t round_k, uint64_t row_start,
+                      double rate, hipStream_t stream) {
+  const uint32_t thr = philox_threshold(rate);
+  const int take_all = rate >= 1.0;
+  const int grid = grad_grid(n_rows);
+  hipLaunchKernelGGL(scan_rows_kernel, dim3(grid), dim3(BLOCK), 0, stream, y,
+                     rowlist, ylist, count_dev, scan_round_dev, n_rows, seed,
+                     round_k, row_start, thr, take_all);
+}
+
+void launch_bump_counter(int* p, hipStream_t stream) {
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, stream, p);
+}
+
+void launch_grad_dense_list(const void* X, const float* w, float* g_part,
+                            const int* rowlist, const float* ylist,
+                            const int* count_dev, long n_rows, int d,
+                            int objective, int x_is_bf16,
+                            hipStream_t stream) {
+  const int grid = grad_grid(n_rows);
+  const int iters = (d + 255) / 256;
+  const size_t smem = (size_t)5 * (iters * 256) * sizeof(float) +
+                      2048 * (sizeof(float) + sizeof(int));
+#define LIST_LAUNCH(XT, CAST, IT)                                            \
+  hipLaunchKernelGGL((grad_dense_list_kernel<XT, 256, 4, IT>), dim3(grid),   \
+                     dim3(256), smem, stream, (CAST)X, w, g_part, rowlist,   \
+                     ylist, count_dev, n_rows, d, objective)
+#define LIST_DISPATCH(XT, CAST)                                              \
+  do {                                                                       \
+    switch (iters) {                                                         \
+      case 1: LIST_LAUNCH(XT, CAST, 1); break;                               \
+      case 2: LIST_LAUNCH(XT, CAST, 2); break;                               \
+      case 3: LIST_LAUNCH(XT, CAST, 3); break;                               \
+      case 4: LIST_LAUNCH(XT, CAST, 4); break;                               \
+      case 5: LIST_LAUNCH(XT, CAST, 5); break;                               \
+      case 6: LIST_LAUNCH(XT, CAST, 6); break;                               \
+      case 7: LIST_LAUNCH(XT, CAST, 7); break;                               \
+      default: LIST_LAUNCH(XT, CAST, 8); break;                              \
+    }                                                                        \
+  } while (0)
+  if (x_is_bf16) LIST_DISPATCH(__hip_bfloat16, const __hip_bfloat16*);
+  else LIST_DISPATCH(float, const float*);
+#undef LIST_DISPATCH
+#undef LIST_LAUNCH
+}
 
 void launch_grad_dense(const void* X, const float* y, const float* w,
                        float* g_out, float* g_part, int* n_out,
