@@ -90,6 +90,7 @@ class GraphEngine:
                            for _ in range(2)]
             self._scan_round = torch.zeros(1, dtype=torch.int32,
                                            device=device)
+            self._ticket = torch.zeros(1, dtype=torch.int32, device=device)
             self._sideB = torch.cuda.Stream(device)
             self._ev_scan = [torch.cuda.Event(), torch.cuda.Event()]
             self._ev_grad = [torch.cuda.Event(), torch.cuda.Event()]
@@ -179,12 +180,13 @@ class GraphEngine:
             self._rowlist[cur].data_ptr(), self._ylist[cur].data_ptr(),
             self._count[cur].data_ptr(), sh.n_rows, cfg.d, self._obj_code,
             1 if sh.X.dtype == torch.bfloat16 else 0, stream)
-        self._core.reduce_partials(self.g_part.data_ptr(),
-                                   self.g.data_ptr(), cfg.d, self._G,
-                                   self._splits, stream)
-        self._core.sgd_update_fused(
-            self.w.data_ptr(), self.g.data_ptr(), self.k_dev.data_ptr(),
-            cfg.gamma, 1.0 / cfg.par_recs, cfg.num_workers, cfg.d, stream)
+        # fused reduce+update: sums partials, applies the step, and the
+        # last-finishing block advances k (atomic ticket)
+        self._core.sgd_reduce_update(
+            self.g_part.data_ptr(), self.w.data_ptr(),
+            self.k_dev.data_ptr(), self._ticket.data_ptr(), cfg.gamma,
+            1.0 / cfg.par_recs, cfg.num_workers, cfg.d, self._G,
+            self._splits, stream)
 
     def _overlap_scan_part(self, nxt: int) -> None:
         """memset count[nxt] -> scan(list[nxt]) -> bump scan_round, on the

@@ -36,6 +36,8 @@ void launch_bump_counter(int*, hipStream_t);
 void launch_grad_dense_list(const void*, const float*, float*, const int*,
                             const float*, const int*, long, int, int, int,
                             hipStream_t);
+void launch_sgd_reduce_update(const float*, float*, int*, int*, float, float,
+                              int, int, int, int, hipStream_t);
 void launch_grad_csr(const int*, const int*, const void*, const float*,
                      const float*, float*, int*, const int*, long, uint64_t,
                      uint32_t, uint64_t, double, int, int, hipStream_t);
@@ -85,6 +87,17 @@ PYBIND11_MODULE(_hip_core, m) {
                            (int*)count, (const int*)scan_round, n_rows, seed,
                            round_k, row_start, rate, (hipStream_t)stream);
           check(hipGetLastError(), "scan_rows launch");
+        });
+
+  m.def("sgd_reduce_update",
+        [](uintptr_t g_part, uintptr_t w, uintptr_t k_dev, uintptr_t ticket,
+           float gamma, float inv_batch, int num_part, int d, int G,
+           int splits, uintptr_t stream) {
+          launch_sgd_reduce_update((const float*)g_part, (float*)w,
+                                   (int*)k_dev, (int*)ticket, gamma,
+                                   inv_batch, num_part, d, G, splits,
+                                   (hipStream_t)stream);
+          check(hipGetLastError(), "sgd_reduce_update launch");
         });
 
   m.def("bump_counter", [](uintptr_t p, uintptr_t stream) {

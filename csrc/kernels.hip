@@ -798,6 +798,54 @@ __global__ void saga_commit_kernel(float* __restrict__ alpha,
   if (i < n) alpha[idx[i]] = e[i];
 }
 
+// Fused reduce+update (overlap graph mode): one kernel sums the per-block
+// partial slabs AND applies the ASGD update. Each block reads k at entry
+// (k is stable for the whole kernel); the last-finishing block (atomic
+// ticket) advances k — every other block has already finished, and the
+// last block read k at its own start, so the increment races nothing.
+__global__ __launch_bounds__(BLOCK) void sgd_reduce_update_kernel(
+    const float* __restrict__ g_part, float* __restrict__ w,
+    int* __restrict__ k_dev, int* __restrict__ ticket, float gamma,
+    float inv_batch, int num_part, int d, int G, int splits) {
+  const int k = *k_dev;
+  const float gamma_k =
+      (float)((double)gamma / sqrt((double)(k / num_part + 1)));
+  const int njc = (d + BLOCK - 1) / BLOCK;
+  const int jc = blockIdx.x % njc;
+  const int sp = blockIdx.x / njc;
+  const int j = jc * BLOCK + threadIdx.x;
+  if (j < d) {
+    const int per = (G + splits - 1) / splits;
+    const int b0 = sp * per;
+    const int b1 = min(G, b0 + per);
+    const float* base = g_part + (size_t)j * G;
+    float s = 0.f;
+    if (((b1 - b0) & 3) == 0 && (b0 & 3) == 0) {
+      const float4* v = reinterpret_cast<const float4*>(base + b0);
+      const int n4 = (b1 - b0) >> 2;
+      for (int q = 0; q < n4; ++q) {
+        const float4 x = v[q];
+        s += x.x + x.y + x.z + x.w;
+      }
+    } else {
+      for (int b = b0; b < b1; ++b) s += base[b];
+    }
+    if (splits == 1) {
+      w[j] -= gamma_k * inv_batch * s;
+    } else {
+      atomicAdd(&w[j], -gamma_k * inv_batch * s);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int t = atomicAdd(ticket, 1);
+    if (t == (int)gridDim.x - 1) {
+      *ticket = 0;
+      *k_dev = k + 1;
+    }
+  }
+}
+
 // Fused device-loop update kernels (graph mode): ONE workgroup applies the
 // update, zeroes the gradient accumulator for the next round, and advances
 // the device round counter — so an unrolled sequence of
@@ -944,9 +992,22 @@ void launch_grad_dense_list(const void* X, const float* w, float* g_part,
                             hipStream_t stream) {
   const int grid = grad_grid(n_rows);
   const int iters = (d + 255) / 256;
+  const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
+  const int depth = dps ? std::atoi(dps) : 4;
   const size_t smem = (size_t)5 * (iters * 256) * sizeof(float) +
                       2048 * (sizeof(float) + sizeof(int));
+#define LIST_LAUNCH_D(XT, CAST, IT, DP)                                      \
+  hipLaunchKernelGGL((grad_dense_list_kernel<XT, 256, DP, IT>), dim3(grid),  \
+                     dim3(256), smem, stream, (CAST)X, w, g_part, rowlist,   \
+                     ylist, count_dev, n_rows, d, objective)
 #define LIST_LAUNCH(XT, CAST, IT)                                            \
+  do {                                                                       \
+    if (depth == 2) LIST_LAUNCH_D(XT, CAST, IT, 2);                          \
+    else if (depth == 6) LIST_LAUNCH_D(XT, CAST, IT, 6);                     \
+    else if (depth == 8) LIST_LAUNCH_D(XT, CAST, IT, 8);                     \
+    else LIST_LAUNCH_D(XT, CAST, IT, 4);                                     \
+  } while (0)
+#define LIST_LAUNCH_OLD(XT, CAST, IT)                                        \
   hipLaunchKernelGGL((grad_dense_list_kernel<XT, 256, 4, IT>), dim3(grid),   \
                      dim3(256), smem, stream, (CAST)X, w, g_part, rowlist,   \
                      ylist, count_dev, n_rows, d, objective)
@@ -967,6 +1028,8 @@ void launch_grad_dense_list(const void* X, const float* w, float* g_part,
   else LIST_DISPATCH(float, const float*);
 #undef LIST_DISPATCH
 #undef LIST_LAUNCH
+#undef LIST_LAUNCH_OLD
+#undef LIST_LAUNCH_D
 }
 
 void launch_grad_dense(const void* X, const float* y, const float* w,
@@ -1099,6 +1162,16 @@ void launch_saga_commit(float* alpha, const int* idx, const float* e, int n,
   const int grid = (n + 255) / 256;
   hipLaunchKernelGGL(saga_commit_kernel, dim3(grid), dim3(256), 0, stream,
                      alpha, idx, e, n);
+}
+
+void launch_sgd_reduce_update(const float* g_part, float* w, int* k_dev,
+                              int* ticket, float gamma, float inv_batch,
+                              int num_part, int d, int G, int splits,
+                              hipStream_t stream) {
+  const int njc = (d + BLOCK - 1) / BLOCK;
+  hipLaunchKernelGGL(sgd_reduce_update_kernel, dim3(njc * splits),
+                     dim3(BLOCK), 0, stream, g_part, w, k_dev, ticket, gamma,
+                     inv_batch, num_part, d, G, splits);
 }
 
 void launch_sgd_update_fused(float* w, float* g, int* k_dev, float gamma,
