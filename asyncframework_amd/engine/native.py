@@ -1,0 +1,104 @@
+"""Python adapter for the native C++ event-loop engine
+(csrc/engine_native.cpp): multi-worker bounded-staleness async on one GPU
+with zero Python in the round loop.
+
+Each worker owns a shard, a HIP stream (created natively), a weight-snapshot
+buffer (the versioned-broadcast semantic) and a gradient accumulator; the
+C++ loop does dispatch -> event poll -> tau filter -> fused update ->
+quorum-gated redispatch, including the reference's straggler model and
+calibration. Use for GPU multi-worker configs (the threaded Python engine
+stays for CPU tests, host-spill history, and as the semantics oracle)."""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .config import EngineConfig
+from .worker import Shard
+
+_OBJ = {"lsq": 0, "logistic": 1}
+_ALGO = {"asgd": 0, "asaga": 1}
+
+
+class NativeLocalEngine:
+    def __init__(self, cfg: EngineConfig, shards: List[Shard],
+                 device: torch.device):
+        from .. import _hip_core
+        self._core = _hip_core
+        assert device.type == "cuda", "native engine is GPU-only"
+        assert cfg.history_placement != "host", \
+            "host-spill history uses the threaded engine"
+        self.cfg = cfg
+        self.device = device
+        self.shards = shards
+        d = cfg.d
+        self.w = torch.zeros(d, dtype=torch.float32, device=device)
+        self.alpha_bar = torch.zeros(d, dtype=torch.float32, device=device)
+        self._bufs = []
+        self._keep = []   # keep tensors alive across the native call
+        self.alpha_tables: List[torch.Tensor] = []
+        for sh in shards:
+            wbuf = torch.zeros(d, dtype=torch.float32, device=device)
+            g = torch.zeros(d, dtype=torch.float32, device=device)
+            ctr = torch.zeros(2, dtype=torch.int32, device=device)
+            wd = dict(sparse=sh.is_sparse, y=sh.y.data_ptr(),
+                      wbuf=wbuf.data_ptr(), g=g.data_ptr(),
+                      ctr=ctr.data_ptr(), n_rows=sh.n_rows,
+                      row_start=sh.row_start, x_is_bf16=0)
+            keep = [wbuf, g, ctr]
+            if sh.is_sparse:
+                wd.update(indptr=sh.indptr.data_ptr(),
+                          indices=sh.indices.data_ptr(),
+                          values=sh.values.data_ptr())
+                wd["x_is_bf16"] = 1 if sh.values.dtype == torch.bfloat16 else 0
+            else:
+                wd.update(X=sh.X.data_ptr())
+                wd["x_is_bf16"] = 1 if sh.X.dtype == torch.bfloat16 else 0
+            if cfg.algo == "asaga":
+                alpha = torch.zeros(sh.n_rows, dtype=torch.float32,
+                                    device=device)
+                cap = sh.n_rows if cfg.batch_rate >= 1.0 else min(
+                    sh.n_rows, int(cfg.batch_rate * sh.n_rows * 3) + 4096)
+                idx_out = torch.zeros(cap, dtype=torch.int32, device=device)
+                e_out = torch.zeros(cap, dtype=torch.float32, device=device)
+                wd.update(alpha=alpha.data_ptr(), idx_out=idx_out.data_ptr(),
+                          e_out=e_out.data_ptr())
+                keep += [alpha, idx_out, e_out]
+                self.alpha_tables.append(alpha)
+            self._keep.extend(keep)
+            self._bufs.append(wd)
+
+    def run(self, num_iterations: Optional[int] = None,
+            mark_lo: int = -1, mark_hi: int = -1,
+            max_wall_s: float = 1800.0) -> Dict:
+        cfg = self.cfg
+        conf = dict(N=cfg.N, d=cfg.d, P=cfg.num_workers,
+                    iters=num_iterations or cfg.num_iterations,
+                    gamma=cfg.gamma, rate=cfg.batch_rate,
+                    bucket_ratio=cfg.bucket_ratio, taw=cfg.taw,
+                    seed=cfg.seed, algo=_ALGO[cfg.algo],
+                    objective=_OBJ[cfg.objective], coeff=cfg.delay_coeff,
+                    calib_window=cfg.calib_factor * cfg.num_workers,
+                    mark_lo=mark_lo, mark_hi=mark_hi,
+                    max_wall_s=max_wall_s)
+        torch.cuda.synchronize()
+        res = self._core.native_local_run(conf, self._bufs,
+                                          self.w.data_ptr(),
+                                          self.alpha_bar.data_ptr())
+        torch.cuda.synchronize()
+        return res
+
+    def bench(self, warmup: int, steps: int,
+              max_wall_s: float = 1800.0) -> Tuple[float, Dict]:
+        """Exactly-K-steps contract: the native loop stamps wall times when
+        k crosses warmup and warmup+steps (after a server-stream sync)."""
+        res = self.run(num_iterations=warmup + steps + 1,
+                       mark_lo=warmup, mark_hi=warmup + steps,
+                       max_wall_s=max_wall_s)
+        t0, t1 = res["mark_lo_t"], res["mark_hi_t"]
+        if not (t1 > t0 > 0):
+            raise RuntimeError(f"native bench marks missing: {res}")
+        return t1 - t0, res

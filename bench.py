@@ -72,7 +72,10 @@ def parse_args():
     p.add_argument("--objective", default="lsq", choices=["lsq", "logistic"])
     p.add_argument("--algo", default="", choices=["", "asgd", "asaga"])
     p.add_argument("--device", default=None, help="override (cpu for debug)")
-    p.add_argument("--engine", default="", choices=["", "graph", "threads"])
+    p.add_argument("--engine", default="",
+                   choices=["", "graph", "threads", "native"])
+    p.add_argument("--workers", type=int, default=0,
+                   help="logical workers (threads/native engines)")
     args = p.parse_args()
     preset = MODELS[args.model]
     args.rows = args.rows or preset["rows"]
@@ -87,7 +90,7 @@ def parse_args():
     args.sync = preset.get("sync", False)
     args.delay_coeff = preset.get("delay_coeff", 0.0)
     args.history = preset.get("history", "device")
-    args.preset_workers = preset.get("workers", 0)
+    args.preset_workers = args.workers or preset.get("workers", 0)
     return args
 
 
@@ -149,6 +152,28 @@ def run_single(args, device):
         data = synthetic_dense(args.rows, args.cols, seed=BASE["seed"],
                                dtype=dt, device=device,
                                objective=args.objective)
+
+    if args.engine == "native" and device.type == "cuda":
+        # C++ event-loop engine: multi-worker async, zero Python per round
+        from asyncframework_amd.engine.native import NativeLocalEngine
+        shards = []
+        for s, t in row_shards(args.rows, n_workers):
+            if args.sparse:
+                indptr, indices, values, y = data
+                base = int(indptr[s])
+                shards.append(Shard(
+                    row_start=s, n_rows=t - s,
+                    indptr=(indptr[s:t + 1] - base).contiguous(),
+                    indices=indices[base:int(indptr[t])],
+                    values=values[base:int(indptr[t])], y=y[s:t]))
+            else:
+                X, y = data
+                shards.append(Shard(row_start=s, n_rows=t - s, X=X[s:t],
+                                    y=y[s:t]))
+        neng = NativeLocalEngine(cfg, shards, device)
+        elapsed, _res = neng.bench(args.warmup, args.steps)
+        emit(args, cfg, elapsed, n_gpus=1)
+        return
 
     if args.engine == "graph" and device.type == "cuda":
         from asyncframework_amd.engine.graph import GraphEngine

@@ -19,7 +19,8 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "hipcc")
 
 SOURCES = [ROOT / "csrc" / "kernels.hip", ROOT / "csrc" / "bindings.cpp",
-           ROOT / "csrc" / "libsvm_parser.cpp"]
+           ROOT / "csrc" / "libsvm_parser.cpp",
+           ROOT / "csrc" / "engine_native.cpp"]
 HEADERS = [ROOT / "csrc" / "philox.h"]
 
 

@@ -18,6 +18,7 @@
 namespace py = pybind11;
 
 void register_libsvm(py::module_& m);  // csrc/libsvm_parser.cpp
+void register_native_engine(py::module_& m);  // csrc/engine_native.cpp
 
 extern "C" {
 void launch_grad_dense(const void*, const float*, const float*, float*,
@@ -213,6 +214,7 @@ PYBIND11_MODULE(_hip_core, m) {
         });
 
   register_libsvm(m);
+  register_native_engine(m);
 
   m.attr("__hip__") = true;
 }

@@ -1,0 +1,495 @@
+// Native async parameter-server engine (single GPU, multi-worker).
+//
+// The MI355X-native replacement for the reference's driver main loop +
+// updater thread + executor pool (SparkASGDThread.scala:153-345,
+// Executor.scala TaskRunner) for co-located workers: ONE host thread drives
+// an event loop over per-worker HIP streams —
+//
+//   dispatch(i): stream_i waits the last update event, copies w into the
+//                worker's snapshot buffer (the versioned-broadcast semantic),
+//                zeroes g_i, launches the fused gradient kernel with the
+//                round's Philox key, records done_i;
+//   poll:        hipEventQuery(done_i) == success -> completion bookkeeping
+//                (arrival clock, staleness, availability) exactly as the
+//                Python engine does;
+//   accept:      tau filter; SGD/SAGA update on the server stream (+ SAGA
+//                commit on the worker stream, ordered before its next
+//                round); requeue; quorum-gated redispatch;
+//   delay:       the reference's straggler model (cloud long-tail / coeff),
+//                implemented as host-side due-times on the dispatch queue —
+//                no thread sleeps, no locks, no GIL (released for the whole
+//                run).
+//
+// This is the same control plane as engine/local.py with the thread handoffs
+// (~100 us each under the GIL) replaced by ~1 us event polls. The RCCL
+// multi-GPU server reuses this loop shape with peers instead of streams.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cmath>
+#include <cstdint>
+#include <deque>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+extern "C" {
+void launch_grad_dense(const void*, const float*, const float*, float*,
+                       float*, int*, const int*, long, int, uint64_t,
+                       uint32_t, uint64_t, double, int, int, hipStream_t);
+void launch_saga_grad_dense(const void*, const float*, const float*, float*,
+                            float*, float*, int*, int*, float*, int*,
+                            const int*, int, long, int, uint64_t, uint32_t,
+                            uint64_t, double, int, int, hipStream_t);
+void launch_grad_csr(const int*, const int*, const void*, const float*,
+                     const float*, float*, int*, const int*, long, uint64_t,
+                     uint32_t, uint64_t, double, int, int, hipStream_t);
+void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
+                          const float*, float*, float*, int*, int*, float*,
+                          int*, const int*, int, long, uint64_t, uint32_t,
+                          uint64_t, double, int, int, hipStream_t);
+void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
+void launch_saga_update(float*, const float*, float*, float, float, float,
+                        int, hipStream_t);
+void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
+}
+
+namespace {
+
+#define HIP_CHECK(x)                                                       \
+  do {                                                                     \
+    hipError_t _e = (x);                                                   \
+    if (_e != hipSuccess)                                                  \
+      throw std::runtime_error(std::string("native engine: ") +            \
+                               hipGetErrorString(_e) + " @ " #x);          \
+  } while (0)
+
+// Host Philox (must match csrc/philox.h / utils/philox.py) for the
+// reproducible straggler draws (uniform01 counters (i,0,round,stream)).
+inline uint32_t philox_host_x0(uint64_t seed, uint32_t c0, uint32_t c1,
+                               uint32_t c2, uint32_t c3) {
+  uint32_t k0 = (uint32_t)(seed & 0xFFFFFFFFull);
+  uint32_t k1 = (uint32_t)(seed >> 32);
+  for (int r = 0; r < 10; ++r) {
+    uint64_t p0 = 0xD2511F53ull * (uint64_t)c0;
+    uint64_t p1 = 0xCD9E8D57ull * (uint64_t)c2;
+    uint32_t hi0 = (uint32_t)(p0 >> 32), lo0 = (uint32_t)p0;
+    uint32_t hi1 = (uint32_t)(p1 >> 32), lo1 = (uint32_t)p1;
+    uint32_t n0 = hi1 ^ c1 ^ k0;
+    uint32_t n1 = lo1;
+    uint32_t n2 = hi0 ^ c3 ^ k1;
+    uint32_t n3 = lo0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  return c0;
+}
+
+inline double uniform01_host(uint64_t seed, uint32_t round_k,
+                             uint32_t stream) {
+  return philox_host_x0(seed, 0u, 0u, round_k, stream) / 4294967296.0;
+}
+
+struct WorkerBuf {
+  // device pointers supplied by Python (torch tensors kept alive there)
+  uintptr_t X = 0, indptr = 0, indices = 0, values = 0, y = 0;
+  uintptr_t alpha = 0, idx_out = 0, e_out = 0;  // SAGA staging
+  uintptr_t wbuf = 0, g = 0, ctr = 0;           // snapshot, grad, counters
+  long n_rows = 0, row_start = 0;
+  bool sparse = false;
+  int x_is_bf16 = 0;
+  // runtime
+  hipStream_t stream = nullptr;
+  hipEvent_t done = nullptr;
+  bool busy = false;
+  int ts = 0;        // arrival clock at dispatch
+  long k_submit = 0; // round index at dispatch
+  bool pending_commit = false;
+  int pending_n = 0;
+  double submit_t = 0, finish_t = 0, waiting_ms = 0;
+  long tasks = 0;
+};
+
+struct EngineCfg {
+  long N = 0;
+  int d = 0, P = 1;
+  long iters = 0;
+  double gamma = 0.01, rate = 0.01, bucket_ratio = 0.7;
+  long taw = 1 << 30;
+  uint64_t seed = 42;
+  int algo = 0;       // 0 asgd, 1 asaga
+  int objective = 0;  // 0 lsq, 1 logistic
+  double coeff = 0.0; // delay model
+  long calib_window = 0;
+  long mark_lo = -1, mark_hi = -1;
+  double max_wall_s = 3600.0;
+};
+
+double now_s() {
+  using clk = std::chrono::steady_clock;
+  return std::chrono::duration<double>(clk::now().time_since_epoch()).count();
+}
+
+struct NativeEngine {
+  EngineCfg cfg;
+  std::vector<WorkerBuf> ws;
+  uintptr_t w = 0, alpha_bar = 0;
+  hipStream_t sstream = nullptr;  // server stream
+  hipEvent_t update_ev = nullptr;
+  // server state
+  long k = 0;
+  int clock = 0;  // arrival clock (ASYNCcontext.CurrentTime)
+  std::deque<int> pendingq;
+  std::deque<std::pair<double, int>> delayed;  // (due time, worker)
+  long applied = 0, rejected = 0;
+  long max_staleness_seen = -1;
+  // delay calibration (reference :177-186,:247-252)
+  double cul_time_ms = 0;
+  long cul_count = 0;
+  double avg_delay_ms = 0;
+  bool delay_flag = false;
+  // straggler sets (reference :124-141)
+  std::vector<int> straggler_kind;  // 0 none, 1 normal, 2 longtail
+  double mark_lo_t = 0, mark_hi_t = 0;
+
+  void init_stragglers() {
+    straggler_kind.assign(cfg.P, 0);
+    const int length = (int)std::lround(0.25 * cfg.P);
+    const int length_normal = (int)std::lround(0.8 * length);
+    const int length_longtail = length - length_normal;
+    for (int c = 0; c < length; ++c) {
+      const int idx = c * 4;
+      if (idx < cfg.P) straggler_kind[idx] = (c < length_longtail) ? 2 : 1;
+    }
+  }
+
+  double delay_ms_for(int wid, long round_k) const {
+    if (!delay_flag || cfg.coeff == 0.0) return 0.0;
+    if (cfg.coeff != -1.0) {
+      if (wid == 0 && cfg.coeff > 0) return std::round(cfg.coeff * avg_delay_ms);
+      return 0.0;
+    }
+    if (straggler_kind[wid] == 2) {
+      const double u = uniform01_host(cfg.seed, (uint32_t)round_k, wid);
+      return std::round((u * 7.5 + 2.5) * avg_delay_ms);
+    }
+    if (straggler_kind[wid] == 1) {
+      const double u = uniform01_host(cfg.seed, (uint32_t)round_k, wid);
+      return std::round((u + 1.5) * avg_delay_ms);
+    }
+    return 0.0;
+  }
+
+  int gate() const {
+    return std::max(1, (int)std::floor(cfg.P * cfg.bucket_ratio));
+  }
+
+  int available() const {
+    int n = 0;
+    for (auto& wk : ws)
+      if (!wk.busy) ++n;
+    return n;
+  }
+
+  void launch_grad(WorkerBuf& wk, long round_key) {
+    // zero gradient + counters, then the fused sample+gradient kernel
+    HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wk.stream));
+    HIP_CHECK(hipMemsetAsync((void*)wk.ctr, 0, 8, wk.stream));
+    if (cfg.algo == 1) {
+      if (wk.sparse)
+        launch_saga_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
+                             (const void*)wk.values, (const float*)wk.y,
+                             (const float*)wk.wbuf, (float*)wk.alpha,
+                             (float*)wk.g, (int*)wk.ctr, (int*)wk.idx_out,
+                             (float*)wk.e_out, (int*)(wk.ctr + 4), nullptr,
+                             0, wk.n_rows, cfg.seed, (uint32_t)round_key,
+                             (uint64_t)wk.row_start, cfg.rate, cfg.objective,
+                             wk.x_is_bf16, wk.stream);
+      else
+        launch_saga_grad_dense((const void*)wk.X, (const float*)wk.y,
+                               (const float*)wk.wbuf, (float*)wk.alpha,
+                               (float*)wk.g, nullptr, (int*)wk.ctr,
+                               (int*)wk.idx_out, (float*)wk.e_out,
+                               (int*)(wk.ctr + 4), nullptr, 0, wk.n_rows,
+                               cfg.d, cfg.seed, (uint32_t)round_key,
+                               (uint64_t)wk.row_start, cfg.rate,
+                               cfg.objective, wk.x_is_bf16, wk.stream);
+    } else {
+      if (wk.sparse)
+        launch_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
+                        (const void*)wk.values, (const float*)wk.y,
+                        (const float*)wk.wbuf, (float*)wk.g, (int*)wk.ctr,
+                        nullptr, wk.n_rows, cfg.seed, (uint32_t)round_key,
+                        (uint64_t)wk.row_start, cfg.rate, cfg.objective,
+                        wk.x_is_bf16, wk.stream);
+      else
+        launch_grad_dense((const void*)wk.X, (const float*)wk.y,
+                          (const float*)wk.wbuf, (float*)wk.g, nullptr,
+                          (int*)wk.ctr, nullptr, wk.n_rows, cfg.d, cfg.seed,
+                          (uint32_t)round_key, (uint64_t)wk.row_start,
+                          cfg.rate, cfg.objective, wk.x_is_bf16, wk.stream);
+    }
+    HIP_CHECK(hipGetLastError());
+  }
+
+  void dispatch(int wid, double t_now) {
+    WorkerBuf& wk = ws[wid];
+    // accept-gated SAGA history commit from the worker's previous round
+    if (cfg.algo == 1 && wk.pending_commit && wk.pending_n > 0) {
+      launch_saga_commit((float*)wk.alpha, (const int*)wk.idx_out,
+                         (const float*)wk.e_out, wk.pending_n, wk.stream);
+      HIP_CHECK(hipGetLastError());
+    }
+    wk.pending_commit = false;
+    wk.pending_n = 0;
+    wk.waiting_ms += (t_now - wk.finish_t) * 1000.0;
+    wk.submit_t = t_now;
+    wk.busy = true;
+    wk.ts = clock;
+    wk.k_submit = k;
+    // versioned weights: snapshot w on the worker stream, ordered after the
+    // latest applied update (ASYNCbroadcast semantics)
+    HIP_CHECK(hipStreamWaitEvent(wk.stream, update_ev, 0));
+    HIP_CHECK(hipMemcpyAsync((void*)wk.wbuf, (const void*)w,
+                             (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
+                             wk.stream));
+    launch_grad(wk, wk.k_submit + 1);  // reference seed+k+1
+    HIP_CHECK(hipEventRecord(wk.done, wk.stream));
+  }
+
+  void maybe_dispatch_pending(double t_now) {
+    if (pendingq.empty()) return;
+    if (available() < gate()) return;
+    // delay calibration activation (reference :247-252)
+    if (!delay_flag && k > cfg.calib_window) {
+      if (cul_count > 0) avg_delay_ms = cul_time_ms / cul_count;
+      delay_flag = true;
+    }
+    const size_t qn = pendingq.size();
+    for (size_t i = 0; i < qn; ++i) {
+      const int wid = pendingq.front();
+      pendingq.pop_front();
+      const double dly = delay_ms_for(wid, k) / 1000.0;
+      if (dly > 0) {
+        ws[wid].busy = true;  // occupied while "straggling"
+        delayed.emplace_back(t_now + dly, wid);
+      } else {
+        dispatch(wid, t_now);
+      }
+    }
+  }
+
+  void on_completion(int wid, double t_now) {
+    WorkerBuf& wk = ws[wid];
+    wk.busy = false;
+    wk.finish_t = t_now;
+    wk.tasks += 1;
+    const int staleness = clock - wk.ts;  // arrival-clock staleness
+    clock += 1;
+    max_staleness_seen = std::max<long>(max_staleness_seen, staleness);
+    const bool accept = (cfg.algo == 1) ? (k - wk.k_submit) <= cfg.taw
+                                        : staleness <= cfg.taw;
+    if (accept) {
+      if (k < cfg.calib_window) {
+        cul_time_ms += (t_now - wk.submit_t) * 1000.0;
+        cul_count += 1;
+      }
+      const double par_recs = cfg.rate * (double)cfg.N / cfg.P;
+      if (cfg.algo == 1) {
+        launch_saga_update((float*)w, (const float*)wk.g, (float*)alpha_bar,
+                           (float)cfg.gamma, (float)(1.0 / par_recs),
+                           (float)(1.0 / cfg.N), cfg.d, sstream);
+        wk.pending_commit = true;
+        wk.pending_n = -1;  // resolved lazily from ctr at commit time
+      } else {
+        const double gamma_k =
+            cfg.gamma / std::sqrt((double)(k / cfg.P + 1));
+        launch_sgd_update((float*)w, (const float*)wk.g, (float)gamma_k,
+                          (float)(1.0 / par_recs), cfg.d, sstream);
+      }
+      HIP_CHECK(hipGetLastError());
+      HIP_CHECK(hipEventRecord(update_ev, sstream));
+      k += 1;
+      applied += 1;
+      if (k == cfg.mark_lo) {
+        HIP_CHECK(hipStreamSynchronize(sstream));
+        mark_lo_t = now_s();
+      }
+      if (k == cfg.mark_hi) {
+        HIP_CHECK(hipStreamSynchronize(sstream));
+        mark_hi_t = now_s();
+      }
+    } else {
+      rejected += 1;
+      wk.pending_commit = false;
+    }
+    pendingq.push_back(wid);
+  }
+
+  void resolve_saga_commit_count(WorkerBuf& wk) {
+    // read the device sampled-count for the staged commit (host copy; only
+    // on accept, once per accepted SAGA round)
+    int n = 0;
+    HIP_CHECK(hipMemcpy(&n, (void*)(wk.ctr + 4), 4, hipMemcpyDeviceToHost));
+    wk.pending_n = n;
+  }
+
+  struct Result {
+    long k = 0, applied = 0, rejected = 0, max_staleness = -1;
+    double elapsed_ms = 0, mark_lo_t = 0, mark_hi_t = 0, avg_delay = 0;
+    std::vector<double> waits;
+  };
+
+  Result run() {
+    HIP_CHECK(hipStreamCreateWithFlags(&sstream, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreateWithFlags(&update_ev, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(update_ev, sstream));
+    for (auto& wk : ws) {
+      HIP_CHECK(hipStreamCreateWithFlags(&wk.stream, hipStreamNonBlocking));
+      HIP_CHECK(hipEventCreateWithFlags(&wk.done, hipEventDisableTiming));
+    }
+    init_stragglers();
+    for (int i = 0; i < cfg.P; ++i) pendingq.push_back(i);
+    const double t0 = now_s();
+    // first dispatch ignores the gate (reference k==0 path)
+    {
+      const size_t qn = pendingq.size();
+      for (size_t i = 0; i < qn; ++i) {
+        const int wid = pendingq.front();
+        pendingq.pop_front();
+        dispatch(wid, t0);
+      }
+    }
+    while (k < cfg.iters) {
+      const double t_now = now_s();
+      if (t_now - t0 > cfg.max_wall_s) break;
+      // release due delayed dispatches
+      while (!delayed.empty() && delayed.front().first <= t_now) {
+        const int wid = delayed.front().second;
+        delayed.pop_front();
+        dispatch(wid, t_now);
+      }
+      // poll completions
+      bool any = false;
+      for (int i = 0; i < cfg.P && k < cfg.iters; ++i) {
+        WorkerBuf& wk = ws[i];
+        if (!wk.busy) continue;
+        if (wk.submit_t > wk.finish_t && wk.done) {
+          const hipError_t q = hipEventQuery(wk.done);
+          if (q == hipSuccess) {
+            // SAGA: fetch the staged count before deciding (cheap, 4 B)
+            const bool will_commit = cfg.algo == 1;
+            on_completion(i, now_s());
+            if (will_commit && wk.pending_commit)
+              resolve_saga_commit_count(wk);
+            any = true;
+          } else if (q != hipErrorNotReady) {
+            HIP_CHECK(q);
+          }
+        }
+      }
+      if (any) maybe_dispatch_pending(now_s());
+    }
+    HIP_CHECK(hipStreamSynchronize(sstream));
+    const double t1 = now_s();
+    // drain in-flight rounds so buffers are quiescent before Python resumes
+    for (auto& wk : ws) HIP_CHECK(hipStreamSynchronize(wk.stream));
+    Result out;
+    out.k = k;
+    out.elapsed_ms = (t1 - t0) * 1000.0;
+    out.applied = applied;
+    out.rejected = rejected;
+    out.max_staleness = max_staleness_seen;
+    out.mark_lo_t = mark_lo_t;
+    out.mark_hi_t = mark_hi_t;
+    out.avg_delay = avg_delay_ms;
+    for (auto& wk : ws) out.waits.push_back(wk.waiting_ms);
+    for (auto& wk : ws) {
+      HIP_CHECK(hipEventDestroy(wk.done));
+      HIP_CHECK(hipStreamDestroy(wk.stream));
+    }
+    HIP_CHECK(hipEventDestroy(update_ev));
+    HIP_CHECK(hipStreamDestroy(sstream));
+    return out;
+  }
+};
+
+}  // namespace
+
+void register_native_engine(py::module_& m) {
+  m.def(
+      "native_local_run",
+      [](py::dict c, py::list workers, uintptr_t w, uintptr_t alpha_bar) {
+        NativeEngine eng;
+        EngineCfg& cfg = eng.cfg;
+        cfg.N = py::cast<long>(c["N"]);
+        cfg.d = py::cast<int>(c["d"]);
+        cfg.P = py::cast<int>(c["P"]);
+        cfg.iters = py::cast<long>(c["iters"]);
+        cfg.gamma = py::cast<double>(c["gamma"]);
+        cfg.rate = py::cast<double>(c["rate"]);
+        cfg.bucket_ratio = py::cast<double>(c["bucket_ratio"]);
+        cfg.taw = py::cast<long>(c["taw"]);
+        cfg.seed = py::cast<uint64_t>(c["seed"]);
+        cfg.algo = py::cast<int>(c["algo"]);
+        cfg.objective = py::cast<int>(c["objective"]);
+        cfg.coeff = py::cast<double>(c["coeff"]);
+        cfg.calib_window = py::cast<long>(c["calib_window"]);
+        cfg.mark_lo = py::cast<long>(c["mark_lo"]);
+        cfg.mark_hi = py::cast<long>(c["mark_hi"]);
+        cfg.max_wall_s = py::cast<double>(c["max_wall_s"]);
+        for (auto item : workers) {
+          py::dict wd = py::cast<py::dict>(item);
+          WorkerBuf wk;
+          wk.sparse = py::cast<bool>(wd["sparse"]);
+          if (wk.sparse) {
+            wk.indptr = py::cast<uintptr_t>(wd["indptr"]);
+            wk.indices = py::cast<uintptr_t>(wd["indices"]);
+            wk.values = py::cast<uintptr_t>(wd["values"]);
+          } else {
+            wk.X = py::cast<uintptr_t>(wd["X"]);
+          }
+          wk.y = py::cast<uintptr_t>(wd["y"]);
+          wk.wbuf = py::cast<uintptr_t>(wd["wbuf"]);
+          wk.g = py::cast<uintptr_t>(wd["g"]);
+          wk.ctr = py::cast<uintptr_t>(wd["ctr"]);
+          wk.n_rows = py::cast<long>(wd["n_rows"]);
+          wk.row_start = py::cast<long>(wd["row_start"]);
+          wk.x_is_bf16 = py::cast<int>(wd["x_is_bf16"]);
+          if (eng.cfg.algo == 1) {
+            wk.alpha = py::cast<uintptr_t>(wd["alpha"]);
+            wk.idx_out = py::cast<uintptr_t>(wd["idx_out"]);
+            wk.e_out = py::cast<uintptr_t>(wd["e_out"]);
+          }
+          eng.ws.push_back(wk);
+        }
+        if ((int)eng.ws.size() != cfg.P)
+          throw std::runtime_error("native engine: P != len(workers)");
+        eng.w = w;
+        eng.alpha_bar = alpha_bar;
+        NativeEngine::Result r;
+        {
+          py::gil_scoped_release rel;  // the event loop never touches Python
+          r = eng.run();
+        }
+        py::dict out;
+        out["k"] = r.k;
+        out["elapsed_ms"] = r.elapsed_ms;
+        out["applied"] = r.applied;
+        out["rejected"] = r.rejected;
+        out["max_staleness"] = r.max_staleness;
+        out["mark_lo_t"] = r.mark_lo_t;
+        out["mark_hi_t"] = r.mark_hi_t;
+        out["avg_delay_ms"] = r.avg_delay;
+        out["waiting_ms"] = r.waits;
+        return out;
+      });
+}
