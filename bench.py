@@ -42,15 +42,18 @@ MODELS = {
     "sync-tiny-cpu": dict(rows=1_000, cols=784, rate=0.3, algo="asgd",
                           sync=True, dtype="fp32", sparse=False,
                           engine="threads", device="cpu", workers=2),
+    # flagship: 32 async workers on streams (the reference's fixed
+    # partitions=32, README.md) driven by the native C++ event loop;
+    # --engine graph gives the single-worker hipGraph variant
     "asgd-mnist8m": dict(rows=8_100_000, cols=784, rate=0.01, algo="asgd",
                          sync=False, dtype="bf16", sparse=False,
-                         engine="graph"),
+                         engine="native", workers=32),
     "asaga-rcv1": dict(rows=697_641, cols=47_236, rate=0.02, algo="asaga",
                        sync=False, dtype="fp32", sparse=True,
-                       engine="graph"),
+                       engine="native", workers=32),
     "asgd-epsilon-delay": dict(rows=400_000, cols=2_000, rate=0.01,
                                algo="asgd", sync=False, dtype="fp32",
-                               sparse=False, engine="threads",
+                               sparse=False, engine="native",
                                delay_coeff=1.0, workers=8),
     "asaga-mnist8m-hostspill": dict(rows=8_100_000, cols=784, rate=0.01,
                                     algo="asaga", sync=False, dtype="bf16",
