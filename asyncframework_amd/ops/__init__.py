@@ -44,7 +44,7 @@ def _require_hip(op: str):
         if os.environ.get("ASYNCAMD_ALLOW_FALLBACK") == "1":
             return None
         raise RuntimeError(
-            f"{op}: HIP extension asyncframework_amd._hip is not built "
+            f"{op}: HIP extension asyncframework_amd._hip_core is not built "
             f"(import error: {_hip_err}). On a GPU box the native kernels are "
             f"mandatory — run `python setup.py build_ext --inplace` "
             f"(PYTORCH_ROCM_ARCH=gfx950). Set ASYNCAMD_ALLOW_FALLBACK=1 only "

@@ -115,6 +115,7 @@ struct WorkerBuf {
   bool pending_commit = false;
   int pending_n = 0;
   double submit_t = 0, finish_t = 0, waiting_ms = 0;
+  double poll_after = 0;  // don't hipEventQuery before this time
   long tasks = 0;
 };
 
@@ -159,6 +160,7 @@ struct NativeEngine {
   // straggler sets (reference :124-141)
   std::vector<int> straggler_kind;  // 0 none, 1 normal, 2 longtail
   double mark_lo_t = 0, mark_hi_t = 0;
+  double ewma_round_s = 0;  // poll gating
 
   void init_stragglers() {
     straggler_kind.assign(cfg.P, 0);
@@ -200,9 +202,11 @@ struct NativeEngine {
   }
 
   void launch_grad(WorkerBuf& wk, long round_key) {
-    // zero gradient + counters, then the fused sample+gradient kernel
+    // zero gradient (+ SAGA staging counters), then the fused kernel;
+    // ASGD never reads the sampled count, so its ctr memset is skipped
     HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wk.stream));
-    HIP_CHECK(hipMemsetAsync((void*)wk.ctr, 0, 8, wk.stream));
+    if (cfg.algo == 1)
+      HIP_CHECK(hipMemsetAsync((void*)wk.ctr, 0, 8, wk.stream));
     if (cfg.algo == 1) {
       if (wk.sparse)
         launch_saga_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
@@ -263,6 +267,9 @@ struct NativeEngine {
                              wk.stream));
     launch_grad(wk, wk.k_submit + 1);  // reference seed+k+1
     HIP_CHECK(hipEventRecord(wk.done, wk.stream));
+    // poll gating: querying every busy event each pass costs ~1-2 us per
+    // query; skip until ~60% of the EWMA round time has elapsed
+    wk.poll_after = t_now + 0.3 * ewma_round_s;
   }
 
   void maybe_dispatch_pending(double t_now) {
@@ -292,6 +299,8 @@ struct NativeEngine {
     wk.busy = false;
     wk.finish_t = t_now;
     wk.tasks += 1;
+    const double rt = t_now - wk.submit_t;
+    ewma_round_s = ewma_round_s == 0 ? rt : 0.9 * ewma_round_s + 0.1 * rt;
     const int staleness = clock - wk.ts;  // arrival-clock staleness
     clock += 1;
     max_staleness_seen = std::max<long>(max_staleness_seen, staleness);
@@ -383,6 +392,7 @@ struct NativeEngine {
         WorkerBuf& wk = ws[i];
         if (!wk.busy) continue;
         if (wk.submit_t > wk.finish_t && wk.done) {
+          if (t_now < wk.poll_after) continue;
           const hipError_t q = hipEventQuery(wk.done);
           if (q == hipSuccess) {
             // SAGA: fetch the staged count before deciding (cheap, 4 B)

@@ -222,13 +222,13 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
 
 #define QCAP 2048
 
-// Inline-asm row loads + hand-counted waits. hipcc's own scheduler defeats
-// the software pipeline (it emitted a vmcnt(0) drain before each dot and a
-// wait immediately after each refill — measured); per the CDNA guide the
-// fix is asm loads whose destinations are unprotected until OUR counted
-// s_waitcnt. Consumers are pinned below the wait by zero-cost empty-asm
-// register ties on each 32-bit component (tied vector operands are not
-// supported by the backend).
+// Row-fragment types for the pipelined loaders. Loads are compiler-visible
+// buffer intrinsics: with unconditional compile-time load counts and
+// sched_barrier(0) fences around the refills (see the pipe kernels), hipcc
+// emits a counted descending vmcnt ladder — a genuine software pipeline
+// with compiler-guaranteed hazards (hand-counted inline-asm waits were
+// tried and abandoned: regalloc preservation copies raced in-flight asm
+// loads).
 struct B16x4 { union { ushort4 s; uint2 u; }; };
 struct F32x4 { union { float4 f; uint4 u; }; };
 
@@ -244,12 +244,6 @@ template <> struct RowVec<float> {
     return out;
   }
   static constexpr int VOFF_SHIFT = 4;  // 16 B per lane-element
-  static __device__ __forceinline__ void tie(T& x) {
-    asm volatile("" : "+v"(x.u.x));
-    asm volatile("" : "+v"(x.u.y));
-    asm volatile("" : "+v"(x.u.z));
-    asm volatile("" : "+v"(x.u.w));
-  }
 };
 template <> struct RowVec<__hip_bfloat16> {
   using T = B16x4;
@@ -262,26 +256,7 @@ template <> struct RowVec<__hip_bfloat16> {
     return out;
   }
   static constexpr int VOFF_SHIFT = 3;  // 8 B per lane-element
-  static __device__ __forceinline__ void tie(T& x) {
-    asm volatile("" : "+v"(x.u.x));
-    asm volatile("" : "+v"(x.u.y));
-  }
 };
-
-template <int V>
-__device__ __forceinline__ void wait_vmcnt_imm() {
-  asm volatile("s_waitcnt vmcnt(%[v])" ::[v] "i"(V));
-}
-
-// Counted wait + pin this buffer's consumers below it. V is conservative-
-// safe at queue tails (fewer outstanding loads than the bound always
-// passes; FIFO vmcnt semantics drain everything older than the bound).
-template <int V, typename RV, typename T, int N>
-__device__ __forceinline__ void wait_and_tie(T (&b)[N]) {
-  wait_vmcnt_imm<V>();
-#pragma unroll
-  for (int i = 0; i < N; ++i) RV::tie(b[i]);
-}
 
 __device__ __forceinline__ void cvt4(const float4& r, float o[4]) {
   o[0] = r.x; o[1] = r.y; o[2] = r.z; o[3] = r.w;
