@@ -41,6 +41,9 @@ def _engine_flags(p: argparse.ArgumentParser) -> None:
                    help="SAGA history table in HBM (device) or pinned host "
                         "DRAM (host spill, BASELINE config 5)")
     p.add_argument("--max-wall-s", type=float, default=None)
+    p.add_argument("--engine", default="threads",
+                   choices=["threads", "native"],
+                   help="native = C++ event-loop engine (GPU, async only)")
 
 
 def _parse13(argv: List[str], prog: str):
@@ -83,7 +86,8 @@ def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:
         workers = runner.build_csr_workers(cfg, *data)
     else:
         workers = runner.build_dense_workers(cfg, *data)
-    res, _srv = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s)
+    res, _srv = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s,
+                                  engine=a.engine)
     runner.final_report(cfg, res, data, sparse, device=a.device)
 
 

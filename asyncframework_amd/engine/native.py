@@ -73,22 +73,39 @@ class NativeLocalEngine:
 
     def run(self, num_iterations: Optional[int] = None,
             mark_lo: int = -1, mark_hi: int = -1,
-            max_wall_s: float = 1800.0) -> Dict:
+            max_wall_s: float = 1800.0,
+            snapshot_every: int = 0) -> Dict:
+        """snapshot_every > 0 records (ms, w) optVars every that many
+        applied updates (the reference's printer_freq loss-curve mechanism)
+        into a device ring, returned as res['opt_vars']."""
         cfg = self.cfg
-        conf = dict(N=cfg.N, d=cfg.d, P=cfg.num_workers,
-                    iters=num_iterations or cfg.num_iterations,
+        iters = num_iterations or cfg.num_iterations
+        snap_ring = None
+        snap_cap = 0
+        if snapshot_every > 0:
+            snap_cap = iters // snapshot_every + 2
+            snap_ring = torch.zeros(snap_cap, cfg.d, dtype=torch.float32,
+                                    device=self.device)
+        conf = dict(N=cfg.N, d=cfg.d, P=cfg.num_workers, iters=iters,
                     gamma=cfg.gamma, rate=cfg.batch_rate,
                     bucket_ratio=cfg.bucket_ratio, taw=cfg.taw,
                     seed=cfg.seed, algo=_ALGO[cfg.algo],
                     objective=_OBJ[cfg.objective], coeff=cfg.delay_coeff,
                     calib_window=cfg.calib_factor * cfg.num_workers,
                     mark_lo=mark_lo, mark_hi=mark_hi,
-                    max_wall_s=max_wall_s)
+                    max_wall_s=max_wall_s, snap_every=snapshot_every,
+                    snap_ring=snap_ring.data_ptr() if snap_ring is not None
+                    else 0, snap_cap=snap_cap)
         torch.cuda.synchronize()
         res = self._core.native_local_run(conf, self._bufs,
                                           self.w.data_ptr(),
                                           self.alpha_bar.data_ptr())
         torch.cuda.synchronize()
+        if snapshot_every > 0:
+            snaps = res["snap_ms"]
+            res["opt_vars"] = [(0, torch.zeros(cfg.d))] + [
+                (int(ms), snap_ring[i].cpu().clone())
+                for i, ms in enumerate(snaps)]
         return res
 
     def bench(self, warmup: int, steps: int,

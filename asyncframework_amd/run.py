@@ -78,7 +78,27 @@ def build_csr_workers(cfg: EngineConfig, indptr, indices, values, y,
 
 def run_engine(cfg: EngineConfig, workers: List[Worker],
                max_wall_s: Optional[float] = None,
-               verbose: bool = True) -> Tuple[RunResult, Server]:
+               verbose: bool = True, engine: str = "threads"):
+    """engine='threads' (the Python mailbox engine — CPU + semantics
+    oracle) or 'native' (the C++ event loop, GPU only, async only)."""
+    if engine == "native":
+        from .engine.native import NativeLocalEngine
+        assert not cfg.sync, "native engine is async-only"
+        neng = NativeLocalEngine(cfg, [w.shard for w in workers],
+                                 workers[0].device)
+        nres = neng.run(max_wall_s=max_wall_s or 1800.0,
+                        snapshot_every=(cfg.printer_freq
+                                        if cfg.snapshot_weights else 0))
+        if verbose:
+            for i in range(0, nres["k"], cfg.printer_freq):
+                print(f"Iteration {i} is finished")
+        waiting = {i: int(ms) for i, ms in enumerate(nres["waiting_ms"])}
+        res = RunResult(k=int(nres["k"]), elapsed_ms=int(nres["elapsed_ms"]),
+                        opt_vars=nres.get("opt_vars", []),
+                        waiting_time=waiting, w=neng.w,
+                        applied=int(nres["applied"]),
+                        rejected=int(nres["rejected"]))
+        return res, neng
     server = Server(cfg, device=workers[0].device)
     delay = DelayInjector(cfg.num_workers, cfg.delay_coeff, cfg.seed,
                           calib_window=cfg.calib_factor * cfg.num_workers)

@@ -59,6 +59,9 @@ void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
 void launch_saga_update(float*, const float*, float*, float, float, float,
                         int, hipStream_t);
 void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
+void launch_sgd_update_zero(float*, float*, float, float, int, hipStream_t);
+void launch_saga_update_zero(float*, float*, float*, float, float, float,
+                             int, hipStream_t);
 }
 
 namespace {
@@ -116,6 +119,7 @@ struct WorkerBuf {
   int pending_n = 0;
   double submit_t = 0, finish_t = 0, waiting_ms = 0;
   double poll_after = 0;  // don't hipEventQuery before this time
+  bool g_dirty = false;   // gradient buffer holds a rejected round's sums
   long tasks = 0;
 };
 
@@ -132,6 +136,9 @@ struct EngineCfg {
   long calib_window = 0;
   long mark_lo = -1, mark_hi = -1;
   double max_wall_s = 3600.0;
+  long snap_every = 0;       // optVars cadence (reference printer_freq)
+  uintptr_t snap_ring = 0;   // [snap_cap][d] device ring
+  long snap_cap = 0;
 };
 
 double now_s() {
@@ -161,6 +168,8 @@ struct NativeEngine {
   std::vector<int> straggler_kind;  // 0 none, 1 normal, 2 longtail
   double mark_lo_t = 0, mark_hi_t = 0;
   double ewma_round_s = 0;  // poll gating
+  std::vector<double> snap_ms;  // host stamps for the optVars snapshots
+  double run_t0 = 0;
 
   void init_stragglers() {
     straggler_kind.assign(cfg.P, 0);
@@ -202,9 +211,13 @@ struct NativeEngine {
   }
 
   void launch_grad(WorkerBuf& wk, long round_key) {
-    // zero gradient (+ SAGA staging counters), then the fused kernel;
-    // ASGD never reads the sampled count, so its ctr memset is skipped
-    HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wk.stream));
+    // g is zeroed by the fused update of the worker's ACCEPTED previous
+    // round; only a rejected round leaves it dirty. ASGD never reads the
+    // sampled count, so its ctr memset is skipped.
+    if (wk.g_dirty) {
+      HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wk.stream));
+      wk.g_dirty = false;
+    }
     if (cfg.algo == 1)
       HIP_CHECK(hipMemsetAsync((void*)wk.ctr, 0, 8, wk.stream));
     if (cfg.algo == 1) {
@@ -311,18 +324,29 @@ struct NativeEngine {
         cul_time_ms += (t_now - wk.submit_t) * 1000.0;
         cul_count += 1;
       }
+      // optVars snapshot (reference SparkASGDThread.scala:195-198: at the
+      // pre-increment k, every printer_freq applied updates)
+      if (cfg.snap_every > 0 && k % cfg.snap_every == 0 &&
+          (long)snap_ms.size() < cfg.snap_cap) {
+        HIP_CHECK(hipMemcpyAsync(
+            (void*)(cfg.snap_ring + (uintptr_t)snap_ms.size() *
+                                        (size_t)cfg.d * 4),
+            (const void*)w, (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
+            sstream));
+        snap_ms.push_back((t_now - run_t0) * 1000.0);
+      }
       const double par_recs = cfg.rate * (double)cfg.N / cfg.P;
       if (cfg.algo == 1) {
-        launch_saga_update((float*)w, (const float*)wk.g, (float*)alpha_bar,
-                           (float)cfg.gamma, (float)(1.0 / par_recs),
-                           (float)(1.0 / cfg.N), cfg.d, sstream);
+        launch_saga_update_zero((float*)w, (float*)wk.g, (float*)alpha_bar,
+                                (float)cfg.gamma, (float)(1.0 / par_recs),
+                                (float)(1.0 / cfg.N), cfg.d, sstream);
         wk.pending_commit = true;
         wk.pending_n = -1;  // resolved lazily from ctr at commit time
       } else {
         const double gamma_k =
             cfg.gamma / std::sqrt((double)(k / cfg.P + 1));
-        launch_sgd_update((float*)w, (const float*)wk.g, (float)gamma_k,
-                          (float)(1.0 / par_recs), cfg.d, sstream);
+        launch_sgd_update_zero((float*)w, (float*)wk.g, (float)gamma_k,
+                               (float)(1.0 / par_recs), cfg.d, sstream);
       }
       HIP_CHECK(hipGetLastError());
       HIP_CHECK(hipEventRecord(update_ev, sstream));
@@ -339,6 +363,7 @@ struct NativeEngine {
     } else {
       rejected += 1;
       wk.pending_commit = false;
+      wk.g_dirty = true;  // dispatch() re-zeroes before the next round
     }
     pendingq.push_back(wid);
   }
@@ -355,6 +380,7 @@ struct NativeEngine {
     long k = 0, applied = 0, rejected = 0, max_staleness = -1;
     double elapsed_ms = 0, mark_lo_t = 0, mark_hi_t = 0, avg_delay = 0;
     std::vector<double> waits;
+    std::vector<double> snap_ms;
   };
 
   Result run() {
@@ -368,6 +394,7 @@ struct NativeEngine {
     init_stragglers();
     for (int i = 0; i < cfg.P; ++i) pendingq.push_back(i);
     const double t0 = now_s();
+    run_t0 = t0;
     // first dispatch ignores the gate (reference k==0 path)
     {
       const size_t qn = pendingq.size();
@@ -421,6 +448,7 @@ struct NativeEngine {
     out.mark_lo_t = mark_lo_t;
     out.mark_hi_t = mark_hi_t;
     out.avg_delay = avg_delay_ms;
+    out.snap_ms = snap_ms;
     for (auto& wk : ws) out.waits.push_back(wk.waiting_ms);
     for (auto& wk : ws) {
       HIP_CHECK(hipEventDestroy(wk.done));
@@ -456,6 +484,9 @@ void register_native_engine(py::module_& m) {
         cfg.mark_lo = py::cast<long>(c["mark_lo"]);
         cfg.mark_hi = py::cast<long>(c["mark_hi"]);
         cfg.max_wall_s = py::cast<double>(c["max_wall_s"]);
+        cfg.snap_every = py::cast<long>(c["snap_every"]);
+        cfg.snap_ring = py::cast<uintptr_t>(c["snap_ring"]);
+        cfg.snap_cap = py::cast<long>(c["snap_cap"]);
         for (auto item : workers) {
           py::dict wd = py::cast<py::dict>(item);
           WorkerBuf wk;
@@ -500,6 +531,7 @@ void register_native_engine(py::module_& m) {
         out["mark_hi_t"] = r.mark_hi_t;
         out["avg_delay_ms"] = r.avg_delay;
         out["waiting_ms"] = r.waits;
+        out["snap_ms"] = r.snap_ms;
         return out;
       });
 }

@@ -103,3 +103,23 @@ def test_native_bench_marks():
     elapsed, res = eng.bench(warmup=50, steps=200, max_wall_s=120)
     assert elapsed > 0
     assert res["k"] >= 250
+
+
+def test_native_cli_driver_contract(capsys):
+    """asgd-thread --engine native: full stdout contract through the C++
+    event loop (snapshots, waiting times, objective sweep)."""
+    from asyncframework_amd.cli import drivers
+    drivers.asgd_thread(["synthetic", "synthetic", "64", "20000", "4", "60",
+                         "0.3", "1000000", "0.05", "0.5", "20", "0", "42",
+                         "--device", "cuda:0", "--engine", "native"])
+    out = capsys.readouterr().out
+    lines = out.splitlines()
+    assert lines[-1] == "finished"
+    assert any(l.startswith("Iteration ") for l in lines)
+    assert "Individual waiting times:" in lines
+    import re
+    sep = max(i for i, l in enumerate(lines) if l.startswith("*********"))
+    csv = [l for l in lines[sep + 1:] if re.match(r"^\d+,[0-9.eE+-]+$", l)]
+    assert len(csv) >= 2
+    objs = [float(l.split(",")[1]) for l in csv]
+    assert objs[-1] < objs[0]
