@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import queue
 import threading
-from typing import Any, Dict, Generic, Optional, TypeVar
+from typing import Dict, Generic, Optional, TypeVar
 
 T = TypeVar("T")
 

@@ -8,7 +8,7 @@ LibSVM lines are ``label idx:val idx:val ...`` with 1-based indices.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import numpy as np
 import torch

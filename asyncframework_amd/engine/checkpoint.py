@@ -15,7 +15,6 @@ from typing import Dict, List, Optional
 
 import torch
 
-from .config import EngineConfig
 from .server import Server
 from .worker import Worker
 

@@ -3,7 +3,7 @@ MI355X engine flags."""
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

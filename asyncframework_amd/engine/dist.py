@@ -29,8 +29,8 @@ import torch.distributed as dist
 from .config import EngineConfig
 from .delay import DelayInjector
 from .local import AsyncEngine, RunResult, SyncEngine, _LocalChannel
-from .messages import (HDR, Dispatch, WorkerResult, pack_dispatch,
-                       pack_result, unpack_dispatch, unpack_result)
+from .messages import (HDR, Dispatch, pack_dispatch, pack_result,
+                       unpack_dispatch, unpack_result)
 from .server import Server
 from .worker import Worker
 

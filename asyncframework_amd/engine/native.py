@@ -11,7 +11,6 @@ stays for CPU tests, host-spill history, and as the semantics oracle)."""
 
 from __future__ import annotations
 
-import time
 from typing import Dict, List, Optional, Tuple
 
 import torch

@@ -6,13 +6,13 @@ compatible")."""
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 
 from . import ops
 from .data.shard import row_shards
-from .data.synthetic import SHAPES, synthetic_csr, synthetic_dense
+from .data.synthetic import synthetic_csr, synthetic_dense
 from .data.libsvm import load_libsvm
 from .engine.config import EngineConfig
 from .engine.delay import DelayInjector
