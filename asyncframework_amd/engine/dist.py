@@ -34,6 +34,24 @@ from .messages import (HDR, Dispatch, pack_dispatch, pack_result,
 from .server import Server
 from .worker import Worker
 
+# ProcessGroup op ENQUEUES are serialized (torch's NCCL bindings are not
+# documented thread-safe); Work.wait() runs outside the lock so peers still
+# progress concurrently. Each pair group has its own communicator, so
+# ordering across groups is unconstrained.
+_PG_LOCK = threading.Lock()
+
+
+def _send(buf, dst, group):
+    with _PG_LOCK:
+        req = dist.isend(buf, dst=dst, group=group)
+    req.wait()
+
+
+def _recv(buf, src, group):
+    with _PG_LOCK:
+        req = dist.irecv(buf, src=src, group=group)
+    req.wait()
+
 
 class _RemoteChannel:
     """Rank-0-side proxy for one remote worker: a dedicated thread that
@@ -78,10 +96,10 @@ class _RemoteChannel:
                     continue
                 msg = self._q.pop(0)
                 pack_dispatch(self._sendbuf, self.d, msg)
-                dist.send(self._sendbuf, dst=self.peer, group=self.group)
+                _send(self._sendbuf, self.peer, self.group)
                 if msg.stop:
                     break
-                dist.recv(self._recvbuf, src=self.peer, group=self.group)
+                _recv(self._recvbuf, self.peer, self.group)
                 if self.stream is not None:
                     self.stream.synchronize()
                 res = unpack_result(self._recvbuf, self.d, self.peer)
@@ -100,13 +118,13 @@ def remote_worker_loop(worker: Worker, cfg: EngineConfig, group,
     buf = torch.zeros(d + HDR, dtype=torch.float32, device=device)
     out = torch.zeros(d + HDR, dtype=torch.float32, device=device)
     while True:
-        dist.recv(buf, src=0, group=group)
+        _recv(buf, 0, group)
         msg = unpack_dispatch(buf, d)  # .tolist() syncs the stream
         if msg.stop:
             break
         res = worker.process(msg)
         pack_result(out, d, res)
-        dist.send(out, dst=0, group=group)
+        _send(out, 0, group)
 
 
 class DistEngine:
