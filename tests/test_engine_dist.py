@@ -80,3 +80,45 @@ def test_dist_async_asaga_gloo():
     k, obj0, obj1 = _run_dist(sync=False, algo="asaga")
     assert k >= 40
     assert obj1 < obj0
+
+
+WORLD3 = 3
+
+
+def _rank3_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD3)
+    try:
+        cfg = EngineConfig(d=16, N=300, num_workers=WORLD3,
+                           num_iterations=45, gamma=0.4, taw=2 ** 30,
+                           batch_rate=0.3, bucket_ratio=0.6,
+                           printer_freq=1000, delay_coeff=0.0, seed=4,
+                           device="cpu", sync=False, algo="asgd",
+                           snapshot_weights=False)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=21)
+        s, t = row_shards(cfg.N, WORLD3)[rank]
+        worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X[s:t],
+                                    y=y[s:t]), cfg,
+                        device=torch.device("cpu"))
+        eng = DistEngine(cfg, worker, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=120)
+        if rank == 0:
+            obj0 = float(((X @ torch.zeros(cfg.d) - y) ** 2).mean())
+            obj1 = float(((X @ res.w - y) ** 2).mean())
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{obj0},{obj1}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dist_three_ranks_gloo():
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_rank3_main, args=(init_file, out_file), nprocs=WORLD3,
+                 join=True)
+        with open(out_file) as f:
+            k, obj0, obj1 = f.read().split(",")
+        assert int(k) >= 45
+        assert float(obj1) < float(obj0)

@@ -100,3 +100,21 @@ def test_graph_csr_runs_and_decreases():
                                                                  device="cuda:0"),
                                                      eng.w]), "lsq")
     assert float(obj[1]) < float(obj[0])
+
+
+def test_graph_overlap_odd_tail_parity():
+    """The overlap graph is captured for buffer parity cur==0 with an even
+    unroll; odd-length step_rounds calls must re-align via tail rounds and
+    still match the sequential reference."""
+    cfg = _cfg(num_iterations=57)  # not a multiple of unroll, odd tails
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=11, device="cuda:0")
+    eng = GraphEngine(cfg, Shard(row_start=0, n_rows=cfg.N, X=X, y=y),
+                      torch.device("cuda:0"), unroll=10)
+    eng.step_rounds(3)      # odd: forces tail rounds + parity realign
+    eng.step_rounds(24)
+    eng.step_rounds(30)
+    torch.cuda.synchronize()
+    assert eng.k == 57
+    w_ref = _seq_ref_asgd(cfg, X, y)
+    rel = float((eng.w - w_ref).norm() / (w_ref.norm() + 1e-12))
+    assert rel < 1e-4, rel
