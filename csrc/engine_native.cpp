@@ -267,6 +267,7 @@ struct NativeEngine {
     }
     wk.pending_commit = false;
     wk.pending_n = 0;
+    if (wk.finish_t == 0) wk.finish_t = t_now;  // first dispatch: no wait
     wk.waiting_ms += (t_now - wk.finish_t) * 1000.0;
     wk.submit_t = t_now;
     wk.busy = true;
