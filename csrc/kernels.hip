@@ -1009,10 +1009,6 @@ void launch_grad_dense_list(const void* X, const float* w, float* g_part,
     else if (depth == 8) LIST_LAUNCH_D(XT, CAST, IT, 8);                     \
     else LIST_LAUNCH_D(XT, CAST, IT, 4);                                     \
   } while (0)
-#define LIST_LAUNCH_OLD(XT, CAST, IT)                                        \
-  hipLaunchKernelGGL((grad_dense_list_kernel<XT, 256, 4, IT>), dim3(grid),   \
-                     dim3(256), smem, stream, (CAST)X, w, g_part, rowlist,   \
-                     ylist, count_dev, n_rows, d, objective)
 #define LIST_DISPATCH(XT, CAST)                                              \
   do {                                                                       \
     switch (iters) {                                                         \
@@ -1030,7 +1026,6 @@ void launch_grad_dense_list(const void* X, const float* w, float* g_part,
   else LIST_DISPATCH(float, const float*);
 #undef LIST_DISPATCH
 #undef LIST_LAUNCH
-#undef LIST_LAUNCH_OLD
 #undef LIST_LAUNCH_D
 }
 
