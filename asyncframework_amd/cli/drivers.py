@@ -44,6 +44,14 @@ def _engine_flags(p: argparse.ArgumentParser) -> None:
     p.add_argument("--engine", default="threads",
                    choices=["threads", "native"],
                    help="native = C++ event-loop engine (GPU, async only)")
+    p.add_argument("--worker-timeout-s", type=float, default=0.0,
+                   help="declare a busy worker dead after this many seconds "
+                        "(0 = off; the reference leaves lost workers busy "
+                        "forever)")
+    p.add_argument("--checkpoint-path", default="",
+                   help="periodic optimizer-state snapshot file")
+    p.add_argument("--checkpoint-every", type=int, default=0,
+                   help="checkpoint every N applied updates (0 = off)")
 
 
 def _parse13(argv: List[str], prog: str):
@@ -74,7 +82,10 @@ def _cfg13(a, algo: str, sync: bool) -> EngineConfig:
         bucket_ratio=float(a.bucketRatio), printer_freq=int(a.printerFreq),
         delay_coeff=float(a.coeff), seed=int(a.seed), algo=algo, sync=sync,
         objective=a.objective, dtype=a.dtype, device=a.device,
-        history_placement=a.history_placement)
+        history_placement=a.history_placement,
+        worker_timeout_s=a.worker_timeout_s,
+        checkpoint_path=a.checkpoint_path,
+        checkpoint_every=a.checkpoint_every)
 
 
 def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:

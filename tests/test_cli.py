@@ -93,3 +93,14 @@ def test_objective_decreases_in_output():
     objs = [float(l.split(",")[1]) for l in csv]
     assert len(objs) >= 2
     assert objs[-1] < objs[0]
+
+
+def test_cli_checkpoint_flag(tmp_path):
+    ck = str(tmp_path / "run.ckpt")
+    args = ARGS13 + ["--checkpoint-path", ck, "--checkpoint-every", "10"]
+    _capture(drivers.asgd_thread, args)
+    import os
+    assert os.path.exists(ck)
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    state = load_checkpoint(ck)
+    assert state["k"] >= 10
