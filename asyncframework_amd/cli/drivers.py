@@ -52,6 +52,9 @@ def _engine_flags(p: argparse.ArgumentParser) -> None:
                    help="periodic optimizer-state snapshot file")
     p.add_argument("--checkpoint-every", type=int, default=0,
                    help="checkpoint every N applied updates (0 = off)")
+    p.add_argument("--resume-from", default="",
+                   help="restore optimizer state from a checkpoint file "
+                        "before running")
 
 
 def _parse13(argv: List[str], prog: str):
@@ -98,7 +101,7 @@ def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:
     else:
         workers = runner.build_dense_workers(cfg, *data)
     res, _srv = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s,
-                                  engine=a.engine)
+                                  engine=a.engine, resume_from=a.resume_from)
     runner.final_report(cfg, res, data, sparse, device=a.device)
 
 

@@ -78,10 +78,13 @@ def build_csr_workers(cfg: EngineConfig, indptr, indices, values, y,
 
 def run_engine(cfg: EngineConfig, workers: List[Worker],
                max_wall_s: Optional[float] = None,
-               verbose: bool = True, engine: str = "threads"):
+               verbose: bool = True, engine: str = "threads",
+               resume_from: str = ""):
     """engine='threads' (the Python mailbox engine — CPU + semantics
-    oracle) or 'native' (the C++ event loop, GPU only, async only)."""
+    oracle) or 'native' (the C++ event loop, GPU only, async only).
+    ``resume_from`` restores a checkpoint (threads engine only)."""
     if engine == "native":
+        assert not resume_from, "resume is a threads-engine feature"
         from .engine.native import NativeLocalEngine
         assert not cfg.sync, "native engine is async-only"
         neng = NativeLocalEngine(cfg, [w.shard for w in workers],
@@ -100,6 +103,9 @@ def run_engine(cfg: EngineConfig, workers: List[Worker],
                         rejected=int(nres["rejected"]))
         return res, neng
     server = Server(cfg, device=workers[0].device)
+    if resume_from:
+        from .engine.checkpoint import load_checkpoint, restore
+        restore(server, workers, load_checkpoint(resume_from))
     delay = DelayInjector(cfg.num_workers, cfg.delay_coeff, cfg.seed,
                           calib_window=cfg.calib_factor * cfg.num_workers)
     eng_cls = SyncEngine if cfg.sync else AsyncEngine

@@ -104,3 +104,15 @@ def test_cli_checkpoint_flag(tmp_path):
     from asyncframework_amd.engine.checkpoint import load_checkpoint
     state = load_checkpoint(ck)
     assert state["k"] >= 10
+
+
+def test_cli_resume_from_checkpoint(tmp_path):
+    ck = str(tmp_path / "resume.ckpt")
+    _capture(drivers.asgd_thread,
+             ARGS13 + ["--checkpoint-path", ck, "--checkpoint-every", "10"])
+    # resume with a larger iteration budget: the run continues from the
+    # checkpointed k rather than restarting
+    args = list(ARGS13)
+    args[5] = "60"  # numIter
+    out = _capture(drivers.asgd_thread, args + ["--resume-from", ck])
+    assert "finished" in out
