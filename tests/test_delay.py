@@ -51,3 +51,11 @@ def test_coeff_zero_disables():
     inj = DelayInjector(4, coeff=0.0, calib_window=0)
     inj.maybe_activate(1)
     assert inj.delay_ms(0, 1) == 0.0
+
+
+def test_activation_without_samples_keeps_zero_delay():
+    inj = DelayInjector(4, coeff=-1.0, calib_window=0)
+    inj.maybe_activate(1)  # no recorded tasks
+    assert inj.flag
+    assert inj.avg_delay_ms == 0.0
+    assert inj.delay_ms(0, 1) == 0.0
