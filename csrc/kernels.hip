@@ -3,7 +3,12 @@
 // Kernel inventory (reference JVM hot loops they replace):
 //   K1 grad_dense      — fused Philox sample mask + per-row dot + scaled
 //                        accumulate (reference gradfun SparkASGDThread.scala:
-//                        423-438 + reducePartition fold RDD.scala:1103-1123).
+//                        423-438 + reducePartition fold RDD.scala:1103-1123);
+//                        three forms: grad_dense_pipe_kernel (LDS row queue
+//                        + depth-4 register pipeline), scan_rows_kernel +
+//                        grad_dense_list_kernel (split form for the graph
+//                        engine's scan/compute overlap), and a generic
+//                        fallback for d > 2048 or d % 4 != 0.
 //   K2 grad_csr        — CSR SpMV-style gradient, wave-per-row
 //                        (reference sparse BLAS.scala:74-90,134-160).
 //   K3 saga_grad_*     — K1/K2 fused with the per-sample history gather and
