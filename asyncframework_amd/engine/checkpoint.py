@@ -19,9 +19,12 @@ from .server import Server
 from .worker import Worker
 
 
-def capture_state(server: Server, workers: Optional[List[Worker]] = None
+def capture_state(server: Server, workers: Optional[List[Worker]] = None,
+                  remote_alpha: Optional[Dict[int, torch.Tensor]] = None
                   ) -> Dict:
-    """Snapshot everything needed to resume (host-resident tensors)."""
+    """Snapshot everything needed to resume (host-resident tensors).
+    ``remote_alpha`` carries history tables gathered from remote ranks by
+    the dist engine's snap sideband (engine/dist.py) keyed by worker id."""
     state = {
         "k": server.k,
         "current_time": server.AC.getCurrentTime(),
@@ -34,14 +37,18 @@ def capture_state(server: Server, workers: Optional[List[Worker]] = None
     for wk in workers or []:
         if wk.alpha is not None:
             state["alpha"][wk.id] = wk.alpha.detach().cpu().clone()
+    for wid, t in (remote_alpha or {}).items():
+        state["alpha"][wid] = t.detach().cpu().clone()
     return state
 
 
 def save_checkpoint(path: str, server: Server,
-                    workers: Optional[List[Worker]] = None) -> None:
+                    workers: Optional[List[Worker]] = None,
+                    remote_alpha: Optional[Dict[int, torch.Tensor]] = None
+                    ) -> None:
     """Atomic write (tmp + rename) so a crash mid-save keeps the previous
     checkpoint valid."""
-    state = capture_state(server, workers)
+    state = capture_state(server, workers, remote_alpha)
     d = os.path.dirname(os.path.abspath(path))
     os.makedirs(d, exist_ok=True)
     fd, tmp = tempfile.mkstemp(dir=d, suffix=".ckpt.tmp")

@@ -60,13 +60,23 @@ class _RemoteChannel:
     reference TaskResultGetter.scala:57)."""
 
     def __init__(self, peer_rank: int, group, server: Server,
-                 cfg: EngineConfig, device: torch.device):
+                 cfg: EngineConfig, device: torch.device,
+                 alpha_rows: int = 0):
         self.peer = peer_rank
         self.group = group
         self.server = server
         self.cfg = cfg
         self.device = device
         self.d = cfg.d
+        # SAGA history sideband (checkpointing): the peer's shard size under
+        # the canonical row_shards(N, P) layout — the dist engine's sharding
+        # contract (run.py builders and bench use the same row_shards)
+        self.alpha_rows = alpha_rows
+        self._alpha_buf = (torch.zeros(alpha_rows, dtype=torch.float32,
+                                       device=device)
+                           if alpha_rows > 0 else None)
+        self._snap_done = threading.Event()
+        self.alpha_snapshot = None
         self._q: "list[Dispatch]" = []
         self._ev = threading.Event()
         self._sendbuf = torch.zeros(cfg.d + HDR, dtype=torch.float32,
@@ -99,6 +109,15 @@ class _RemoteChannel:
                 _send(self._sendbuf, self.peer, self.group)
                 if msg.stop:
                     break
+                if msg.snap == 1:
+                    # checkpoint sideband: the peer replies with its SAGA
+                    # history table instead of a gradient
+                    _recv(self._alpha_buf, self.peer, self.group)
+                    if self.stream is not None:
+                        self.stream.synchronize()
+                    self.alpha_snapshot = self._alpha_buf.detach().cpu().clone()
+                    self._snap_done.set()
+                    continue
                 _recv(self._recvbuf, self.peer, self.group)
                 if self.stream is not None:
                     self.stream.synchronize()
@@ -108,6 +127,29 @@ class _RemoteChannel:
 
     def join(self, timeout=None):
         self.thread.join(timeout)
+
+    # -- SAGA history checkpoint sideband ------------------------------------
+    def request_alpha(self) -> None:
+        """Enqueue a snapshot request; the proxy thread serializes it after
+        any in-flight round (the strict dispatch->result alternation keeps
+        the pair protocol deadlock-free)."""
+        self._snap_done.clear()
+        self.dispatch(Dispatch(w=None, snap=1))
+
+    def wait_alpha(self, timeout: float = 60.0):
+        if self._snap_done.wait(timeout):
+            return self.alpha_snapshot
+        return None
+
+    def push_alpha(self, table: torch.Tensor) -> None:
+        """Resume path: send a restored history table to the peer. Must be
+        called BEFORE start() (the proxy thread is not running, so direct
+        sends on the pair group cannot interleave with a round)."""
+        assert not self.thread.is_alive()
+        pack_dispatch(self._sendbuf, self.d, Dispatch(w=None, snap=2))
+        _send(self._sendbuf, self.peer, self.group)
+        buf = table.to(dtype=torch.float32, device=self.device)
+        _send(buf, self.peer, self.group)
 
 
 def remote_worker_loop(worker: Worker, cfg: EngineConfig, group,
@@ -122,6 +164,18 @@ def remote_worker_loop(worker: Worker, cfg: EngineConfig, group,
         msg = unpack_dispatch(buf, d)  # .tolist() syncs the stream
         if msg.stop:
             break
+        if msg.snap == 1:  # checkpoint: reply with the SAGA history table
+            assert worker.alpha is not None, "snap on a non-SAGA worker"
+            _send(worker.alpha.to(dtype=torch.float32, device=device),
+                  0, group)
+            continue
+        if msg.snap == 2:  # resume: receive a restored history table
+            assert worker.alpha is not None, "snap on a non-SAGA worker"
+            ab = torch.zeros(worker.shard.n_rows, dtype=torch.float32,
+                             device=device)
+            _recv(ab, 0, group)
+            worker.alpha.copy_(ab.to(worker.alpha.device))
+            continue
         res = worker.process(msg)
         pack_result(out, d, res)
         _send(out, 0, group)
@@ -150,14 +204,27 @@ class DistEngine:
             self.pair_groups[i] = dist.new_group([0, i])
 
     def run(self, max_wall_s: Optional[float] = None,
-            verbose: bool = True) -> Optional[RunResult]:
+            verbose: bool = True,
+            resume_from: str = "") -> Optional[RunResult]:
         cfg = self.cfg
         if self.rank == 0:
+            from ..data.shard import row_shards
+            shards = row_shards(cfg.N, self.world)
             server = Server(cfg, device=self.device)
             channels: List = [_LocalChannel(self.local_worker, server)]
             for i in range(1, self.world):
-                channels.append(_RemoteChannel(i, self.pair_groups[i],
-                                               server, cfg, self.device))
+                s, t = shards[i]
+                channels.append(_RemoteChannel(
+                    i, self.pair_groups[i], server, cfg, self.device,
+                    alpha_rows=(t - s) if cfg.algo == "asaga" else 0))
+            if resume_from:
+                from .checkpoint import load_checkpoint, restore
+                state = load_checkpoint(resume_from)
+                restore(server, [self.local_worker], state)
+                # push remote ranks' history tables before any round starts
+                for i in range(1, self.world):
+                    if cfg.algo == "asaga" and i in state.get("alpha", {}):
+                        channels[i].push_alpha(state["alpha"][i])
             eng_cls = SyncEngine if cfg.sync else AsyncEngine
             eng = eng_cls(cfg, server=server, channels=channels,
                           delay=self.delay)

@@ -124,6 +124,24 @@ class AsyncEngine:
     def _local_workers(self):
         return [ch.worker for ch in self.channels if hasattr(ch, "worker")]
 
+    def _gather_remote_alpha(self):
+        """Pull SAGA history tables from remote ranks (dist engine snap
+        sideband). Requests fan out first so peers snapshot concurrently;
+        each table is a consistent point-in-time view of that worker's
+        commit-on-accept history (cross-worker skew is inherent to an async
+        checkpoint — same model as the in-process snapshot, which also
+        observes workers mid-round)."""
+        remotes = [(wid, ch) for wid, ch in enumerate(self.channels)
+                   if hasattr(ch, "request_alpha") and ch.alpha_rows > 0]
+        for _, ch in remotes:
+            ch.request_alpha()
+        out = {}
+        for wid, ch in remotes:
+            t = ch.wait_alpha()
+            if t is not None:
+                out[wid] = t
+        return out
+
     def _reap_dead_workers(self):
         """Failure detection the reference lacks (a lost task
         leaves a worker permanently busy): a worker whose round exceeds
@@ -178,7 +196,8 @@ class AsyncEngine:
                     and srv.k % cfg.checkpoint_every == 0):
                 from .checkpoint import save_checkpoint
                 save_checkpoint(cfg.checkpoint_path, srv,
-                                self._local_workers())
+                                self._local_workers(),
+                                remote_alpha=self._gather_remote_alpha())
         else:
             srv.last_accept[wid] = False
             self.pending.append(wid)

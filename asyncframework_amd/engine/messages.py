@@ -21,13 +21,17 @@ class Dispatch:
     (sampling seed round, reference ``sample(false,b,seed+k+1)``,
     SparkASGDThread.scala:314); ``accept_prev`` commits the worker's staged
     SAGA history from its previous round; ``delay_s`` the injected straggler
-    delay; ``stop`` ends the worker loop."""
+    delay; ``stop`` ends the worker loop; ``snap`` is a history-table
+    sideband op for the dist engine (1 = worker SENDS its SAGA alpha table
+    to the server for a checkpoint, 2 = worker RECEIVES a table to restore
+    on resume) — no gradient round runs for snap messages."""
     w: Optional[torch.Tensor]
     ts: int = 0
     k_submit: int = 0
     accept_prev: bool = True
     delay_s: float = 0.0
     stop: bool = False
+    snap: int = 0
 
 
 @dataclass
@@ -45,7 +49,7 @@ class WorkerResult:
 # Header layout for the packed wire format (dist engine). The payload tensor
 # is [d + HDR] floats: payload[:d] = w or g, payload[d:] = header.
 HDR = 8
-H_TS, H_K, H_ACCEPT, H_STOP, H_DELAY, H_NROWS, H_ELAPSED, H_PAD = range(8)
+H_TS, H_K, H_ACCEPT, H_STOP, H_DELAY, H_NROWS, H_ELAPSED, H_SNAP = range(8)
 
 
 def _hdr_tensor(vals) -> torch.Tensor:
@@ -59,7 +63,8 @@ def pack_dispatch(buf: torch.Tensor, d: int, msg: Dispatch) -> None:
     # tensor are one tiny H2D each — measured to dominate dispatch latency)
     hdr = _hdr_tensor([float(msg.ts), float(msg.k_submit),
                        1.0 if msg.accept_prev else 0.0,
-                       1.0 if msg.stop else 0.0, msg.delay_s, 0.0, 0.0, 0.0])
+                       1.0 if msg.stop else 0.0, msg.delay_s, 0.0, 0.0,
+                       float(msg.snap)])
     buf[d:].copy_(hdr, non_blocking=False)
 
 
@@ -67,7 +72,7 @@ def unpack_dispatch(buf: torch.Tensor, d: int) -> Dispatch:
     h = buf[d:].tolist()
     return Dispatch(w=buf[:d], ts=int(h[H_TS]), k_submit=int(h[H_K]),
                     accept_prev=h[H_ACCEPT] > 0.5, stop=h[H_STOP] > 0.5,
-                    delay_s=float(h[H_DELAY]))
+                    delay_s=float(h[H_DELAY]), snap=int(h[H_SNAP]))
 
 
 def pack_result(buf: torch.Tensor, d: int, res: WorkerResult) -> None:
