@@ -1,7 +1,10 @@
 """Multi-process engine: one process per GPU over torch.distributed.
 
 MI355X topology: rank 0 hosts the parameter server (master
-weights in its HBM) *and* worker 0; every other rank is a pure worker. Data
+weights in its HBM) *and* the first M workers; every other rank hosts M pure
+workers (M = num_workers / world_size — logical workers are decoupled from
+ranks, the reference's partitions != executors model; each worker has its
+own HIP stream and its own pair communicator). Data
 movement is point-to-point send/recv on per-pair process groups — under the
 "nccl" backend that is RCCL over xGMI (each server<->worker pair gets its own
 communicator, so the server can service all peers concurrently from per-peer
@@ -61,8 +64,12 @@ class _RemoteChannel:
 
     def __init__(self, peer_rank: int, group, server: Server,
                  cfg: EngineConfig, device: torch.device,
-                 alpha_rows: int = 0):
+                 alpha_rows: int = 0, wid: Optional[int] = None):
         self.peer = peer_rank
+        # global logical-worker id; with several workers per rank this is
+        # NOT the peer rank (results must be credited to the worker, not
+        # the process)
+        self.wid = peer_rank if wid is None else wid
         self.group = group
         self.server = server
         self.cfg = cfg
@@ -121,7 +128,7 @@ class _RemoteChannel:
                 _recv(self._recvbuf, self.peer, self.group)
                 if self.stream is not None:
                     self.stream.synchronize()
-                res = unpack_result(self._recvbuf, self.d, self.peer)
+                res = unpack_result(self._recvbuf, self.d, self.wid)
                 res.g = self._recvbuf[:self.d].clone()
                 self.server.on_completion(res)
 
@@ -184,57 +191,95 @@ def remote_worker_loop(worker: Worker, cfg: EngineConfig, group,
 class DistEngine:
     """Orchestrates the multi-process run. Call from every rank with an
     initialized default process group; returns a RunResult on rank 0 and
-    None elsewhere."""
+    None elsewhere.
 
-    def __init__(self, cfg: EngineConfig, local_worker: Worker,
-                 device: torch.device,
+    Logical workers are decoupled from ranks (the reference's partitions !=
+    executors model): each rank hosts M = num_workers / world_size workers,
+    global worker id = rank * M + j. Every remote worker gets its OWN pair
+    process group (own communicator), so a rank with several workers
+    services them concurrently from independent proxy/worker threads with
+    no cross-channel ordering constraints — on GPU each worker also has its
+    own HIP stream, so co-located rounds overlap. Sharding contract:
+    ``row_shards(cfg.N, cfg.num_workers)``, worker wid owns shard wid."""
+
+    def __init__(self, cfg: EngineConfig, local_workers, device: torch.device,
                  delay: Optional[DelayInjector] = None):
         assert dist.is_initialized()
         self.cfg = cfg
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
-        assert cfg.num_workers == self.world, \
-            "dist mode: one worker per rank (num_workers == world_size)"
+        if isinstance(local_workers, Worker):
+            local_workers = [local_workers]
+        self.local_workers: List[Worker] = local_workers
+        assert cfg.num_workers % self.world == 0, \
+            "num_workers must be a multiple of world_size"
+        self.M = cfg.num_workers // self.world
+        assert len(local_workers) == self.M, \
+            f"rank {self.rank}: expected {self.M} local workers"
         self.device = device
-        self.local_worker = local_worker
         self.delay = delay
-        # pair groups: every rank participates in every new_group call
+        # per-REMOTE-WORKER pair groups; every rank participates in every
+        # new_group call (collective), same order on all ranks
         self.pair_groups = {}
-        for i in range(1, self.world):
-            self.pair_groups[i] = dist.new_group([0, i])
+        for wid in range(self.M, cfg.num_workers):
+            self.pair_groups[wid] = dist.new_group([0, wid // self.M])
+
+    # -- rank-0 construction (exposed so bench.py can set marks) -------------
+    def build_engine(self):
+        """Rank 0 only: construct (engine, server, channels), not started."""
+        from ..data.shard import row_shards
+        cfg = self.cfg
+        shards = row_shards(cfg.N, cfg.num_workers)
+        server = Server(cfg, device=self.device)
+        channels: List = [_LocalChannel(w, server)
+                          for w in self.local_workers]
+        for wid in range(self.M, cfg.num_workers):
+            s, t = shards[wid]
+            channels.append(_RemoteChannel(
+                wid // self.M, self.pair_groups[wid], server, cfg,
+                self.device,
+                alpha_rows=(t - s) if cfg.algo == "asaga" else 0,
+                wid=wid))
+        eng_cls = SyncEngine if cfg.sync else AsyncEngine
+        eng = eng_cls(cfg, server=server, channels=channels,
+                      delay=self.delay)
+        return eng, server, channels
+
+    def worker_loop(self) -> None:
+        """Non-zero ranks: one thread per hosted worker (each on its own
+        pair group; on GPU each Worker owns its own HIP stream)."""
+        threads = []
+        for j, w in enumerate(self.local_workers):
+            wid = self.rank * self.M + j
+            th = threading.Thread(
+                target=remote_worker_loop,
+                args=(w, self.cfg, self.pair_groups[wid], self.device),
+                daemon=True, name=f"rworker-{wid}")
+            th.start()
+            threads.append(th)
+        for th in threads:
+            th.join()
 
     def run(self, max_wall_s: Optional[float] = None,
             verbose: bool = True,
             resume_from: str = "") -> Optional[RunResult]:
         cfg = self.cfg
         if self.rank == 0:
-            from ..data.shard import row_shards
-            shards = row_shards(cfg.N, self.world)
-            server = Server(cfg, device=self.device)
-            channels: List = [_LocalChannel(self.local_worker, server)]
-            for i in range(1, self.world):
-                s, t = shards[i]
-                channels.append(_RemoteChannel(
-                    i, self.pair_groups[i], server, cfg, self.device,
-                    alpha_rows=(t - s) if cfg.algo == "asaga" else 0))
+            eng, server, channels = self.build_engine()
             if resume_from:
                 from .checkpoint import load_checkpoint, restore
                 state = load_checkpoint(resume_from)
-                restore(server, [self.local_worker], state)
+                restore(server, self.local_workers, state)
                 # push remote ranks' history tables before any round starts
-                for i in range(1, self.world):
-                    if cfg.algo == "asaga" and i in state.get("alpha", {}):
-                        channels[i].push_alpha(state["alpha"][i])
-            eng_cls = SyncEngine if cfg.sync else AsyncEngine
-            eng = eng_cls(cfg, server=server, channels=channels,
-                          delay=self.delay)
+                for wid in range(self.M, cfg.num_workers):
+                    if cfg.algo == "asaga" and wid in state.get("alpha", {}):
+                        channels[wid].push_alpha(state["alpha"][wid])
             eng.verbose = verbose
             res = eng.run(max_wall_s=max_wall_s)
             dist.barrier()
             return res
         else:
-            remote_worker_loop(self.local_worker, cfg,
-                               self.pair_groups[self.rank], self.device)
+            self.worker_loop()
             dist.barrier()
             return None
 

@@ -214,31 +214,36 @@ def run_single(args, device):
 
 
 def run_dist(args, device, rank, world):
+    """N>1: one process per GPU over RCCL. Logical workers stay at the
+    preset's count (the reference's FIXED partitions=32 model): each rank
+    hosts M = preset_workers/world workers on independent HIP streams and
+    per-worker pair communicators, so the whole-node worker pool (and the
+    quorum-gate semantics) is identical at every N — only the hardware
+    underneath it scales."""
     import torch.distributed as dist
-    from asyncframework_amd.engine.dist import (DistEngine, _RemoteChannel,
-                                                remote_worker_loop)
-    from asyncframework_amd.engine.local import _LocalChannel
-    cfg = make_cfg(args, world, device)
+    from asyncframework_amd.engine.dist import DistEngine
+    M = max(1, (args.preset_workers or world) // world)
+    P = M * world
+    cfg = make_cfg(args, P, device)
     dt = cfg.torch_dtype()
-    s, t = row_shards(args.rows, world)[rank]
-    X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + rank,
-                           dtype=dt, device=device,
-                           objective=args.objective)
-    worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X, y=y), cfg,
-                    device=device)
+    shards = row_shards(args.rows, P)
+    workers = []
+    for j in range(M):
+        wid = rank * M + j
+        s, t = shards[wid]
+        X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + wid,
+                               dtype=dt, device=device,
+                               objective=args.objective)
+        workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s, X=X,
+                                         y=y), cfg, device=device))
     dist.init_process_group("nccl" if device.type == "cuda" else "gloo",
                             rank=rank, world_size=world)
     dist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
-    deng = DistEngine(cfg, worker, device)
+    deng = DistEngine(cfg, workers, device)
     if rank == 0:
-        server = Server(cfg, device=device)
-        channels = [_LocalChannel(worker, server)]
-        for i in range(1, world):
-            channels.append(_RemoteChannel(i, deng.pair_groups[i], server,
-                                           cfg, device))
-        eng = AsyncEngine(cfg, server=server, channels=channels)
+        eng, server, channels = deng.build_engine()
         eng.verbose = False
         eng.mark_at = {args.warmup, args.warmup + args.steps}
         eng.run(max_wall_s=1800)
@@ -252,7 +257,7 @@ def run_dist(args, device, rank, world):
         else:
             emit(args, cfg, t1 - t0, n_gpus=world)
     else:
-        remote_worker_loop(worker, cfg, deng.pair_groups[rank], device)
+        deng.worker_loop()
         dist.barrier()
         if device.type == "cuda":
             torch.cuda.synchronize()
