@@ -1,1 +1,2 @@
 from .context import ASYNCcontext, RDDPartialRes, workerState  # noqa: F401
+from .rdd import AsyncRDD, ASYNCbroadcast  # noqa: F401
