@@ -175,7 +175,12 @@ class AsyncEngine:
             self.staleness_seen.append(srv.k - pr.getStaleness())
         else:
             self.staleness_seen.append(pr.getStaleness())
+        from ..utils.trace import get_tracer
+        tr = get_tracer()
         if self.accepts_now(pr):
+            if tr is not None:
+                tr.instant("accept", 0, args={"k": srv.k, "wid": wid,
+                           "staleness": self.staleness_seen[-1]})
             self.accepted_staleness.append(self.staleness_seen[-1])
             srv.finish_time[wid] = now
             sub = srv.submit_time.get(wid)
@@ -199,6 +204,9 @@ class AsyncEngine:
                                 self._local_workers(),
                                 remote_alpha=self._gather_remote_alpha())
         else:
+            if tr is not None:
+                tr.instant("reject", 0, args={"k": srv.k, "wid": wid,
+                           "staleness": self.staleness_seen[-1]})
             srv.last_accept[wid] = False
             self.pending.append(wid)
             self.rejected += 1
@@ -219,6 +227,11 @@ class AsyncEngine:
         qsize = len(self.pending)
         for _ in range(qsize):
             workers_list.append(self.pending.popleft())
+        from ..utils.trace import get_tracer
+        tr = get_tracer()
+        if tr is not None:
+            tr.instant("dispatch", 0, args={"k": srv.k,
+                                            "wids": list(workers_list)})
         self.delay.maybe_activate(srv.k)
         w_snap = srv.w.detach().clone()
         now = time.perf_counter()
@@ -281,6 +294,9 @@ class AsyncEngine:
             ch.dispatch(Dispatch(w=None, stop=True))
         for ch in self.channels:
             ch.join(timeout=10.0)
+        from ..utils.trace import get_tracer, stop_trace
+        if get_tracer() is not None:
+            stop_trace()  # flush the event-log JSON (ASYNCAMD_TRACE)
         return RunResult(k=srv.k, elapsed_ms=elapsed, opt_vars=srv.opt_vars,
                          waiting_time=srv.waiting_time, w=srv.w,
                          staleness_seen=self.staleness_seen,

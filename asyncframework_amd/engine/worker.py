@@ -106,6 +106,13 @@ class Worker:
             if self.is_cuda:
                 self.stream.synchronize()
         elapsed_ms = (time.perf_counter() - t0) * 1000.0
+        from ..utils.trace import get_tracer
+        tr = get_tracer()
+        if tr is not None:
+            tr.complete("round", self.id, tr.now_us() - elapsed_ms * 1000.0,
+                        elapsed_ms * 1000.0,
+                        args={"k_submit": msg.k_submit, "n": n,
+                              "delay_s": msg.delay_s})
         return WorkerResult(worker_id=self.id, g=g, ts=msg.ts,
                             k_submit=msg.k_submit, nrows=n,
                             elapsed_ms=elapsed_ms)

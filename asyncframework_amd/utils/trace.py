@@ -1,0 +1,109 @@
+"""Chrome-trace (Perfetto) event logging — the observability substrate.
+
+The reference's substrate writes scheduler event-log JSON consumed by the
+Spark History Server (reference core/.../scheduler/EventLoggingListener.scala:55,
+LiveListenerBus.scala:44); the stdout timing tables are the framework-level
+channel (SparkASGDThread.scala:115-119,331-338 — kept byte-exact in
+utils/logfmt). This module is the MI355X rebuild of the substrate channel:
+engines emit dispatch/compute/accept/reject/update events into the
+chrome://tracing "Trace Event Format" (one JSON object, ``traceEvents``
+array), viewable in Perfetto — alongside rocprofv3 for the kernel level.
+
+Off the hot path: when disabled every hook is a single attribute check; when
+enabled, events append to an in-memory list under a lock and are written once
+at ``stop()``.
+
+Activate with ``ASYNCAMD_TRACE=/path/trace.json`` (written at engine
+shutdown) or programmatically via ``start_trace``/``stop_trace``.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+_T0_NS = time.perf_counter_ns()
+
+
+def _now_us() -> float:
+    return (time.perf_counter_ns() - _T0_NS) / 1000.0
+
+
+class Tracer:
+    """Collects Trace Event Format events (ph 'X' complete / 'i' instant)."""
+
+    def __init__(self, path: str = ""):
+        self.path = path
+        self.enabled = True
+        self._events: List[Dict[str, Any]] = []
+        self._lock = threading.Lock()
+
+    # pid = engine role, tid = worker id / server lane
+    def complete(self, name: str, tid: int, t0_us: float, dur_us: float,
+                 args: Optional[Dict[str, Any]] = None,
+                 pid: str = "workers") -> None:
+        ev = {"name": name, "ph": "X", "ts": t0_us, "dur": dur_us,
+              "pid": pid, "tid": tid}
+        if args:
+            ev["args"] = args
+        with self._lock:
+            self._events.append(ev)
+
+    def instant(self, name: str, tid: int,
+                args: Optional[Dict[str, Any]] = None,
+                pid: str = "server") -> None:
+        ev = {"name": name, "ph": "i", "ts": _now_us(), "s": "t",
+              "pid": pid, "tid": tid}
+        if args:
+            ev["args"] = args
+        with self._lock:
+            self._events.append(ev)
+
+    def now_us(self) -> float:
+        return _now_us()
+
+    def save(self, path: str = "") -> str:
+        path = path or self.path
+        with self._lock:
+            doc = {"traceEvents": list(self._events),
+                   "displayTimeUnit": "ms"}
+        with open(path, "w") as f:
+            json.dump(doc, f)
+        return path
+
+
+_tracer: Optional[Tracer] = None
+_tracer_lock = threading.Lock()
+
+
+def get_tracer() -> Optional[Tracer]:
+    """The active tracer, or None (the common, zero-cost case). Lazily
+    honors ASYNCAMD_TRACE on first call."""
+    global _tracer
+    if _tracer is None:
+        path = os.environ.get("ASYNCAMD_TRACE", "")
+        if path:
+            with _tracer_lock:
+                if _tracer is None:
+                    _tracer = Tracer(path)
+    return _tracer
+
+
+def start_trace(path: str) -> Tracer:
+    global _tracer
+    with _tracer_lock:
+        _tracer = Tracer(path)
+    return _tracer
+
+
+def stop_trace() -> Optional[str]:
+    """Write and deactivate the current trace; returns the file path."""
+    global _tracer
+    with _tracer_lock:
+        t, _tracer = _tracer, None
+    if t is None:
+        return None
+    return t.save() if t.path else None
