@@ -14,7 +14,7 @@ Default model = BASELINE config 2: ASGD on synthetic mnist8m-shape
 stays FIXED as N grows (sharded over ranks) => strong scaling, matching the
 reference's fixed-dataset/more-partitions semantics. The other BASELINE
 configs (1,3,4,5) are selectable via --model (single-node; N>1 supports the
-flagship dense ASGD/ASAGA models).
+flagship ASGD/ASAGA models, dense and CSR).
 """
 
 from __future__ import annotations
@@ -231,11 +231,18 @@ def run_dist(args, device, rank, world):
     for j in range(M):
         wid = rank * M + j
         s, t = shards[wid]
-        X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + wid,
-                               dtype=dt, device=device,
-                               objective=args.objective)
-        workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s, X=X,
-                                         y=y), cfg, device=device))
+        if args.sparse:
+            indptr, indices, values, yv = synthetic_csr(
+                t - s, args.cols, seed=BASE["seed"] + wid, device=device,
+                dtype=dt)
+            sh = Shard(row_start=s, n_rows=t - s, indptr=indptr,
+                       indices=indices, values=values, y=yv)
+        else:
+            X, y = synthetic_dense(t - s, args.cols, seed=BASE["seed"] + wid,
+                                   dtype=dt, device=device,
+                                   objective=args.objective)
+            sh = Shard(row_start=s, n_rows=t - s, X=X, y=y)
+        workers.append(Worker(wid, sh, cfg, device=device))
     dist.init_process_group("nccl" if device.type == "cuda" else "gloo",
                             rank=rank, world_size=world)
     dist.barrier()
