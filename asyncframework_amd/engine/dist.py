@@ -223,6 +223,21 @@ class DistEngine:
         self.pair_groups = {}
         for wid in range(self.M, cfg.num_workers):
             self.pair_groups[wid] = dist.new_group([0, wid // self.M])
+        self._warmup_pair_comms()
+
+    def _warmup_pair_comms(self) -> None:
+        """Eagerly create every pair communicator in a DETERMINISTIC order
+        before any proxy/worker thread runs. NCCL/RCCL communicator init is
+        a blocking collective that otherwise happens lazily on the first
+        isend/irecv — with many pair groups first-used concurrently from
+        threads in arbitrary per-rank order, two ranks can enter different
+        comm inits first and deadlock. A barrier per group, ascending wid,
+        from both members (rank 0 walks all peers sequentially, each peer
+        walks only its own groups — pairwise order is consistent, and rank 0
+        being the only shared participant rules out cycles)."""
+        for wid in sorted(self.pair_groups):
+            if self.rank in (0, wid // self.M):
+                dist.barrier(group=self.pair_groups[wid])
 
     # -- rank-0 construction (exposed so bench.py can set marks) -------------
     def build_engine(self):
