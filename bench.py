@@ -248,6 +248,26 @@ def run_dist(args, device, rank, world):
     dist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
+    if os.environ.get("ASYNCAMD_DIST_ENGINE", "") == "native":
+        # opt-in C++ rank-0 server (csrc/server_dist.cpp); wire-compatible
+        # with the Python path — flip the default after round-2 multi-GPU
+        # validation
+        from asyncframework_amd.engine.dist_native import NativeDistEngine
+        neng = NativeDistEngine(
+            cfg, workers, device,
+            mark_at=[args.warmup, args.warmup + args.steps])
+        res = neng.run(verbose=False, max_wall_s=1800)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        if rank == 0:
+            t0 = neng.marks.get(args.warmup)
+            t1 = neng.marks.get(args.warmup + args.steps)
+            if t0 is None or t1 is None or t1 <= t0:
+                print(json.dumps({"error": "marks missing"}))
+            else:
+                emit(args, cfg, t1 - t0, n_gpus=world)
+        dist.destroy_process_group()
+        return
     deng = DistEngine(cfg, workers, device)
     if rank == 0:
         eng, server, channels = deng.build_engine()

@@ -38,21 +38,40 @@ def needs_build(out: Path) -> bool:
 
 def build(force: bool = False, verbose: bool = True) -> Path:
     out = so_path()
-    if not force and not needs_build(out):
+    if force or needs_build(out):
+        import pybind11
+        inc_py = sysconfig.get_paths()["include"]
+        inc_pb = pybind11.get_include()
+        cmd = [
+            HIPCC, f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-shared",
+            "-fPIC", "-DNDEBUG", f"-I{inc_py}", f"-I{inc_pb}",
+            *[str(s) for s in SOURCES], "-o", str(out),
+        ]
+        if verbose:
+            print("[build_hip]", " ".join(cmd))
+        subprocess.run(cmd, check=True)
+    elif verbose:
+        print(f"[build_hip] up to date: {out}")
+    build_dist_core(verbose=verbose)
+    return out
+
+
+DIST_SRC = ROOT / "csrc" / "server_dist.cpp"
+DIST_DIR = ROOT / "asyncframework_amd" / "_dist_build"
+
+
+def build_dist_core(verbose: bool = True) -> Path:
+    """The C++ dist server is a torch extension (needs c10d headers) built
+    into its own in-tree .so so _hip_core stays torch-free."""
+    out = DIST_DIR / "_dist_core.so"
+    if out.exists() and out.stat().st_mtime >= DIST_SRC.stat().st_mtime:
         if verbose:
             print(f"[build_hip] up to date: {out}")
         return out
-    import pybind11
-    inc_py = sysconfig.get_paths()["include"]
-    inc_pb = pybind11.get_include()
-    cmd = [
-        HIPCC, f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-shared",
-        "-fPIC", "-DNDEBUG", f"-I{inc_py}", f"-I{inc_pb}",
-        *[str(s) for s in SOURCES], "-o", str(out),
-    ]
-    if verbose:
-        print("[build_hip]", " ".join(cmd))
-    subprocess.run(cmd, check=True)
+    from torch.utils.cpp_extension import load
+    DIST_DIR.mkdir(parents=True, exist_ok=True)
+    load(name="_dist_core", sources=[str(DIST_SRC)],
+         build_directory=str(DIST_DIR), verbose=verbose)
     return out
 
 

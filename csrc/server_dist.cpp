@@ -1,0 +1,476 @@
+// Native multi-GPU parameter server: the rank-0 control plane of the dist
+// engine in C++ (ROADMAP item 1).
+//
+// The Python dist engine (engine/dist.py) spends ~100-250 us of GIL-bound
+// Python per update (proxy threads, pack/unpack, server bookkeeping); this
+// server runs the same protocol with zero Python on the hot path:
+//
+//   * one C++ channel thread per REMOTE worker drives the pair
+//     ProcessGroup (send dispatch -> recv result) — under the "nccl"
+//     backend that is RCCL point-to-point over xGMI; under gloo it runs on
+//     CPU, which is how the CPU test tier exercises THIS exact loop.
+//   * rank-0-local workers call in through local_next_dispatch /
+//     local_deliver (GIL released while blocked), so the Python worker
+//     objects (HIP gradient kernels on their own streams) plug in
+//     unchanged.
+//   * tau filter, quorum gate, straggler model, update rules and
+//     bookkeeping mirror engine/local.py::AsyncEngine + engine/server.py
+//     exactly (reference semantics: SparkASGDThread.scala:153-345,
+//     SparkASAGAThread.scala:191,217-220, RDD.scala:1144-1165).
+//
+// WIRE-COMPATIBLE with engine/dist.py's remote_worker_loop: the packed
+// [d + 8] float layout of engine/messages.py (H_TS..H_SNAP), tag 0, peer
+// group-rank 1. Worker ranks keep running the validated Python loop.
+//
+// Updates use aten tensor ops (w.add_ etc.) — identical numerics to
+// ops.torch_ref on CPU and cuBLAS-free elementwise HIP kernels on GPU; the
+// fused-update HIP kernels remain available for the single-GPU native
+// engine. Checkpoint snap sideband is not yet wired here (use the Python
+// dist engine for checkpointed runs).
+
+#include <torch/extension.h>
+#include <torch/csrc/distributed/c10d/ProcessGroup.hpp>
+
+#include <atomic>
+#include <chrono>
+#include <cmath>
+#include <condition_variable>
+#include <cstdint>
+#include <deque>
+#include <map>
+#include <mutex>
+#include <set>
+#include <thread>
+#include <vector>
+
+namespace {
+
+constexpr int HDR = 8;
+enum { H_TS = 0, H_K, H_ACCEPT, H_STOP, H_DELAY, H_NROWS, H_ELAPSED, H_SNAP };
+
+inline double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+// Host Philox x0 (bit-identical to csrc/philox.h / utils/philox.py) for the
+// reproducible straggler draws: uniform01 counters (0, 0, round, stream).
+inline uint32_t philox_host_x0(uint64_t seed, uint32_t c0, uint32_t c1,
+                               uint32_t c2, uint32_t c3) {
+  uint32_t k0 = (uint32_t)(seed & 0xFFFFFFFFull);
+  uint32_t k1 = (uint32_t)(seed >> 32);
+  for (int r = 0; r < 10; ++r) {
+    uint64_t p0 = 0xD2511F53ull * (uint64_t)c0;
+    uint64_t p1 = 0xCD9E8D57ull * (uint64_t)c2;
+    uint32_t hi0 = (uint32_t)(p0 >> 32), lo0 = (uint32_t)p0;
+    uint32_t hi1 = (uint32_t)(p1 >> 32), lo1 = (uint32_t)p1;
+    c0 = hi1 ^ c1 ^ k0;
+    c1 = lo1;
+    c2 = hi0 ^ c3 ^ k1;
+    c3 = lo0;
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  return c0;
+}
+
+inline double uniform01_host(uint64_t seed, uint32_t round_k,
+                             uint32_t stream) {
+  return philox_host_x0(seed, 0u, 0u, round_k, stream) / 4294967296.0;
+}
+
+struct DSCfg {
+  int64_t d = 0, P = 0, M = 0;  // dims, logical workers, rank-0 locals
+  int64_t N = 0, num_iter = 0, printer_freq = 100;
+  double gamma = 0.01, batch_rate = 0.1;
+  int64_t taw = 1 << 30;
+  int64_t gate = 0;
+  double coeff = 0.0;
+  uint64_t seed = 42;
+  int64_t calib_window = 0;
+  bool asaga = false;
+  bool snapshot_weights = false;
+  double par_recs() const { return batch_rate * (double)N / (double)P; }
+};
+
+struct DispatchMsg {
+  at::Tensor w;  // shared snapshot (all dispatches of a round share one)
+  int64_t ts = 0, k = 0;
+  bool accept = true, stop = false;
+  double delay_s = 0.0;
+};
+
+struct Slot {
+  std::mutex m;
+  std::condition_variable cv;
+  bool has = false;
+  DispatchMsg msg;
+};
+
+class DistServer {
+ public:
+  DistServer(DSCfg cfg, at::Tensor w0,
+             std::vector<c10::intrusive_ptr<c10d::ProcessGroup>> pair_pgs,
+             std::vector<int64_t> mark_at)
+      : cfg_(cfg), w_(w0), pgs_(std::move(pair_pgs)) {
+    TORCH_CHECK(w_.dtype() == at::kFloat && w_.numel() == cfg_.d);
+    TORCH_CHECK((int64_t)pgs_.size() == cfg_.P - cfg_.M,
+                "one pair ProcessGroup per remote worker, wid order");
+    if (cfg_.asaga)
+      alpha_bar_ = at::zeros({cfg_.d}, w_.options());
+    avail_.assign(cfg_.P, 1);
+    last_accept_.assign(cfg_.P, 1);
+    submit_t_.assign(cfg_.P, 0.0);
+    finish_t_.assign(cfg_.P, 0.0);
+    waiting_ms_.assign(cfg_.P, 0);
+    for (int64_t i = 0; i < cfg_.P; ++i)
+      slots_.emplace_back(new Slot());
+    for (auto m : mark_at) mark_at_.insert(m);
+    init_stragglers();
+  }
+
+  // ---- lifecycle ---------------------------------------------------------
+  void start() {
+    t0_ = now_s();
+    if (cfg_.snapshot_weights)
+      opt_ms_.push_back(0), opt_w_.push_back(w_.detach().cpu().clone());
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (int64_t wid = 0; wid < cfg_.P; ++wid) pending_.push_back((int)wid);
+      try_dispatch(/*first=*/true);
+    }
+    for (int64_t wid = cfg_.M; wid < cfg_.P; ++wid)
+      threads_.emplace_back(&DistServer::channel_loop, this, (int)wid);
+  }
+
+  bool wait_done(double timeout_s) {
+    std::unique_lock<std::mutex> lk(mu_);
+    done_cv_.wait_for(lk, std::chrono::duration<double>(timeout_s),
+                      [&] { return done_; });
+    if (!done_) finish_locked();  // wall-clock cap: stop everything
+    return done_;
+  }
+
+  void join() {
+    for (auto& t : threads_)
+      if (t.joinable()) t.join();
+    threads_.clear();
+  }
+
+  // ---- rank-0-local worker API (called from Python worker threads) -------
+  // returns (w, ts, k_submit, accept_prev, delay_s, stop)
+  std::tuple<at::Tensor, int64_t, int64_t, bool, double, bool>
+  local_next_dispatch(int64_t wid) {
+    Slot& s = *slots_[wid];
+    std::unique_lock<std::mutex> lk(s.m);
+    s.cv.wait(lk, [&] { return s.has; });
+    s.has = false;
+    DispatchMsg m = s.msg;
+    return {m.w, m.ts, m.k, m.accept, m.delay_s, m.stop};
+  }
+
+  void local_deliver(int64_t wid, at::Tensor g, int64_t ts, int64_t k_submit,
+                     double elapsed_ms) {
+    deliver((int)wid, g, ts, k_submit, elapsed_ms);
+  }
+
+  // ---- results -----------------------------------------------------------
+  int64_t k() const { return k_; }
+  int64_t applied() const { return applied_; }
+  int64_t rejected() const { return rejected_; }
+  int64_t elapsed_ms() const { return (int64_t)((now_s() - t0_) * 1000.0); }
+  std::vector<int64_t> waiting_ms() const { return waiting_ms_; }
+  std::map<int64_t, double> marks() const { return marks_; }
+  std::vector<int64_t> opt_ms() const { return opt_ms_; }
+  std::vector<at::Tensor> opt_w() const { return opt_w_; }
+  at::Tensor weights() const { return w_; }
+  int64_t max_staleness_seen() const { return max_staleness_; }
+
+  // straggler-model probe for cross-checking against engine/delay.py
+  static double delay_probe(int64_t P, double coeff, uint64_t seed,
+                            double avg_ms, int64_t wid, int64_t round_k) {
+    DSCfg c;
+    c.P = P;
+    c.coeff = coeff;
+    c.seed = seed;
+    DistServer* tmp = nullptr;
+    (void)tmp;
+    // replicate delay_ms_for with an ad-hoc straggler table
+    std::vector<int> kind(P, 0);
+    const int length = (int)std::lround(0.25 * (double)P);
+    const int length_normal = (int)std::lround(0.8 * length);
+    const int length_longtail = length - length_normal;
+    for (int cc = 0; cc < length; ++cc) {
+      const int idx = cc * 4;
+      if (idx < P) kind[idx] = (cc < length_longtail) ? 2 : 1;
+    }
+    if (coeff == 0.0) return 0.0;
+    if (coeff != -1.0)
+      return (wid == 0 && coeff > 0) ? std::round(coeff * avg_ms) : 0.0;
+    if (kind[wid] == 2)
+      return std::round(
+          (uniform01_host(seed, (uint32_t)round_k, (uint32_t)wid) * 7.5 +
+           2.5) * avg_ms);
+    if (kind[wid] == 1)
+      return std::round(
+          (uniform01_host(seed, (uint32_t)round_k, (uint32_t)wid) + 1.5) *
+          avg_ms);
+    return 0.0;
+  }
+
+ private:
+  // ---- remote channel: one thread per remote worker ----------------------
+  void channel_loop(int wid) {
+    auto pg = pgs_[wid - cfg_.M];
+    at::Tensor buf = at::zeros({cfg_.d + HDR}, w_.options());
+    at::Tensor hdr_host = at::zeros({HDR}, at::kFloat);
+    Slot& s = *slots_[wid];
+    std::vector<at::Tensor> v{buf};
+    while (true) {
+      DispatchMsg m;
+      {
+        std::unique_lock<std::mutex> lk(s.m);
+        s.cv.wait(lk, [&] { return s.has; });
+        s.has = false;
+        m = s.msg;
+      }
+      // pack (engine/messages.py layout)
+      if (m.w.defined()) buf.narrow(0, 0, cfg_.d).copy_(m.w);
+      hdr_host[H_TS] = (float)m.ts;
+      hdr_host[H_K] = (float)m.k;
+      hdr_host[H_ACCEPT] = m.accept ? 1.f : 0.f;
+      hdr_host[H_STOP] = m.stop ? 1.f : 0.f;
+      hdr_host[H_DELAY] = (float)m.delay_s;
+      hdr_host[H_NROWS] = 0.f;
+      hdr_host[H_ELAPSED] = 0.f;
+      hdr_host[H_SNAP] = 0.f;
+      buf.narrow(0, cfg_.d, HDR).copy_(hdr_host);
+      pg->send(v, /*group-rank of peer*/ 1, /*tag*/ 0)->wait();
+      if (m.stop) break;
+      pg->recv(v, 1, 0)->wait();
+      // header comes back to host (this is also the completion sync point)
+      at::Tensor h = buf.narrow(0, cfg_.d, HDR).cpu();
+      const float* hp = h.data_ptr<float>();
+      deliver(wid, buf.narrow(0, 0, cfg_.d), (int64_t)hp[H_TS],
+              (int64_t)hp[H_K], (double)hp[H_ELAPSED]);
+    }
+  }
+
+  // ---- completion path (engine/local.py::_process_result +
+  //      engine/server.py::on_completion/accepts/apply) --------------------
+  void deliver(int wid, at::Tensor g, int64_t ts, int64_t k_submit,
+               double elapsed_ms) {
+    (void)k_submit;
+    (void)elapsed_ms;
+    std::lock_guard<std::mutex> lk(mu_);
+    if (done_) return;  // late result after shutdown: dropped (as in Python)
+    const double t_now = now_s();
+    const int64_t staleness = clock_ - ts;  // arrival clock
+    clock_ += 1;
+    max_staleness_ = std::max(max_staleness_, staleness);
+    avail_[wid] = 1;
+    const bool accept =
+        cfg_.asaga ? (k_ - ts) <= cfg_.taw : staleness <= cfg_.taw;
+    if (accept) {
+      finish_t_[wid] = t_now;
+      // delay calibration sample (reference :177-186; Python records wall
+      // round time on accepted results only)
+      cul_time_ms_ += (t_now - submit_t_[wid]) * 1000.0;
+      cul_count_ += 1;
+      apply(g);
+      last_accept_[wid] = 1;
+      if (k_ % cfg_.printer_freq == 0 && cfg_.snapshot_weights) {
+        opt_ms_.push_back((int64_t)((t_now - t0_) * 1000.0));
+        opt_w_.push_back(w_.detach().cpu().clone());
+      }
+      k_ += 1;
+      applied_ += 1;
+      if (mark_at_.count(k_)) marks_[k_] = now_s();
+    } else {
+      last_accept_[wid] = 0;
+      rejected_ += 1;
+    }
+    pending_.push_back(wid);
+    if (k_ >= cfg_.num_iter) {
+      finish_locked();
+      return;
+    }
+    try_dispatch(false);
+  }
+
+  void apply(const at::Tensor& g) {
+    at::Tensor gg = g;
+    if (gg.device() != w_.device()) gg = gg.to(w_.device());
+    if (cfg_.asaga) {
+      // w -= gamma*(g/parRecs); w -= gamma*alphaBar; alphaBar += g/N
+      // (SparkASAGAThread.scala:217-220)
+      w_.add_(gg, -cfg_.gamma / cfg_.par_recs());
+      w_.add_(alpha_bar_, -cfg_.gamma);
+      alpha_bar_.add_(gg, 1.0 / (double)cfg_.N);
+    } else {
+      // gamma/sqrt(k/P + 1), Scala INT division (SparkASGDThread.scala:190)
+      const double gamma_k =
+          cfg_.gamma / std::sqrt((double)(k_ / cfg_.P + 1));
+      w_.add_(gg, -gamma_k / cfg_.par_recs());
+    }
+  }
+
+  // ---- dispatch (engine/local.py::_dispatch_pending) ---------------------
+  void try_dispatch(bool first) {
+    if (pending_.empty()) return;
+    int64_t avail_n = 0;
+    for (auto a : avail_) avail_n += a;
+    const int64_t init_workers = first ? cfg_.P : avail_n;
+    if (init_workers < cfg_.gate) return;
+    if (!delay_flag_ && k_ > cfg_.calib_window) {
+      if (cul_count_ > 0) avg_delay_ms_ = cul_time_ms_ / (double)cul_count_;
+      delay_flag_ = true;
+    }
+    at::Tensor w_snap = w_.detach().clone();
+    const double t_now = now_s();
+    const size_t qn = pending_.size();
+    for (size_t i = 0; i < qn; ++i) {
+      const int wid = pending_.front();
+      pending_.pop_front();
+      const double prev_fin = finish_t_[wid] == 0.0 ? t_now : finish_t_[wid];
+      waiting_ms_[wid] += (int64_t)((t_now - prev_fin) * 1000.0);
+      submit_t_[wid] = t_now;
+      avail_[wid] = 0;
+      DispatchMsg m;
+      m.w = w_snap;
+      m.ts = clock_;
+      m.k = k_;
+      m.accept = last_accept_[wid] != 0;
+      m.delay_s = delay_ms_for(wid, k_) / 1000.0;
+      m.stop = false;
+      fill_slot(wid, m);
+    }
+  }
+
+  void fill_slot(int wid, const DispatchMsg& m) {
+    Slot& s = *slots_[wid];
+    std::lock_guard<std::mutex> lk(s.m);
+    s.msg = m;
+    s.has = true;
+    s.cv.notify_one();
+  }
+
+  void finish_locked() {
+    if (done_) return;
+    done_ = true;
+    elapsed_final_ms_ = (int64_t)((now_s() - t0_) * 1000.0);
+    DispatchMsg stop;
+    stop.stop = true;
+    stop.w = at::Tensor();
+    for (int64_t wid = 0; wid < cfg_.P; ++wid) fill_slot((int)wid, stop);
+    done_cv_.notify_all();
+  }
+
+  // ---- straggler model (reference SparkASGDThread.scala:124-141,287-312) -
+  void init_stragglers() {
+    straggler_kind_.assign(cfg_.P, 0);
+    const int length = (int)std::lround(0.25 * (double)cfg_.P);
+    const int length_normal = (int)std::lround(0.8 * length);
+    const int length_longtail = length - length_normal;
+    for (int c = 0; c < length; ++c) {
+      const int idx = c * 4;
+      if (idx < cfg_.P) straggler_kind_[idx] = (c < length_longtail) ? 2 : 1;
+    }
+  }
+
+  double delay_ms_for(int wid, int64_t round_k) const {
+    if (!delay_flag_ || cfg_.coeff == 0.0) return 0.0;
+    if (cfg_.coeff != -1.0) {
+      if (wid == 0 && cfg_.coeff > 0)
+        return std::round(cfg_.coeff * avg_delay_ms_);
+      return 0.0;
+    }
+    if (straggler_kind_[wid] == 2) {
+      const double u =
+          uniform01_host(cfg_.seed, (uint32_t)round_k, (uint32_t)wid);
+      return std::round((u * 7.5 + 2.5) * avg_delay_ms_);
+    }
+    if (straggler_kind_[wid] == 1) {
+      const double u =
+          uniform01_host(cfg_.seed, (uint32_t)round_k, (uint32_t)wid);
+      return std::round((u + 1.5) * avg_delay_ms_);
+    }
+    return 0.0;
+  }
+
+  DSCfg cfg_;
+  at::Tensor w_, alpha_bar_;
+  std::vector<c10::intrusive_ptr<c10d::ProcessGroup>> pgs_;
+  std::vector<std::unique_ptr<Slot>> slots_;
+  std::vector<std::thread> threads_;
+
+  std::mutex mu_;
+  std::condition_variable done_cv_;
+  bool done_ = false;
+  int64_t k_ = 0, clock_ = 0, applied_ = 0, rejected_ = 0;
+  int64_t max_staleness_ = -1;
+  int64_t elapsed_final_ms_ = 0;
+  std::vector<uint8_t> avail_, last_accept_;
+  std::deque<int> pending_;
+  std::vector<double> submit_t_, finish_t_;
+  std::vector<int64_t> waiting_ms_;
+  std::set<int64_t> mark_at_;
+  std::map<int64_t, double> marks_;
+  std::vector<int64_t> opt_ms_;
+  std::vector<at::Tensor> opt_w_;
+  double t0_ = 0;
+  // delay calibration
+  double cul_time_ms_ = 0, avg_delay_ms_ = 0;
+  int64_t cul_count_ = 0;
+  bool delay_flag_ = false;
+  std::vector<int> straggler_kind_;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  py::class_<DSCfg>(m, "DSCfg")
+      .def(py::init<>())
+      .def_readwrite("d", &DSCfg::d)
+      .def_readwrite("P", &DSCfg::P)
+      .def_readwrite("M", &DSCfg::M)
+      .def_readwrite("N", &DSCfg::N)
+      .def_readwrite("num_iter", &DSCfg::num_iter)
+      .def_readwrite("printer_freq", &DSCfg::printer_freq)
+      .def_readwrite("gamma", &DSCfg::gamma)
+      .def_readwrite("batch_rate", &DSCfg::batch_rate)
+      .def_readwrite("taw", &DSCfg::taw)
+      .def_readwrite("gate", &DSCfg::gate)
+      .def_readwrite("coeff", &DSCfg::coeff)
+      .def_readwrite("seed", &DSCfg::seed)
+      .def_readwrite("calib_window", &DSCfg::calib_window)
+      .def_readwrite("asaga", &DSCfg::asaga)
+      .def_readwrite("snapshot_weights", &DSCfg::snapshot_weights);
+
+  py::class_<DistServer>(m, "DistServer")
+      .def(py::init<DSCfg, at::Tensor,
+                    std::vector<c10::intrusive_ptr<c10d::ProcessGroup>>,
+                    std::vector<int64_t>>())
+      .def("start", &DistServer::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("wait_done", &DistServer::wait_done,
+           py::call_guard<py::gil_scoped_release>())
+      .def("join", &DistServer::join,
+           py::call_guard<py::gil_scoped_release>())
+      .def("local_next_dispatch", &DistServer::local_next_dispatch,
+           py::call_guard<py::gil_scoped_release>())
+      .def("local_deliver", &DistServer::local_deliver,
+           py::call_guard<py::gil_scoped_release>())
+      .def("k", &DistServer::k)
+      .def("applied", &DistServer::applied)
+      .def("rejected", &DistServer::rejected)
+      .def("elapsed_ms", &DistServer::elapsed_ms)
+      .def("waiting_ms", &DistServer::waiting_ms)
+      .def("marks", &DistServer::marks)
+      .def("opt_ms", &DistServer::opt_ms)
+      .def("opt_w", &DistServer::opt_w)
+      .def("weights", &DistServer::weights)
+      .def("max_staleness_seen", &DistServer::max_staleness_seen)
+      .def_static("delay_probe", &DistServer::delay_probe);
+}

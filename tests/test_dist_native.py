@@ -1,0 +1,111 @@
+"""The C++ dist server (csrc/server_dist.cpp) over gloo on CPU, world=2 —
+the SAME C++ loop that drives RCCL on GPU (only the ProcessGroup backend
+differs). Mirrors tests/test_engine_dist.py's scenarios."""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from asyncframework_amd.data.shard import row_shards
+from asyncframework_amd.data.synthetic import synthetic_dense
+from asyncframework_amd.engine.config import EngineConfig
+from asyncframework_amd.engine.dist_native import (NativeDistEngine,
+                                                   dist_core_available)
+from asyncframework_amd.engine.worker import Shard, Worker
+
+pytestmark = pytest.mark.skipif(not dist_core_available(),
+                                reason="_dist_core.so not built")
+
+WORLD = 2
+
+
+def _rank_main(rank, init_file, out_file, algo, P):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        cfg = EngineConfig(d=24, N=400, num_workers=P, num_iterations=60,
+                           gamma=0.5 if algo == "asgd" else 0.05,
+                           taw=2 ** 30, batch_rate=0.3, bucket_ratio=0.5,
+                           printer_freq=20, delay_coeff=0.0, seed=42,
+                           device="cpu", sync=False, algo=algo)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+        M = P // WORLD
+        workers = []
+        for j in range(M):
+            wid = rank * M + j
+            s, t = row_shards(cfg.N, P)[wid]
+            workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                             X=X[s:t], y=y[s:t]), cfg,
+                                  device=torch.device("cpu")))
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"),
+                               mark_at=[10, 50])
+        res = eng.run(verbose=False, max_wall_s=120)
+        if rank == 0:
+            obj0 = float(((X @ torch.zeros(cfg.d) - y) ** 2).mean())
+            obj1 = float(((X @ res.w - y) ** 2).mean())
+            n_snap = len(res.opt_vars)
+            n_marks = len(eng.marks)
+            n_wait = sum(1 for v in res.waiting_time.values() if v >= 0)
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{obj0},{obj1},{res.applied},"
+                        f"{n_snap},{n_marks},{n_wait}")
+    finally:
+        dist.destroy_process_group()
+
+
+def _run(algo, P):
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_rank_main, args=(init_file, out_file, algo, P),
+                 nprocs=WORLD, join=True)
+        with open(out_file) as f:
+            return f.read().split(",")
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_asgd_converges():
+    k, obj0, obj1, applied, n_snap, n_marks, n_wait = _run("asgd", P=2)
+    assert int(k) >= 60
+    assert float(obj1) < float(obj0)
+    assert int(applied) >= 60
+    assert int(n_snap) >= 3    # 0 + every printer_freq=20
+    assert int(n_marks) == 2   # both bench marks stamped
+    assert int(n_wait) == 2
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_asaga_converges():
+    k, obj0, obj1, applied, *_ = _run("asaga", P=2)
+    assert int(k) >= 60
+    assert float(obj1) < float(obj0)
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_multi_worker_per_rank():
+    k, obj0, obj1, applied, n_snap, n_marks, n_wait = _run("asgd", P=4)
+    assert int(k) >= 60
+    assert float(obj1) < float(obj0)
+    assert int(n_wait) == 4    # every logical worker dispatched
+
+
+def test_delay_model_matches_python_injector():
+    """The C++ straggler model is bit-compatible with engine/delay.py
+    (same Philox draws, same selection/calibration rules)."""
+    from asyncframework_amd.engine.delay import DelayInjector
+    from asyncframework_amd.engine.dist_native import _load
+    core = _load()
+    for P in (4, 8, 32):
+        for coeff in (-1.0, 0.0, 1.0, 2.5):
+            inj = DelayInjector(P, coeff=coeff, seed=42, calib_window=0)
+            inj._cul_time, inj._cul_count = 100.0, 1
+            inj.maybe_activate(1)
+            for wid in range(P):
+                for rk in (0, 3, 17):
+                    assert core.DistServer.delay_probe(
+                        P, coeff, 42, 100.0, wid, rk) == \
+                        inj.delay_ms(wid, rk), (P, coeff, wid, rk)
