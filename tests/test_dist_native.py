@@ -109,3 +109,87 @@ def test_delay_model_matches_python_injector():
                     assert core.DistServer.delay_probe(
                         P, coeff, 42, 100.0, wid, rk) == \
                         inj.delay_ms(wid, rk), (P, coeff, wid, rk)
+
+
+def _tau_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        # taw=0: only zero-staleness results accepted — heavy rejection
+        # traffic; the server must still reach num_iterations (the
+        # reference silently drops over-tau results and requeues,
+        # SparkASGDThread.scala:202-205)
+        cfg = EngineConfig(d=12, N=160, num_workers=4, num_iterations=25,
+                           gamma=0.2, taw=0, batch_rate=0.3,
+                           bucket_ratio=0.25, printer_freq=1000,
+                           delay_coeff=0.0, seed=7, device="cpu",
+                           sync=False, algo="asgd", snapshot_weights=False)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=3)
+        M = 2
+        workers = []
+        for j in range(M):
+            wid = rank * M + j
+            s, t = row_shards(cfg.N, 4)[wid]
+            workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                             X=X[s:t], y=y[s:t]), cfg,
+                                  device=torch.device("cpu")))
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=60)
+        if rank == 0:
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{res.applied},{res.rejected}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_tau_zero_torture():
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_tau_rank_main, args=(init_file, out_file), nprocs=WORLD,
+                 join=True)
+        with open(out_file) as f:
+            k, applied, rejected = map(int, f.read().split(","))
+        assert k >= 25
+        assert applied >= 25
+        # with taw=0 and 4 concurrent workers rejections must occur
+        assert rejected > 0
+
+
+def _wall_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        # unreachable iteration budget: the wall-clock cap must stop the
+        # run cleanly on every rank (no hang, partial results returned)
+        cfg = EngineConfig(d=12, N=160, num_workers=2, num_iterations=10 ** 9,
+                           gamma=0.1, taw=2 ** 30, batch_rate=0.3,
+                           bucket_ratio=0.5, printer_freq=1 << 30,
+                           delay_coeff=0.0, seed=7, device="cpu",
+                           sync=False, algo="asgd", snapshot_weights=False)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=3)
+        s, t = row_shards(cfg.N, 2)[rank]
+        worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X[s:t],
+                                    y=y[s:t]), cfg,
+                        device=torch.device("cpu"))
+        eng = NativeDistEngine(cfg, [worker], torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=2.0)
+        if rank == 0:
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{res.elapsed_ms}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_wall_clock_cap():
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_wall_rank_main, args=(init_file, out_file), nprocs=WORLD,
+                 join=True)
+        with open(out_file) as f:
+            k, elapsed = map(int, f.read().split(","))
+        assert k > 0          # made progress
+        assert elapsed < 30_000  # stopped near the cap, not the timeout
