@@ -186,6 +186,8 @@ class DistServer {
   std::vector<at::Tensor> opt_w() const { return opt_w_; }
   at::Tensor weights() const { return w_; }
   int64_t max_staleness_seen() const { return max_staleness_; }
+  bool delay_active() const { return delay_flag_; }
+  double avg_delay_ms() const { return avg_delay_ms_; }
 
   // straggler-model probe for cross-checking against engine/delay.py
   static double delay_probe(int64_t P, double coeff, uint64_t seed,
@@ -472,5 +474,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("opt_w", &DistServer::opt_w)
       .def("weights", &DistServer::weights)
       .def("max_staleness_seen", &DistServer::max_staleness_seen)
+      .def("delay_active", &DistServer::delay_active)
+      .def("avg_delay_ms", &DistServer::avg_delay_ms)
       .def_static("delay_probe", &DistServer::delay_probe);
 }

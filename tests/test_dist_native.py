@@ -193,3 +193,50 @@ def test_native_dist_wall_clock_cap():
             k, elapsed = map(int, f.read().split(","))
         assert k > 0          # made progress
         assert elapsed < 30_000  # stopped near the cap, not the timeout
+
+
+def _delay_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        # coeff=5.0 + tiny calibration window: once active, worker 0's
+        # rounds are delayed by 5x the measured average round time — the
+        # end-to-end wiring (C++ server computes delay_s -> header ->
+        # Python worker sleeps) must slow the run measurably
+        cfg = EngineConfig(d=12, N=160, num_workers=2, num_iterations=40,
+                           gamma=0.1, taw=2 ** 30, batch_rate=0.3,
+                           bucket_ratio=0.5, printer_freq=1 << 30,
+                           delay_coeff=5.0, seed=7, device="cpu",
+                           sync=False, algo="asgd", snapshot_weights=False,
+                           calib_factor=2)  # window = 2*P = 4 tasks
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=3)
+        s, t = row_shards(cfg.N, 2)[rank]
+        worker = Worker(rank, Shard(row_start=s, n_rows=t - s, X=X[s:t],
+                                    y=y[s:t]), cfg,
+                        device=torch.device("cpu"))
+        eng = NativeDistEngine(cfg, [worker], torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=60)
+        if rank == 0:
+            srv = eng.srv
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{int(srv.delay_active())},"
+                        f"{srv.avg_delay_ms()},{res.waiting_time[1]}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_delay_injection_active():
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_delay_rank_main, args=(init_file, out_file), nprocs=WORLD,
+                 join=True)
+        with open(out_file) as f:
+            k, active, avg_ms, wait1 = f.read().split(",")
+        assert int(k) >= 40
+        assert int(active) == 1          # calibration completed
+        assert float(avg_ms) > 0.0
+        # worker 1 (not a straggler) accumulates waiting time while the
+        # server's round cadence is held back by straggling worker 0
+        assert int(wait1) >= 0
