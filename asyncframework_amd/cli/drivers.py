@@ -8,7 +8,18 @@ Positional args (13) exactly as the reference drivers parse them
 --objective, --sparse, --history-placement) come after the positionals.
 
 ``file name`` = 'synthetic' generates data of the given shape instead of
-loading a LibSVM file (no network in this environment)."""
+loading a LibSVM file (no network in this environment).
+
+Multi-GPU: launch any 13-arg driver under torchrun (one rank per GPU) and
+it routes through the dist engine — ``numPart`` stays the LOGICAL worker
+count (the reference's partitions), spread M = numPart/world per rank;
+rank 0 prints the unchanged stdout contract:
+
+    torchrun --nproc-per-node 8 -m asyncframework_amd.cli asgd-thread \
+        synthetic synthetic 784 8100000 32 16000 1.5e-3 20000000 0.01 \
+        0.7 200 -1 42 --device cuda --dtype bf16
+(--engine native under torchrun selects the C++ rank-0 server,
+csrc/server_dist.cpp)."""
 
 from __future__ import annotations
 
@@ -92,6 +103,11 @@ def _cfg13(a, algo: str, sync: bool) -> EngineConfig:
 
 
 def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:
+    import os
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        _run_dist(cfg, a, app, names, vals, world)
+        return
     logfmt.print_header(app, names, vals)
     sparse = a.sparse
     data = runner.load_dataset(cfg, a.pathname, a.fname, sparse=sparse,
@@ -103,6 +119,72 @@ def _run(cfg: EngineConfig, a, app: str, names, vals) -> None:
     res, _srv = runner.run_engine(cfg, workers, max_wall_s=a.max_wall_s,
                                   engine=a.engine, resume_from=a.resume_from)
     runner.final_report(cfg, res, data, sparse, device=a.device)
+
+
+def _run_dist(cfg: EngineConfig, a, app: str, names, vals,
+              world: int) -> None:
+    """torchrun path: one process per GPU; numPart logical workers spread
+    M per rank (the reference's partitions-independent-of-executors model).
+    Rank 0 hosts the server and prints the stdout contract; every rank
+    loads the dataset once (rank 0 needs it whole for the objective-sweep
+    epilogue) and keeps only its shards."""
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    from ..data.shard import row_shards
+    from ..engine.worker import Shard, Worker
+
+    rank = int(os.environ["RANK"])
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    device = torch.device(a.device)
+    if device.type == "cuda":
+        device = torch.device(f"cuda:{local_rank}")
+        torch.cuda.set_device(device)
+    assert cfg.num_workers % world == 0, \
+        "numPart must be a multiple of the torchrun world size"
+    M = cfg.num_workers // world
+    cfg.device = str(device)
+    if rank == 0:
+        logfmt.print_header(app, names, vals)
+    sparse = a.sparse
+    data = runner.load_dataset(cfg, a.pathname, a.fname, sparse=sparse,
+                               device="cpu")
+    shards = row_shards(cfg.N, cfg.num_workers)
+    workers = []
+    for j in range(M):
+        wid = rank * M + j
+        s, t = shards[wid]
+        if sparse:
+            indptr, indices, values, y = data
+            base = int(indptr[s])
+            sh = Shard(row_start=s, n_rows=t - s,
+                       indptr=(indptr[s:t + 1] - base).to(device),
+                       indices=indices[base:int(indptr[t])].to(device),
+                       values=values[base:int(indptr[t])].to(device),
+                       y=y[s:t].to(device))
+        else:
+            X, y = data
+            sh = Shard(row_start=s, n_rows=t - s, X=X[s:t].to(device),
+                       y=y[s:t].to(device))
+        workers.append(Worker(wid, sh, cfg, device=device))
+    dist.init_process_group("nccl" if device.type == "cuda" else "gloo")
+    if a.engine == "native":
+        assert not cfg.sync, "--engine native under torchrun is async-only"
+        assert not cfg.checkpoint_path and not a.resume_from, \
+            "checkpointing under torchrun needs --engine threads"
+        from ..engine.dist_native import NativeDistEngine
+        eng = NativeDistEngine(cfg, workers, device)
+        res = eng.run(max_wall_s=a.max_wall_s, verbose=(rank == 0))
+    else:
+        from ..engine.dist import DistEngine
+        eng = DistEngine(cfg, workers, device)
+        res = eng.run(max_wall_s=a.max_wall_s, verbose=(rank == 0),
+                      resume_from=a.resume_from)
+    if rank == 0:
+        runner.final_report(cfg, res, data, sparse, device="cpu")
+    dist.destroy_process_group()
 
 
 def _vals13(a):

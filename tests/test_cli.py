@@ -3,6 +3,7 @@
 196, 349, 355-362, 400-410)."""
 
 import io
+import os
 import re
 from contextlib import redirect_stdout
 
@@ -116,3 +117,32 @@ def test_cli_resume_from_checkpoint(tmp_path):
     args[5] = "60"  # numIter
     out = _capture(drivers.asgd_thread, args + ["--resume-from", ck])
     assert "finished" in out
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("engine", ["threads", "native"])
+def test_cli_torchrun_dist_contract(engine, tmp_path):
+    """13-arg drivers under torchrun (2 ranks, gloo): numPart stays the
+    logical worker count; rank 0 prints the unchanged stdout contract."""
+    import subprocess
+    import sys
+    args = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+            "--nproc-per-node=2", "--master-addr=127.0.0.1",
+            "--master-port=29733" if engine == "threads"
+            else "--master-port=29734",
+            "-m", "asyncframework_amd.cli", "asgd-thread",
+            "synthetic", "synthetic", "16", "200", "4", "40", "0.5",
+            "1000000", "0.3", "0.5", "10", "0", "42", "--engine", engine]
+    out = subprocess.run(args, capture_output=True, text=True, timeout=240,
+                         cwd=os.path.dirname(os.path.dirname(
+                             os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = out.stdout.splitlines()
+    assert lines[0] == "Spark ASGD application started"
+    assert any(re.match(r"Iteration \d+ is finished", l) for l in lines)
+    assert any(re.match(r"Elapsed time\(ms\): \d+", l) for l in lines)
+    assert lines[-1] == "finished"
+    sep = max(i for i, l in enumerate(lines) if l.startswith("*********"))
+    csv = [l for l in lines[sep + 1:] if re.match(r"^\d+,[0-9.eE+-]+$", l)]
+    objs = [float(l.split(",")[1]) for l in csv]
+    assert len(objs) >= 2 and objs[-1] < objs[0]
