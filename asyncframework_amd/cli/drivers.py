@@ -170,6 +170,17 @@ def _run_dist(cfg: EngineConfig, a, app: str, names, vals,
                        y=y[s:t].to(device))
         workers.append(Worker(wid, sh, cfg, device=device))
     dist.init_process_group("nccl" if device.type == "cuda" else "gloo")
+    if cfg.algo == "mllib":
+        # the MLlib treeAggregate baseline maps to all_reduce:
+        # replicated weights, the collective IS the barrier
+        from ..engine.dist_sync import AllReduceSyncEngine
+        eng = AllReduceSyncEngine(cfg, workers, device)
+        res = eng.run(max_wall_s=a.max_wall_s, verbose=(rank == 0))
+        if rank == 0:
+            runner.final_report(cfg, res, data, sparse, device="cpu")
+        dist.barrier()
+        dist.destroy_process_group()
+        return
     if a.engine == "native":
         assert not cfg.sync, "--engine native under torchrun is async-only"
         assert not cfg.checkpoint_path and not a.resume_from, \
@@ -232,9 +243,15 @@ def sgd_mllib(argv: Optional[List[str]] = None) -> None:
         batch_rate=float(a.b), algo="mllib", sync=True, printer_freq=100,
         objective=a.objective, dtype=a.dtype, device=a.device, seed=42,
         delay_coeff=0.0)
-    logfmt.print_header("MLlib SGD", ARG_NAMES_8,
-                        [a.pathname, a.fname, a.d, a.N, a.numPart, a.numIter,
-                         a.gamma, a.b])
+    import os
+    vals = [a.pathname, a.fname, a.d, a.N, a.numPart, a.numIter, a.gamma,
+            a.b]
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        a.resume_from = ""
+        _run_dist(cfg, a, "MLlib SGD", ARG_NAMES_8, vals, world)
+        return
+    logfmt.print_header("MLlib SGD", ARG_NAMES_8, vals)
     data = runner.load_dataset(cfg, a.pathname, a.fname, sparse=a.sparse,
                                device=a.device)
     if a.sparse:
