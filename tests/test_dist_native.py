@@ -240,3 +240,50 @@ def test_native_dist_delay_injection_active():
         # worker 1 (not a straggler) accumulates waiting time while the
         # server's round cadence is held back by straggling worker 0
         assert int(wait1) >= 0
+
+
+def _world1_run(device: str):
+    """world=1: no remote channels — the C++ server drives M local Python
+    workers through the slot API only. On GPU this validates the server's
+    aten-update path + GIL-released slot handoff with cuda tensors on a
+    single device (the multi-GPU channel path shares all of this code)."""
+    import tempfile as _tf
+    with _tf.TemporaryDirectory() as td:
+        dist.init_process_group("gloo", init_method=f"file://{td}/i",
+                                rank=0, world_size=1)
+        try:
+            dev = torch.device(device)
+            cfg = EngineConfig(d=32, N=400, num_workers=4,
+                               num_iterations=80, gamma=0.5, taw=2 ** 30,
+                               batch_rate=0.2, bucket_ratio=0.5,
+                               printer_freq=40, delay_coeff=0.0, seed=42,
+                               device=device, sync=False, algo="asgd")
+            X, y = synthetic_dense(cfg.N, cfg.d, seed=11, device=dev)
+            workers = []
+            for wid in range(4):
+                s, t = row_shards(cfg.N, 4)[wid]
+                workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                                 X=X[s:t], y=y[s:t]), cfg,
+                                      device=dev))
+            eng = NativeDistEngine(cfg, workers, dev, mark_at=[20, 70])
+            res = eng.run(verbose=False, max_wall_s=120)
+            obj0 = float(((X.float() @ torch.zeros(cfg.d, device=dev) - y)
+                          ** 2).mean())
+            obj1 = float(((X.float() @ res.w - y) ** 2).mean())
+            assert res.k >= 80
+            assert obj1 < obj0
+            assert len(eng.marks) == 2
+            assert len(res.opt_vars) >= 2
+        finally:
+            dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_world1_cpu():
+    _world1_run("cpu")
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_native_dist_world1_gpu():
+    _world1_run("cuda:0")
