@@ -248,6 +248,8 @@ def _world1_run(device: str):
     aten-update path + GIL-released slot handoff with cuda tensors on a
     single device (the multi-GPU channel path shares all of this code)."""
     import tempfile as _tf
+    # GPU boxes may not resolve their own hostname; pin gloo to loopback
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
     with _tf.TemporaryDirectory() as td:
         dist.init_process_group("gloo", init_method=f"file://{td}/i",
                                 rank=0, world_size=1)
