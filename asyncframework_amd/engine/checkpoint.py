@@ -42,13 +42,10 @@ def capture_state(server: Server, workers: Optional[List[Worker]] = None,
     return state
 
 
-def save_checkpoint(path: str, server: Server,
-                    workers: Optional[List[Worker]] = None,
-                    remote_alpha: Optional[Dict[int, torch.Tensor]] = None
-                    ) -> None:
+def save_state(path: str, state: Dict) -> None:
     """Atomic write (tmp + rename) so a crash mid-save keeps the previous
-    checkpoint valid."""
-    state = capture_state(server, workers, remote_alpha)
+    checkpoint valid. ``state`` uses the capture_state schema — all engines
+    (threads, dist, native dist) write interchangeable checkpoints."""
     d = os.path.dirname(os.path.abspath(path))
     os.makedirs(d, exist_ok=True)
     fd, tmp = tempfile.mkstemp(dir=d, suffix=".ckpt.tmp")
@@ -59,6 +56,13 @@ def save_checkpoint(path: str, server: Server,
     finally:
         if os.path.exists(tmp):
             os.unlink(tmp)
+
+
+def save_checkpoint(path: str, server: Server,
+                    workers: Optional[List[Worker]] = None,
+                    remote_alpha: Optional[Dict[int, torch.Tensor]] = None
+                    ) -> None:
+    save_state(path, capture_state(server, workers, remote_alpha))
 
 
 def load_checkpoint(path: str) -> Dict:

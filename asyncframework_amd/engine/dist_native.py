@@ -11,7 +11,10 @@ proxy threads + server loop move to C++ (the Python engine spends
 
 Status: CPU(gloo)-validated this round; first multi-GPU RCCL execution is
 round 2 (the Python dist engine remains the default N>1 path until then).
-Checkpoint/resume stays on the Python dist engine."""
+Checkpoint/resume is supported: a rank-0 monitor thread snapshots
+(w, alpha_bar, k, clock, history tables — remote ones gathered through the
+C++ snap sideband) with the same schema as the threads engine, so
+checkpoints are interchangeable across engines."""
 
 from __future__ import annotations
 
@@ -23,10 +26,11 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 
+from ..data.shard import row_shards
 from .config import EngineConfig
-from .dist import DistEngine
+from .dist import DistEngine, _send
 from .local import RunResult
-from .messages import Dispatch
+from .messages import HDR, Dispatch, pack_dispatch
 from .worker import Worker
 
 _mod = None
@@ -75,7 +79,7 @@ class NativeDistEngine:
         self.marks = {}
         self.srv = None
 
-    def _make_cfg(self):
+    def _make_cfg(self, k0: int = 0, clock0: int = 0):
         core = _load()
         c = core.DSCfg()
         cfg = self.cfg
@@ -94,6 +98,8 @@ class NativeDistEngine:
         c.calib_window = cfg.calib_factor * cfg.num_workers
         c.asaga = cfg.algo == "asaga"
         c.snapshot_weights = cfg.snapshot_weights
+        c.k0 = k0
+        c.clock0 = clock0
         return c
 
     def _local_worker_loop(self, wid: int, worker: Worker):
@@ -112,19 +118,94 @@ class NativeDistEngine:
             srv.local_deliver(wid, g.contiguous(), res.ts, res.k_submit,
                               res.elapsed_ms)
 
+    # ---- checkpointing (same schema as engine/checkpoint.py) --------------
+    def _gather_state(self):
+        cfg, srv = self.cfg, self.srv
+        state = {
+            "k": srv.k(),
+            "current_time": srv.clock(),
+            "w": srv.weights().detach().cpu().clone(),
+            "alpha_bar": (srv.alpha_bar().detach().cpu().clone()
+                          if cfg.algo == "asaga" else None),
+            "cfg": cfg.__dict__.copy(),
+            "alpha": {},
+        }
+        for j, wk in enumerate(self.base.local_workers):
+            if wk.alpha is not None:
+                state["alpha"][j] = wk.alpha.detach().cpu().clone()
+        if cfg.algo == "asaga":
+            for wid in range(self.M, cfg.num_workers):
+                srv.request_alpha_snapshot(wid)
+            for wid in range(self.M, cfg.num_workers):
+                if srv.wait_alpha(wid, 15.0):
+                    t = srv.get_alpha(wid)
+                    if t is not None:  # empty = peer already shut down
+                        state["alpha"][wid] = t
+        return state
+
+    def _checkpoint_monitor(self, stop_ev: threading.Event):
+        from .checkpoint import save_state
+        cfg = self.cfg
+        next_k = cfg.checkpoint_every
+        while not stop_ev.is_set():
+            k = self.srv.k()
+            if k >= next_k and k < cfg.num_iterations:
+                # (a snap queued after shutdown would never be answered —
+                # the run's final state is still captured by the last
+                # periodic snapshot, as in the threads engine)
+                save_state(cfg.checkpoint_path, self._gather_state())
+                next_k += cfg.checkpoint_every
+            stop_ev.wait(0.05)
+
+    def _push_alpha(self, wid: int, table: torch.Tensor) -> None:
+        """Resume: send a restored history table to wid's rank BEFORE the
+        C++ channel threads start (plain sends on the pair group are
+        race-free then); the Python worker loop handles snap=2."""
+        pg = self.base.pair_groups[wid]
+        buf = torch.zeros(self.cfg.d + HDR, dtype=torch.float32,
+                          device=self.device)
+        pack_dispatch(buf, self.cfg.d, Dispatch(w=None, snap=2))
+        peer = wid // self.M
+        _send(buf, peer, pg)
+        _send(table.to(dtype=torch.float32, device=self.device), peer, pg)
+
     def run(self, max_wall_s: Optional[float] = None,
-            verbose: bool = True) -> Optional[RunResult]:
+            verbose: bool = True,
+            resume_from: str = "") -> Optional[RunResult]:
         cfg = self.cfg
         if self.rank != 0:
             self.base.worker_loop()
             dist.barrier()
             return None
         core = _load()
+        k0 = clock0 = 0
         w0 = torch.zeros(cfg.d, dtype=torch.float32, device=self.device)
+        state = None
+        if resume_from:
+            from .checkpoint import load_checkpoint
+            state = load_checkpoint(resume_from)
+            k0 = int(state["k"])
+            clock0 = int(state["current_time"])
+            w0 = state["w"].to(self.device)
+            for j, wk in enumerate(self.base.local_workers):
+                if wk.alpha is not None and j in state["alpha"]:
+                    wk.alpha.copy_(state["alpha"][j].to(wk.alpha.device))
         pgs = [self.base.pair_groups[wid]
                for wid in range(self.M, cfg.num_workers)]
-        self.srv = core.DistServer(self._make_cfg(), w0, pgs,
-                                   [int(m) for m in self.mark_at])
+        shards = row_shards(cfg.N, cfg.num_workers)
+        alpha_rows = [(shards[wid][1] - shards[wid][0])
+                      if cfg.algo == "asaga" else 0
+                      for wid in range(self.M, cfg.num_workers)]
+        self.srv = core.DistServer(self._make_cfg(k0, clock0), w0, pgs,
+                                   [int(m) for m in self.mark_at],
+                                   alpha_rows=alpha_rows)
+        if state is not None:
+            if cfg.algo == "asaga" and state.get("alpha_bar") is not None:
+                self.srv.alpha_bar().copy_(
+                    state["alpha_bar"].to(self.device))
+            for wid in range(self.M, cfg.num_workers):
+                if cfg.algo == "asaga" and wid in state["alpha"]:
+                    self._push_alpha(wid, state["alpha"][wid])
         self.srv.start()
         threads = []
         for j, wk in enumerate(self.base.local_workers):
@@ -133,7 +214,17 @@ class NativeDistEngine:
                                   name=f"lworker-{j}")
             th.start()
             threads.append(th)
+        ckpt_stop = None
+        if cfg.checkpoint_every > 0 and cfg.checkpoint_path:
+            ckpt_stop = threading.Event()
+            ckpt_th = threading.Thread(
+                target=self._checkpoint_monitor, args=(ckpt_stop,),
+                daemon=True, name="ckpt-monitor")
+            ckpt_th.start()
         self.srv.wait_done(max_wall_s or 1800.0)
+        if ckpt_stop is not None:
+            ckpt_stop.set()
+            ckpt_th.join(timeout=30.0)
         self.srv.join()
         for th in threads:
             th.join(timeout=10.0)

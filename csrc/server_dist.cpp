@@ -25,8 +25,11 @@
 // Updates use aten tensor ops (w.add_ etc.) — identical numerics to
 // ops.torch_ref on CPU and cuBLAS-free elementwise HIP kernels on GPU; the
 // fused-update HIP kernels remain available for the single-GPU native
-// engine. Checkpoint snap sideband is not yet wired here (use the Python
-// dist engine for checkpointed runs).
+// engine. The checkpoint snap sideband (engine/dist.py protocol: snap=1 ->
+// worker sends its SAGA history table) is wired: request_alpha_snapshot /
+// wait_alpha gather remote tables mid-run; resume pushes tables back from
+// Python BEFORE start() (no channel thread is running yet, so plain
+// dist.send on the pair group is race-free).
 
 #include <torch/extension.h>
 #include <torch/csrc/distributed/c10d/ProcessGroup.hpp>
@@ -47,6 +50,26 @@ namespace {
 
 constexpr int HDR = 8;
 enum { H_TS = 0, H_K, H_ACCEPT, H_STOP, H_DELAY, H_NROWS, H_ELAPSED, H_SNAP };
+
+// ProcessGroup op ENQUEUES are serialized exactly like engine/dist.py's
+// _PG_LOCK (torch's backend bindings are not documented thread-safe for
+// concurrent enqueue); Work::wait runs OUTSIDE the lock so channels still
+// progress concurrently.
+std::mutex g_pg_mu;
+
+inline c10::intrusive_ptr<c10d::Work> pg_send(
+    const c10::intrusive_ptr<c10d::ProcessGroup>& pg,
+    std::vector<at::Tensor>& v, int dst, int tag) {
+  std::lock_guard<std::mutex> lk(g_pg_mu);
+  return pg->send(v, dst, tag);
+}
+
+inline c10::intrusive_ptr<c10d::Work> pg_recv(
+    const c10::intrusive_ptr<c10d::ProcessGroup>& pg,
+    std::vector<at::Tensor>& v, int src, int tag) {
+  std::lock_guard<std::mutex> lk(g_pg_mu);
+  return pg->recv(v, src, tag);
+}
 
 inline double now_s() {
   return std::chrono::duration<double>(
@@ -91,6 +114,7 @@ struct DSCfg {
   int64_t calib_window = 0;
   bool asaga = false;
   bool snapshot_weights = false;
+  int64_t k0 = 0, clock0 = 0;  // resume-from-checkpoint initial state
   double par_recs() const { return batch_rate * (double)N / (double)P; }
 };
 
@@ -99,24 +123,31 @@ struct DispatchMsg {
   int64_t ts = 0, k = 0;
   bool accept = true, stop = false;
   double delay_s = 0.0;
+  int snap = 0;  // 1 = fetch the peer's SAGA history (checkpoint sideband)
 };
 
 struct Slot {
   std::mutex m;
   std::condition_variable cv;
-  bool has = false;
-  DispatchMsg msg;
+  std::deque<DispatchMsg> q;  // regular round + snap ops interleave FIFO
 };
 
 class DistServer {
  public:
   DistServer(DSCfg cfg, at::Tensor w0,
              std::vector<c10::intrusive_ptr<c10d::ProcessGroup>> pair_pgs,
-             std::vector<int64_t> mark_at)
+             std::vector<int64_t> mark_at,
+             std::vector<int64_t> alpha_rows = {})
       : cfg_(cfg), w_(w0), pgs_(std::move(pair_pgs)) {
     TORCH_CHECK(w_.dtype() == at::kFloat && w_.numel() == cfg_.d);
     TORCH_CHECK((int64_t)pgs_.size() == cfg_.P - cfg_.M,
                 "one pair ProcessGroup per remote worker, wid order");
+    k_ = cfg_.k0;
+    clock_ = cfg_.clock0;
+    alpha_rows_ = alpha_rows;
+    alpha_rows_.resize(cfg_.P - cfg_.M, 0);
+    alpha_snap_.resize(cfg_.P - cfg_.M);
+    alpha_ready_.assign(cfg_.P - cfg_.M, 0);
     if (cfg_.asaga)
       alpha_bar_ = at::zeros({cfg_.d}, w_.options());
     avail_.assign(cfg_.P, 1);
@@ -137,8 +168,13 @@ class DistServer {
       opt_ms_.push_back(0), opt_w_.push_back(w_.detach().cpu().clone());
     {
       std::lock_guard<std::mutex> lk(mu_);
-      for (int64_t wid = 0; wid < cfg_.P; ++wid) pending_.push_back((int)wid);
-      try_dispatch(/*first=*/true);
+      if (k_ >= cfg_.num_iter) {  // resumed with the budget already spent
+        finish_locked();
+      } else {
+        for (int64_t wid = 0; wid < cfg_.P; ++wid)
+          pending_.push_back((int)wid);
+        try_dispatch(/*first=*/true);
+      }
     }
     for (int64_t wid = cfg_.M; wid < cfg_.P; ++wid)
       threads_.emplace_back(&DistServer::channel_loop, this, (int)wid);
@@ -164,11 +200,46 @@ class DistServer {
   local_next_dispatch(int64_t wid) {
     Slot& s = *slots_[wid];
     std::unique_lock<std::mutex> lk(s.m);
-    s.cv.wait(lk, [&] { return s.has; });
-    s.has = false;
-    DispatchMsg m = s.msg;
+    s.cv.wait(lk, [&] { return !s.q.empty(); });
+    DispatchMsg m = s.q.front();
+    s.q.pop_front();
     return {m.w, m.ts, m.k, m.accept, m.delay_s, m.stop};
   }
+
+  // ---- checkpoint sideband (SAGA history gather, engine/dist.py snap=1) --
+  void request_alpha_snapshot(int64_t wid) {
+    TORCH_CHECK(wid >= cfg_.M && wid < cfg_.P, "remote wids only");
+    TORCH_CHECK(alpha_rows_[wid - cfg_.M] > 0, "no history table for wid");
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      if (done_) {
+        // shutdown already queued a stop: the peer can no longer answer —
+        // resolve as empty so the checkpoint thread never blocks
+        alpha_snap_[wid - cfg_.M] = at::Tensor();
+        alpha_ready_[wid - cfg_.M] = 1;
+        alpha_cv_.notify_all();
+        return;
+      }
+      alpha_ready_[wid - cfg_.M] = 0;
+    }
+    DispatchMsg m;
+    m.snap = 1;
+    fill_slot((int)wid, m);
+  }
+
+  bool wait_alpha(int64_t wid, double timeout_s) {
+    std::unique_lock<std::mutex> lk(mu_);
+    return alpha_cv_.wait_for(lk, std::chrono::duration<double>(timeout_s),
+                              [&] { return alpha_ready_[wid - cfg_.M] != 0; });
+  }
+
+  at::Tensor get_alpha(int64_t wid) {
+    std::lock_guard<std::mutex> lk(mu_);
+    return alpha_snap_[wid - cfg_.M];
+  }
+
+  int64_t clock() const { return clock_; }
+  at::Tensor alpha_bar() const { return alpha_bar_; }
 
   void local_deliver(int64_t wid, at::Tensor g, int64_t ts, int64_t k_submit,
                      double elapsed_ms) {
@@ -233,9 +304,9 @@ class DistServer {
       DispatchMsg m;
       {
         std::unique_lock<std::mutex> lk(s.m);
-        s.cv.wait(lk, [&] { return s.has; });
-        s.has = false;
-        m = s.msg;
+        s.cv.wait(lk, [&] { return !s.q.empty(); });
+        m = s.q.front();
+        s.q.pop_front();
       }
       // pack (engine/messages.py layout)
       if (m.w.defined()) buf.narrow(0, 0, cfg_.d).copy_(m.w);
@@ -246,11 +317,42 @@ class DistServer {
       hdr_host[H_DELAY] = (float)m.delay_s;
       hdr_host[H_NROWS] = 0.f;
       hdr_host[H_ELAPSED] = 0.f;
-      hdr_host[H_SNAP] = 0.f;
+      hdr_host[H_SNAP] = (float)m.snap;
       buf.narrow(0, cfg_.d, HDR).copy_(hdr_host);
-      pg->send(v, /*group-rank of peer*/ 1, /*tag*/ 0)->wait();
-      if (m.stop) break;
-      pg->recv(v, 1, 0)->wait();
+      pg_send(pg, v, /*group-rank of peer*/ 1, /*tag*/ 0)->wait();
+      if (m.stop) {
+        // resolve any snaps queued behind the stop as empty. Lock order:
+        // never hold s.m while taking mu_ (finish_locked holds mu_ and
+        // fills slots -> s.m), so inspect the queue first, then mark.
+        bool pending_snap = false;
+        {
+          std::lock_guard<std::mutex> lk(s.m);
+          for (auto& rest : s.q)
+            if (rest.snap == 1) pending_snap = true;
+          s.q.clear();
+        }
+        if (pending_snap) {
+          std::lock_guard<std::mutex> lk2(mu_);
+          alpha_snap_[wid - cfg_.M] = at::Tensor();
+          alpha_ready_[wid - cfg_.M] = 1;
+        }
+        alpha_cv_.notify_all();
+        break;
+      }
+      if (m.snap == 1) {
+        // checkpoint sideband: the peer replies with its history table
+        at::Tensor ab = at::zeros({alpha_rows_[wid - cfg_.M]}, w_.options());
+        std::vector<at::Tensor> av{ab};
+        pg_recv(pg, av, 1, 0)->wait();
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          alpha_snap_[wid - cfg_.M] = ab.cpu();
+          alpha_ready_[wid - cfg_.M] = 1;
+        }
+        alpha_cv_.notify_all();
+        continue;
+      }
+      pg_recv(pg, v, 1, 0)->wait();
       // header comes back to host (this is also the completion sync point)
       at::Tensor h = buf.narrow(0, cfg_.d, HDR).cpu();
       const float* hp = h.data_ptr<float>();
@@ -267,6 +369,10 @@ class DistServer {
     (void)elapsed_ms;
     std::lock_guard<std::mutex> lk(mu_);
     if (done_) return;  // late result after shutdown: dropped (as in Python)
+    if (k_ >= cfg_.num_iter) {  // budget already reached (Python loop
+      finish_locked();          // checks BEFORE processing — mirror that)
+      return;
+    }
     const double t_now = now_s();
     const int64_t staleness = clock_ - ts;  // arrival clock
     clock_ += 1;
@@ -353,8 +459,7 @@ class DistServer {
   void fill_slot(int wid, const DispatchMsg& m) {
     Slot& s = *slots_[wid];
     std::lock_guard<std::mutex> lk(s.m);
-    s.msg = m;
-    s.has = true;
+    s.q.push_back(m);
     s.cv.notify_one();
   }
 
@@ -407,6 +512,12 @@ class DistServer {
   std::vector<std::unique_ptr<Slot>> slots_;
   std::vector<std::thread> threads_;
 
+  // LOCKING RULE: no pybind entry point may BLOCK on mu_ (or a Slot
+  // mutex) while holding the GIL — C++ threads holding mu_ can need the
+  // GIL transiently (destroying an at::Tensor whose Python wrapper exists
+  // acquires the GIL for the pyobj decref), so a GIL-holding mu_ waiter
+  // closes a deadlock cycle. Every binding that touches mu_/slots is
+  // wrapped in py::gil_scoped_release; plain getters read fields lock-free.
   std::mutex mu_;
   std::condition_variable done_cv_;
   bool done_ = false;
@@ -418,6 +529,10 @@ class DistServer {
   std::vector<double> submit_t_, finish_t_;
   std::vector<int64_t> waiting_ms_;
   std::set<int64_t> mark_at_;
+  std::vector<int64_t> alpha_rows_;
+  std::vector<at::Tensor> alpha_snap_;
+  std::vector<uint8_t> alpha_ready_;
+  std::condition_variable alpha_cv_;
   std::map<int64_t, double> marks_;
   std::vector<int64_t> opt_ms_;
   std::vector<at::Tensor> opt_w_;
@@ -448,12 +563,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def_readwrite("seed", &DSCfg::seed)
       .def_readwrite("calib_window", &DSCfg::calib_window)
       .def_readwrite("asaga", &DSCfg::asaga)
-      .def_readwrite("snapshot_weights", &DSCfg::snapshot_weights);
+      .def_readwrite("snapshot_weights", &DSCfg::snapshot_weights)
+      .def_readwrite("k0", &DSCfg::k0)
+      .def_readwrite("clock0", &DSCfg::clock0);
 
   py::class_<DistServer>(m, "DistServer")
       .def(py::init<DSCfg, at::Tensor,
                     std::vector<c10::intrusive_ptr<c10d::ProcessGroup>>,
-                    std::vector<int64_t>>())
+                    std::vector<int64_t>, std::vector<int64_t>>(),
+           py::arg("cfg"), py::arg("w0"), py::arg("pair_pgs"),
+           py::arg("mark_at"),
+           py::arg("alpha_rows") = std::vector<int64_t>{})
       .def("start", &DistServer::start,
            py::call_guard<py::gil_scoped_release>())
       .def("wait_done", &DistServer::wait_done,
@@ -476,5 +596,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("max_staleness_seen", &DistServer::max_staleness_seen)
       .def("delay_active", &DistServer::delay_active)
       .def("avg_delay_ms", &DistServer::avg_delay_ms)
+      .def("clock", &DistServer::clock)
+      .def("alpha_bar", &DistServer::alpha_bar)
+      .def("request_alpha_snapshot", &DistServer::request_alpha_snapshot,
+           py::call_guard<py::gil_scoped_release>())
+      .def("wait_alpha", &DistServer::wait_alpha,
+           py::call_guard<py::gil_scoped_release>())
+      .def("get_alpha", &DistServer::get_alpha,
+           py::call_guard<py::gil_scoped_release>())
       .def_static("delay_probe", &DistServer::delay_probe);
 }

@@ -289,3 +289,74 @@ def test_native_dist_world1_cpu():
 @pytest.mark.timeout(300)
 def test_native_dist_world1_gpu():
     _world1_run("cuda:0")
+
+
+def _ckpt_rank_main(rank, init_file, ck_path, out_file, resume):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        cfg = EngineConfig(d=24, N=400, num_workers=4,
+                           num_iterations=3000, gamma=0.05, taw=2 ** 30,
+                           batch_rate=0.3, bucket_ratio=0.5,
+                           printer_freq=1 << 30, delay_coeff=0.0, seed=42,
+                           device="cpu", sync=False, algo="asaga",
+                           snapshot_weights=False)
+        if resume:
+            from asyncframework_amd.engine.checkpoint import load_checkpoint
+            cfg.num_iterations = load_checkpoint(ck_path)["k"]
+        else:
+            cfg.checkpoint_path = ck_path
+            cfg.checkpoint_every = 500
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+        M = 2
+        workers = []
+        for j in range(M):
+            wid = rank * M + j
+            s, t = row_shards(cfg.N, 4)[wid]
+            workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                             X=X[s:t], y=y[s:t]), cfg,
+                                  device=torch.device("cpu")))
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=120,
+                      resume_from=ck_path if resume else "")
+        if resume and rank == 1:
+            torch.save([w.alpha.clone() for w in workers],
+                       out_file + ".alpha1")
+        if rank == 0:
+            torch.save({"k": res.k, "w": res.w.clone()}, out_file)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_checkpoint_and_resume(tmp_path):
+    """The C++ snap sideband: remote SAGA tables land in mid-run
+    checkpoints; resume pushes everything back (checkpoints share the
+    threads-engine schema)."""
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    ck = str(tmp_path / "n.ckpt")
+    out1 = str(tmp_path / "o1")
+    mp.spawn(_ckpt_rank_main, args=(str(tmp_path / "i1"), ck, out1, False),
+             nprocs=WORLD, join=True)
+    state = load_checkpoint(ck)
+    assert set(state["alpha"].keys()) == {0, 1, 2, 3}
+    for wid, (s, t) in enumerate(row_shards(400, 4)):
+        assert state["alpha"][wid].shape == (t - s,)
+    assert float(sum(a.abs().sum() for a in state["alpha"].values())) > 0
+    # the native monitor snapshots asynchronously at-or-after each
+    # checkpoint_every boundary (k keeps advancing during the gather),
+    # unlike the threads engine's exact-multiple synchronous snapshots
+    assert state["k"] >= 500
+    assert state["current_time"] >= state["k"]
+    assert state["alpha_bar"] is not None
+
+    out2 = str(tmp_path / "o2")
+    mp.spawn(_ckpt_rank_main, args=(str(tmp_path / "i2"), ck, out2, True),
+             nprocs=WORLD, join=True)
+    r = torch.load(out2, weights_only=False)
+    # resumed with budget == checkpoint k: no further updates applied
+    assert r["k"] == state["k"]
+    assert torch.allclose(r["w"], state["w"])
+    alpha1 = torch.load(out2 + ".alpha1", weights_only=False)
+    assert torch.equal(alpha1[0], state["alpha"][2])
+    assert torch.equal(alpha1[1], state["alpha"][3])
