@@ -183,11 +183,10 @@ def _run_dist(cfg: EngineConfig, a, app: str, names, vals,
         return
     if a.engine == "native":
         assert not cfg.sync, "--engine native under torchrun is async-only"
-        assert not cfg.checkpoint_path and not a.resume_from, \
-            "checkpointing under torchrun needs --engine threads"
         from ..engine.dist_native import NativeDistEngine
         eng = NativeDistEngine(cfg, workers, device)
-        res = eng.run(max_wall_s=a.max_wall_s, verbose=(rank == 0))
+        res = eng.run(max_wall_s=a.max_wall_s, verbose=(rank == 0),
+                      resume_from=a.resume_from)
     else:
         from ..engine.dist import DistEngine
         eng = DistEngine(cfg, workers, device)
