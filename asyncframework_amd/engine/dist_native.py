@@ -100,6 +100,8 @@ class NativeDistEngine:
         c.snapshot_weights = cfg.snapshot_weights
         c.k0 = k0
         c.clock0 = clock0
+        c.bucket_ratio = cfg.bucket_ratio
+        c.worker_timeout_s = cfg.worker_timeout_s
         return c
 
     def _local_worker_loop(self, wid: int, worker: Worker):

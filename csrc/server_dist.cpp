@@ -115,6 +115,9 @@ struct DSCfg {
   bool asaga = false;
   bool snapshot_weights = false;
   int64_t k0 = 0, clock0 = 0;  // resume-from-checkpoint initial state
+  double bucket_ratio = 0.93;   // for the alive-scaled quorum gate
+  double worker_timeout_s = 0;  // failure detection (0 = off), matching
+                                // engine/local.py::_reap_dead_workers
   double par_recs() const { return batch_rate * (double)N / (double)P; }
 };
 
@@ -151,6 +154,7 @@ class DistServer {
     if (cfg_.asaga)
       alpha_bar_ = at::zeros({cfg_.d}, w_.options());
     avail_.assign(cfg_.P, 1);
+    dead_.assign(cfg_.P, 0);
     last_accept_.assign(cfg_.P, 1);
     submit_t_.assign(cfg_.P, 0.0);
     finish_t_.assign(cfg_.P, 0.0);
@@ -181,17 +185,43 @@ class DistServer {
   }
 
   bool wait_done(double timeout_s) {
+    const double deadline = now_s() + timeout_s;
     std::unique_lock<std::mutex> lk(mu_);
-    done_cv_.wait_for(lk, std::chrono::duration<double>(timeout_s),
-                      [&] { return done_; });
+    while (!done_ && now_s() < deadline) {
+      done_cv_.wait_for(lk, std::chrono::milliseconds(100),
+                        [&] { return done_; });
+      if (!done_ && cfg_.worker_timeout_s > 0) reap_dead_locked();
+    }
     if (!done_) finish_locked();  // wall-clock cap: stop everything
     return done_;
   }
 
   void join() {
-    for (auto& t : threads_)
-      if (t.joinable()) t.join();
+    // a dead peer's channel thread can be blocked in recv forever — wait
+    // up to 10 s for clean exits, then detach the stragglers (they only
+    // wake, if ever, while this object is still alive in the engine)
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      threads_exit_cv_.wait_for(lk, std::chrono::seconds(10), [&] {
+        return threads_exited_ >= (int64_t)threads_.size();
+      });
+    }
+    for (auto& t : threads_) {
+      if (!t.joinable()) continue;
+      if (threads_exited_ >= (int64_t)threads_.size())
+        t.join();
+      else
+        t.detach();
+    }
     threads_.clear();
+  }
+
+  std::string channel_error() const { return channel_error_; }
+
+  int64_t dead_workers() const {
+    int64_t n = 0;
+    for (auto d : dead_) n += d;
+    return n;
   }
 
   // ---- rank-0-local worker API (called from Python worker threads) -------
@@ -295,6 +325,28 @@ class DistServer {
  private:
   // ---- remote channel: one thread per remote worker ----------------------
   void channel_loop(int wid) {
+    try {
+      channel_loop_body(wid);
+    } catch (const std::exception& e) {
+      // transport failure (e.g. a dead peer closed the connection while we
+      // were blocked in recv): the worker stays/becomes declared dead and
+      // the channel retires — never let the exception reach
+      // std::terminate on a detached thread
+      std::lock_guard<std::mutex> lk(mu_);
+      if (wid >= 0 && wid < (int)dead_.size()) dead_[wid] = 1;
+      channel_error_ = e.what();
+    } catch (...) {
+      std::lock_guard<std::mutex> lk(mu_);
+      channel_error_ = "unknown channel transport error";
+    }
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      threads_exited_ += 1;
+    }
+    threads_exit_cv_.notify_all();
+  }
+
+  void channel_loop_body(int wid) {
     auto pg = pgs_[wid - cfg_.M];
     at::Tensor buf = at::zeros({cfg_.d + HDR}, w_.options());
     at::Tensor hdr_host = at::zeros({HDR}, at::kFloat);
@@ -378,6 +430,7 @@ class DistServer {
     clock_ += 1;
     max_staleness_ = std::max(max_staleness_, staleness);
     avail_[wid] = 1;
+    dead_[wid] = 0;  // a late result resurrects a declared-dead worker
     const bool accept =
         cfg_.asaga ? (k_ - ts) <= cfg_.taw : staleness <= cfg_.taw;
     if (accept) {
@@ -424,13 +477,33 @@ class DistServer {
     }
   }
 
+  // ---- failure detection (engine/local.py::_reap_dead_workers) -----------
+  void reap_dead_locked() {
+    const double now = now_s();
+    bool changed = false;
+    for (int64_t wid = 0; wid < cfg_.P; ++wid) {
+      if (dead_[wid] || avail_[wid]) continue;
+      if (submit_t_[wid] > 0 &&
+          now - submit_t_[wid] > cfg_.worker_timeout_s) {
+        dead_[wid] = 1;
+        changed = true;
+      }
+    }
+    if (changed) try_dispatch(false);  // a gate stall may now unblock
+  }
+
   // ---- dispatch (engine/local.py::_dispatch_pending) ---------------------
   void try_dispatch(bool first) {
     if (pending_.empty()) return;
-    int64_t avail_n = 0;
+    int64_t avail_n = 0, ndead = 0;
     for (auto a : avail_) avail_n += a;
+    for (auto d : dead_) ndead += d;
+    const int64_t alive = cfg_.P - ndead;
+    const int64_t gate = std::min(
+        cfg_.gate,
+        std::max((int64_t)1, (int64_t)(alive * cfg_.bucket_ratio)));
     const int64_t init_workers = first ? cfg_.P : avail_n;
-    if (init_workers < cfg_.gate) return;
+    if (init_workers < gate) return;
     if (!delay_flag_ && k_ > cfg_.calib_window) {
       if (cul_count_ > 0) avg_delay_ms_ = cul_time_ms_ / (double)cul_count_;
       delay_flag_ = true;
@@ -524,7 +597,10 @@ class DistServer {
   int64_t k_ = 0, clock_ = 0, applied_ = 0, rejected_ = 0;
   int64_t max_staleness_ = -1;
   int64_t elapsed_final_ms_ = 0;
-  std::vector<uint8_t> avail_, last_accept_;
+  std::vector<uint8_t> avail_, last_accept_, dead_;
+  int64_t threads_exited_ = 0;
+  std::condition_variable threads_exit_cv_;
+  std::string channel_error_;
   std::deque<int> pending_;
   std::vector<double> submit_t_, finish_t_;
   std::vector<int64_t> waiting_ms_;
@@ -565,7 +641,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def_readwrite("asaga", &DSCfg::asaga)
       .def_readwrite("snapshot_weights", &DSCfg::snapshot_weights)
       .def_readwrite("k0", &DSCfg::k0)
-      .def_readwrite("clock0", &DSCfg::clock0);
+      .def_readwrite("clock0", &DSCfg::clock0)
+      .def_readwrite("bucket_ratio", &DSCfg::bucket_ratio)
+      .def_readwrite("worker_timeout_s", &DSCfg::worker_timeout_s);
 
   py::class_<DistServer>(m, "DistServer")
       .def(py::init<DSCfg, at::Tensor,
@@ -594,6 +672,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("opt_w", &DistServer::opt_w)
       .def("weights", &DistServer::weights)
       .def("max_staleness_seen", &DistServer::max_staleness_seen)
+      .def("dead_workers", &DistServer::dead_workers)
+      .def("channel_error", &DistServer::channel_error)
       .def("delay_active", &DistServer::delay_active)
       .def("avg_delay_ms", &DistServer::avg_delay_ms)
       .def("clock", &DistServer::clock)

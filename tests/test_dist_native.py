@@ -360,3 +360,71 @@ def test_native_dist_checkpoint_and_resume(tmp_path):
     alpha1 = torch.load(out2 + ".alpha1", weights_only=False)
     assert torch.equal(alpha1[0], state["alpha"][2])
     assert torch.equal(alpha1[1], state["alpha"][3])
+
+
+def _faildet_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    # bucket_ratio=1.0: the gate demands ALL FOUR workers, so once wid 3
+    # goes silent every dispatch stalls — only the reaper (worker_timeout_s)
+    # can unblock the run by shrinking the alive pool
+    cfg = EngineConfig(d=16, N=400, num_workers=4, num_iterations=60,
+                       gamma=0.3, taw=2 ** 30, batch_rate=0.3,
+                       bucket_ratio=1.0, printer_freq=1 << 30,
+                       delay_coeff=0.0, seed=42, device="cpu", sync=False,
+                       algo="asgd", snapshot_weights=False,
+                       worker_timeout_s=0.5)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+    M = 2
+    workers = []
+    for j in range(M):
+        wid = rank * M + j
+        s, t = row_shards(cfg.N, 4)[wid]
+        workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                         X=X[s:t], y=y[s:t]), cfg,
+                              device=torch.device("cpu")))
+    if rank == 0:
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=60)
+        with open(out_file, "w") as f:
+            f.write(f"{res.k},{eng.srv.dead_workers()}")
+        # do NOT destroy the pg: the dead peer's channel never drains and
+        # destroy could block; process teardown cleans up (failure path)
+        os._exit(0)
+    else:
+        # worker rank with a BLACK-HOLE worker: wid 3 accepts one dispatch
+        # and never replies (the lost-task scenario the reference leaves
+        # hanging forever)
+        import threading as _th
+        from asyncframework_amd.engine.dist import (_recv,
+                                                    remote_worker_loop)
+        from asyncframework_amd.engine.messages import HDR
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+
+        def black_hole():
+            buf = torch.zeros(cfg.d + HDR, dtype=torch.float32)
+            _recv(buf, 0, eng.base.pair_groups[3])
+            import time as _t
+            _t.sleep(3600)
+
+        th_dead = _th.Thread(target=black_hole, daemon=True)
+        th_dead.start()
+        remote_worker_loop(workers[0], cfg, eng.base.pair_groups[2],
+                           torch.device("cpu"))
+        dist.barrier()  # meet rank 0's end-of-run barrier
+        os._exit(0)
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_failure_detection():
+    """A lost remote worker is declared dead after worker_timeout_s and
+    excluded from the quorum gate; the run completes on the survivors."""
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "init")
+        out_file = os.path.join(td, "out")
+        mp.spawn(_faildet_rank_main, args=(init_file, out_file),
+                 nprocs=WORLD, join=True)
+        with open(out_file) as f:
+            k, dead = map(int, f.read().split(","))
+        assert k >= 60
+        assert dead == 1
