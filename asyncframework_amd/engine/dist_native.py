@@ -102,6 +102,8 @@ class NativeDistEngine:
         c.clock0 = clock0
         c.bucket_ratio = cfg.bucket_ratio
         c.worker_timeout_s = cfg.worker_timeout_s
+        from ..utils.trace import get_tracer
+        c.trace = get_tracer() is not None
         return c
 
     def _local_worker_loop(self, wid: int, worker: Worker):
@@ -231,6 +233,16 @@ class NativeDistEngine:
         for th in threads:
             th.join(timeout=10.0)
         k = self.srv.k()
+        from ..utils.trace import get_tracer, stop_trace
+        tr = get_tracer()
+        if tr is not None:
+            names = {0: "dispatch", 1: "accept", 2: "reject"}
+            for ts_s, wid, kind, kk, st in self.srv.trace_events():
+                tr.instant(names[int(kind)], 0,
+                           args={"k": int(kk), "wid": int(wid),
+                                 "staleness": int(st)},
+                           ts_us=tr.monotonic_s_to_us(ts_s))
+            stop_trace()
         if verbose:
             for i in range(0, k, cfg.printer_freq):
                 print(f"Iteration {i} is finished")

@@ -54,13 +54,20 @@ class Tracer:
 
     def instant(self, name: str, tid: int,
                 args: Optional[Dict[str, Any]] = None,
-                pid: str = "server") -> None:
-        ev = {"name": name, "ph": "i", "ts": _now_us(), "s": "t",
+                pid: str = "server",
+                ts_us: Optional[float] = None) -> None:
+        ev = {"name": name, "ph": "i",
+              "ts": _now_us() if ts_us is None else ts_us, "s": "t",
               "pid": pid, "tid": tid}
         if args:
             ev["args"] = args
         with self._lock:
             self._events.append(ev)
+
+    def monotonic_s_to_us(self, t_s: float) -> float:
+        """Convert a CLOCK_MONOTONIC-epoch stamp in seconds (C++
+        steady_clock — same clock as perf_counter) to this trace's us."""
+        return t_s * 1e6 - _T0_NS / 1000.0
 
     def now_us(self) -> float:
         return _now_us()

@@ -116,6 +116,7 @@ struct DSCfg {
   bool snapshot_weights = false;
   int64_t k0 = 0, clock0 = 0;  // resume-from-checkpoint initial state
   double bucket_ratio = 0.93;   // for the alive-scaled quorum gate
+  bool trace = false;           // record dispatch/accept/reject events
   double worker_timeout_s = 0;  // failure detection (0 = off), matching
                                 // engine/local.py::_reap_dead_workers
   double par_recs() const { return batch_rate * (double)N / (double)P; }
@@ -217,6 +218,18 @@ class DistServer {
   }
 
   std::string channel_error() const { return channel_error_; }
+
+  // (ts_seconds_monotonic, wid, kind 0=dispatch/1=accept/2=reject, k,
+  //  staleness) — merged into the Perfetto log by the Python adapter
+  std::vector<std::tuple<double, int64_t, int64_t, int64_t, int64_t>>
+  trace_events() {
+    std::lock_guard<std::mutex> lk(mu_);
+    std::vector<std::tuple<double, int64_t, int64_t, int64_t, int64_t>> out;
+    out.reserve(trace_ev_.size());
+    for (auto& e : trace_ev_)
+      out.emplace_back(e.ts, e.wid, e.kind, e.k, e.staleness);
+    return out;
+  }
 
   int64_t dead_workers() const {
     int64_t n = 0;
@@ -433,6 +446,10 @@ class DistServer {
     dead_[wid] = 0;  // a late result resurrects a declared-dead worker
     const bool accept =
         cfg_.asaga ? (k_ - ts) <= cfg_.taw : staleness <= cfg_.taw;
+    if (cfg_.trace)
+      trace_ev_.push_back({t_now, (int32_t)wid, accept ? (int8_t)1
+                                                       : (int8_t)2,
+                           (int32_t)k_, (int32_t)staleness});
     if (accept) {
       finish_t_[wid] = t_now;
       // delay calibration sample (reference :177-186; Python records wall
@@ -518,6 +535,9 @@ class DistServer {
       waiting_ms_[wid] += (int64_t)((t_now - prev_fin) * 1000.0);
       submit_t_[wid] = t_now;
       avail_[wid] = 0;
+      if (cfg_.trace)
+        trace_ev_.push_back({t_now, (int32_t)wid, (int8_t)0, (int32_t)k_,
+                             0});
       DispatchMsg m;
       m.w = w_snap;
       m.ts = clock_;
@@ -601,6 +621,13 @@ class DistServer {
   int64_t threads_exited_ = 0;
   std::condition_variable threads_exit_cv_;
   std::string channel_error_;
+  struct TraceEv {
+    double ts;
+    int32_t wid;
+    int8_t kind;
+    int32_t k, staleness;
+  };
+  std::vector<TraceEv> trace_ev_;
   std::deque<int> pending_;
   std::vector<double> submit_t_, finish_t_;
   std::vector<int64_t> waiting_ms_;
@@ -643,6 +670,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def_readwrite("k0", &DSCfg::k0)
       .def_readwrite("clock0", &DSCfg::clock0)
       .def_readwrite("bucket_ratio", &DSCfg::bucket_ratio)
+      .def_readwrite("trace", &DSCfg::trace)
       .def_readwrite("worker_timeout_s", &DSCfg::worker_timeout_s);
 
   py::class_<DistServer>(m, "DistServer")
@@ -674,6 +702,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("max_staleness_seen", &DistServer::max_staleness_seen)
       .def("dead_workers", &DistServer::dead_workers)
       .def("channel_error", &DistServer::channel_error)
+      .def("trace_events", &DistServer::trace_events,
+           py::call_guard<py::gil_scoped_release>())
       .def("delay_active", &DistServer::delay_active)
       .def("avg_delay_ms", &DistServer::avg_delay_ms)
       .def("clock", &DistServer::clock)

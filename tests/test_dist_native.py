@@ -428,3 +428,32 @@ def test_native_dist_failure_detection():
             k, dead = map(int, f.read().split(","))
         assert k >= 60
         assert dead == 1
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_trace_events(tmp_path):
+    """With ASYNCAMD_TRACE active, the C++ server's dispatch/accept events
+    are merged into the Perfetto log (world=1, all-local)."""
+    import json
+
+    from asyncframework_amd.utils import trace
+    p = str(tmp_path / "nd_trace.json")
+    trace.start_trace(p)
+    try:
+        _world1_run("cpu")
+    finally:
+        trace.stop_trace()
+    with open(p) as f:
+        evs = json.load(f)["traceEvents"]
+    names = {e["name"] for e in evs}
+    assert {"dispatch", "accept"} <= names
+    accepts = [e for e in evs if e["name"] == "accept"]
+    assert len(accepts) >= 80
+    assert all(e["args"]["staleness"] >= 0 for e in accepts)
+    # worker-side spans come from the Python Worker hook on the same clock
+    rounds = [e for e in evs if e["name"] == "round"]
+    assert len(rounds) >= 80
+    # the merged clocks must interleave sensibly (same monotonic base)
+    ats = sorted(e["ts"] for e in accepts)
+    rts = sorted(e["ts"] for e in rounds)
+    assert abs(ats[0] - rts[0]) < 60_000_000  # within a minute
