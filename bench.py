@@ -79,6 +79,10 @@ def parse_args():
                    choices=["", "graph", "threads", "native"])
     p.add_argument("--workers", type=int, default=0,
                    help="logical workers (threads/native engines)")
+    p.add_argument("--dist-engine", default="",
+                   choices=["", "python", "native"],
+                   help="N>1 control plane: python (default) or the C++ "
+                        "server (also via ASYNCAMD_DIST_ENGINE=native)")
     args = p.parse_args()
     preset = MODELS[args.model]
     args.rows = args.rows or preset["rows"]
@@ -248,7 +252,9 @@ def run_dist(args, device, rank, world):
     dist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
-    if os.environ.get("ASYNCAMD_DIST_ENGINE", "") == "native":
+    dist_engine = (args.dist_engine
+                   or os.environ.get("ASYNCAMD_DIST_ENGINE", ""))
+    if dist_engine == "native":
         # opt-in C++ rank-0 server (csrc/server_dist.cpp); wire-compatible
         # with the Python path — flip the default after round-2 multi-GPU
         # validation
