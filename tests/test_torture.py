@@ -48,3 +48,55 @@ def test_torture_tau_zero_many_workers():
     res = eng.run(max_wall_s=120)
     assert res.k >= cfg.num_iterations
     assert max(eng.accepted_staleness) <= 0
+
+
+def test_torture_delay_plus_tight_tau():
+    """SURVEY §5.2's torture recipe: straggler injection (cloud model,
+    tiny calibration window) + tight tau together; the run must still
+    reach its iteration budget and never apply an over-tau result."""
+    from asyncframework_amd.data.synthetic import synthetic_dense
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.engine.local import AsyncEngine
+    from asyncframework_amd.run import build_dense_workers
+    cfg = EngineConfig(d=12, N=240, num_workers=6, num_iterations=60,
+                       gamma=0.1, taw=2, batch_rate=0.2, bucket_ratio=0.3,
+                       printer_freq=1 << 30, delay_coeff=-1.0, seed=3,
+                       device="cpu", snapshot_weights=False,
+                       calib_factor=2)  # calibration after 12 tasks
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=9)
+    eng = AsyncEngine(cfg, build_dense_workers(cfg, X, y))
+    eng.verbose = False
+    res = eng.run(max_wall_s=120)
+    assert res.k >= cfg.num_iterations
+    assert max(eng.accepted_staleness) <= cfg.taw
+    assert eng.delay.flag  # calibration really activated mid-run
+
+
+def test_concurrent_context_hammer():
+    """ASYNCcontext under concurrent producers + clock bumps: no lost
+    results, monotone clock (the mailbox is the one synchronized point,
+    as in the reference's JobWaiter.taskSucceeded)."""
+    import threading
+
+    from asyncframework_amd.core.context import ASYNCcontext, RDDPartialRes
+    AC = ASYNCcontext()
+    NP, PER = 8, 500
+
+    def producer(pid):
+        for i in range(PER):
+            AC.put(RDDPartialRes(data=(pid, i), ts=0, id=pid))
+            AC.add2currentTime(1)
+
+    threads = [threading.Thread(target=producer, args=(p,))
+               for p in range(NP)]
+    for t in threads:
+        t.start()
+    seen = set()
+    for _ in range(NP * PER):
+        pr = AC.ASYNCcollectAll(timeout=30)
+        seen.add(pr.gettaskResult())
+    for t in threads:
+        t.join()
+    assert len(seen) == NP * PER
+    assert AC.getCurrentTime() == NP * PER
+    assert not AC.hasNext()
