@@ -112,6 +112,11 @@ class AsyncEngine:
         self._stop = threading.Event()
         self._pending_ev = threading.Event()
         self._pending_ev.set()
+        # brief GIL-yielding spin before the blocking mailbox wait: wins
+        # ~100 us/update when results arrive back-to-back (single-process
+        # GPU engines), but CONTENDS the GIL against co-located worker and
+        # proxy threads when rounds are long — dist mode disables it
+        self.spin_budget = 100
         self.staleness_seen: List[int] = []
         self.accepted_staleness: List[int] = []
         self.applied = 0
@@ -271,7 +276,7 @@ class AsyncEngine:
             # mutex and contends with the producing worker threads.
             _dq = srv.AC.ResultList.queue
             spin = 0
-            while not _dq and spin < 100:
+            while not _dq and spin < self.spin_budget:
                 time.sleep(0)
                 spin += 1
             try:
