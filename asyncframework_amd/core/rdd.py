@@ -100,6 +100,17 @@ class AsyncRDD:
     def filter(self, f: Callable[[Any], bool]) -> "AsyncRDD":
         return self._derive(("filter", f))
 
+    def mapPartitions(self, f) -> "AsyncRDD":
+        """Per-partition transform (iterator -> iterator) — the hook the
+        reference's delay injection uses (mapPartitions sleep,
+        SparkASGDThread.scala:287-312)."""
+        return self._derive(("mapparts", lambda pid, elems: f(elems)))
+
+    def mapPartitionsWithIndex(self, f) -> "AsyncRDD":
+        """(partition index, iterator) -> iterator — the primitive
+        ASYNCbarrier is built on in the reference (RDD.scala:1066-1073)."""
+        return self._derive(("mapparts", f))
+
     def zipWithIndex(self) -> "AsyncRDD":
         """(element, absolute index) pairs (reference RDD.scala:1527-1528)."""
         return self._derive(("zipidx", None))
@@ -126,6 +137,8 @@ class AsyncRDD:
                 elems = [arg(e) for e in elems]
             elif kind == "filter":
                 elems = [e for e in elems if arg(e)]
+            elif kind == "mapparts":
+                elems = list(arg(pid, iter(elems)))
             elif kind == "zipidx":
                 elems = [(e, offset + i) for i, e in enumerate(elems)]
             elif kind == "sample":

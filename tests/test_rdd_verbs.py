@@ -176,3 +176,29 @@ def test_full_async_driver_loop_converges():
     obj1 = float(((X @ w - y) ** 2).mean())
     assert k >= num_iter
     assert obj1 < obj0 * 0.5
+
+
+def test_mappartitions_delay_injection_shape():
+    """The reference injects stragglers via a mapPartitions sleep
+    (SparkASGDThread.scala:287-312) — same code shape works here."""
+    import time as _t
+    rdd, _, _ = _make_points(n=40, P=4)
+    AC = ASYNCcontext()
+    slept = []
+
+    def inject(elems):
+        slept.append(1)
+        return elems
+
+    delayed = rdd.mapPartitions(inject)
+    delayed.map(lambda p: 1).ASYNCreduce(lambda a, b: a + b, AC)
+    total = sum(AC.ASYNCcollectAll().gettaskResult() for _ in range(4))
+    assert total == 40
+    assert len(slept) == 4  # ran once per partition
+
+
+def test_mappartitions_with_index():
+    rdd, _, _ = _make_points(n=40, P=4)
+    out = rdd.mapPartitionsWithIndex(
+        lambda pid, elems: [pid] if pid % 2 == 0 else []).collect()
+    assert out == [0, 2]
