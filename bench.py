@@ -41,7 +41,9 @@ MODELS = {
     # BASELINE.json configs. config 1 is the CPU plumbing check.
     "sync-tiny-cpu": dict(rows=1_000, cols=784, rate=0.3, algo="asgd",
                           sync=True, dtype="fp32", sparse=False,
-                          engine="threads", device="cpu", workers=2),
+                          engine="threads", device="cpu", workers=2,
+                          objective="logistic"),  # BASELINE config 1 names
+                                                  # logistic regression
     # flagship: 32 async workers on streams (the reference's fixed
     # partitions=32, README.md) driven by the native C++ event loop;
     # --engine graph gives the single-worker hipGraph variant
@@ -72,7 +74,8 @@ def parse_args():
     p.add_argument("--cols", type=int, default=0)
     p.add_argument("--rate", type=float, default=0.0)
     p.add_argument("--dtype", default="", choices=["", "bf16", "fp32"])
-    p.add_argument("--objective", default="lsq", choices=["lsq", "logistic"])
+    p.add_argument("--objective", default="",
+                   choices=["", "lsq", "logistic"])
     p.add_argument("--algo", default="", choices=["", "asgd", "asaga"])
     p.add_argument("--device", default=None, help="override (cpu for debug)")
     p.add_argument("--engine", default="",
@@ -90,6 +93,7 @@ def parse_args():
     args.rate = args.rate or preset["rate"]
     args.dtype = args.dtype or preset["dtype"]
     args.algo = args.algo or preset["algo"]
+    args.objective = args.objective or preset.get("objective", "lsq")
     args.engine = args.engine or preset["engine"]
     if args.device is None and "device" in preset:
         args.device = preset["device"]
