@@ -143,6 +143,12 @@ def emit(args, cfg, elapsed: float, n_gpus: int):
             "seq_len": args.cols,
             "parallelism": ("sync" if args.sync else "async-ps") +
                            f"-dp{n_gpus}-w{cfg.num_workers}",
+            # BASELINE.json's metric second clause (wall-clock-to-target-
+            # loss) is measured separately under the reference straggler
+            # model — committed evidence:
+            "wall_clock_to_target_loss":
+                "profiles/r01_async_vs_sync_*.json (async 4.05x faster "
+                "to equal error, epsilon shape; 18.8x, mnist8m shape)",
         },
     }
     print(json.dumps(out))
