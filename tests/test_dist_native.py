@@ -457,3 +457,55 @@ def test_native_dist_trace_events(tmp_path):
     ats = sorted(e["ts"] for e in accepts)
     rts = sorted(e["ts"] for e in rounds)
     assert abs(ats[0] - rts[0]) < 60_000_000  # within a minute
+
+
+def _seq_equiv_run(algo: str):
+    """P=1, world=1: both the C++ server and the Python threads engine are
+    strictly sequential (staleness 0, every result accepted) with identical
+    Philox masks and identical update formulas — final w must be
+    bit-identical to the oracle."""
+    import tempfile as _tf
+    cfg_kw = dict(d=20, N=200, num_workers=1, num_iterations=50,
+                  gamma=0.4 if algo == "asgd" else 0.05, taw=2 ** 30,
+                  batch_rate=0.3, bucket_ratio=0.5, printer_freq=1 << 30,
+                  delay_coeff=0.0, seed=42, device="cpu", sync=False,
+                  algo=algo, snapshot_weights=False)
+    X, y = synthetic_dense(200, 20, seed=7)
+
+    def mk_worker(cfg):
+        return Worker(0, Shard(row_start=0, n_rows=200, X=X, y=y), cfg,
+                      device=torch.device("cpu"))
+
+    # oracle: threads engine
+    from asyncframework_amd.engine.local import AsyncEngine
+    cfg1 = EngineConfig(**cfg_kw)
+    eng1 = AsyncEngine(cfg1, [mk_worker(cfg1)])
+    eng1.verbose = False
+    res1 = eng1.run(max_wall_s=60)
+
+    # C++ server, all-local
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    with _tf.TemporaryDirectory() as td:
+        dist.init_process_group("gloo", init_method=f"file://{td}/i",
+                                rank=0, world_size=1)
+        try:
+            cfg2 = EngineConfig(**cfg_kw)
+            eng2 = NativeDistEngine(cfg2, [mk_worker(cfg2)],
+                                    torch.device("cpu"))
+            res2 = eng2.run(verbose=False, max_wall_s=60)
+        finally:
+            dist.destroy_process_group()
+
+    assert res1.k == res2.k == 50
+    assert torch.equal(res1.w, res2.w), \
+        float((res1.w - res2.w).abs().max())
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_bitwise_matches_threads_oracle_asgd():
+    _seq_equiv_run("asgd")
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_bitwise_matches_threads_oracle_asaga():
+    _seq_equiv_run("asaga")
