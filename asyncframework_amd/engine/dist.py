@@ -126,10 +126,14 @@ class _RemoteChannel:
                     self._snap_done.set()
                     continue
                 _recv(self._recvbuf, self.peer, self.group)
-                if self.stream is not None:
-                    self.stream.synchronize()
                 res = unpack_result(self._recvbuf, self.d, self.wid)
                 res.g = self._recvbuf[:self.d].clone()
+                if self.stream is not None:
+                    # the clone is enqueued on THIS channel's stream; the
+                    # server applies updates on its own stream, so the copy
+                    # must be complete before delivery (host sync here also
+                    # covers the recv itself)
+                    self.stream.synchronize()
                 self.server.on_completion(res)
 
     def join(self, timeout=None):
