@@ -91,6 +91,11 @@ class Worker:
         round_key = msg.k_submit + 1  # reference seed+k+1
         ctx = torch.cuda.stream(self.stream) if self.is_cuda else _nullctx()
         with ctx:
+            if self.is_cuda:
+                # msg.w was snapshotted on the dispatcher's stream (engine
+                # thread / C++ server): order this worker's reads after it
+                self.stream.wait_stream(
+                    torch.cuda.default_stream(self.device))
             w = msg.w
             if w.device != self.device:
                 w = w.to(self.device, non_blocking=True)
