@@ -293,7 +293,12 @@ class DistServer {
   int64_t k() const { return k_; }
   int64_t applied() const { return applied_; }
   int64_t rejected() const { return rejected_; }
-  int64_t elapsed_ms() const { return (int64_t)((now_s() - t0_) * 1000.0); }
+  int64_t elapsed_ms() const {
+    // frozen at finish (matches the Python engines, which stamp elapsed
+    // when the run loop exits, not when results are collected)
+    return done_ ? elapsed_final_ms_
+                 : (int64_t)((now_s() - t0_) * 1000.0);
+  }
   std::vector<int64_t> waiting_ms() const { return waiting_ms_; }
   std::map<int64_t, double> marks() const { return marks_; }
   std::vector<int64_t> opt_ms() const { return opt_ms_; }
