@@ -146,3 +146,38 @@ def test_cli_torchrun_dist_contract(engine, tmp_path):
     csv = [l for l in lines[sep + 1:] if re.match(r"^\d+,[0-9.eE+-]+$", l)]
     objs = [float(l.split(",")[1]) for l in csv]
     assert len(objs) >= 2 and objs[-1] < objs[0]
+
+
+@pytest.mark.timeout(300)
+def test_cli_torchrun_dist_checkpoint_resume(tmp_path):
+    """Checkpoint + resume through the torchrun CLI path (asaga, threads
+    dist engine): the checkpoint carries BOTH ranks' history tables and a
+    resumed run restores them."""
+    import subprocess
+    import sys
+    ck = str(tmp_path / "dist.ckpt")
+
+    def cmd(port, extra):
+        return [sys.executable, "-m", "torch.distributed.run",
+                "--nnodes=1", "--nproc-per-node=2",
+                "--master-addr=127.0.0.1", f"--master-port={port}",
+                "-m", "asyncframework_amd.cli", "asaga-thread",
+                "synthetic", "synthetic", "16", "200", "4", "80", "0.05",
+                "1000000", "0.3", "0.5", "20", "0", "42"] + extra
+
+    cwd = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(cmd(29741, ["--checkpoint-path", ck,
+                                     "--checkpoint-every", "25"]),
+                         capture_output=True, text=True, timeout=240,
+                         cwd=cwd)
+    assert out.returncode == 0, out.stderr[-2000:]
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    state = load_checkpoint(ck)
+    assert set(state["alpha"].keys()) == {0, 1, 2, 3}
+    assert state["k"] >= 25
+
+    out2 = subprocess.run(cmd(29742, ["--resume-from", ck]),
+                          capture_output=True, text=True, timeout=240,
+                          cwd=cwd)
+    assert out2.returncode == 0, out2.stderr[-2000:]
+    assert out2.stdout.splitlines()[-1] == "finished"
