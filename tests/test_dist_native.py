@@ -509,3 +509,52 @@ def test_native_dist_bitwise_matches_threads_oracle_asgd():
 @pytest.mark.timeout(300)
 def test_native_dist_bitwise_matches_threads_oracle_asaga():
     _seq_equiv_run("asaga")
+
+
+def _combo_rank_main(rank, init_file, ck_path, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    try:
+        # everything at once: cloud stragglers + tight-ish tau + ASAGA +
+        # periodic checkpoints + multi-worker-per-rank
+        cfg = EngineConfig(d=16, N=400, num_workers=4, num_iterations=800,
+                           gamma=0.02, taw=3, batch_rate=0.2,
+                           bucket_ratio=0.5, printer_freq=1 << 30,
+                           delay_coeff=-1.0, seed=5, device="cpu",
+                           sync=False, algo="asaga", snapshot_weights=False,
+                           calib_factor=5, checkpoint_path=ck_path,
+                           checkpoint_every=200)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+        workers = []
+        for j in range(2):
+            wid = rank * 2 + j
+            s, t = row_shards(cfg.N, 4)[wid]
+            workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                             X=X[s:t], y=y[s:t]), cfg,
+                                  device=torch.device("cpu")))
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=120)
+        if rank == 0:
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{res.rejected},"
+                        f"{int(eng.srv.delay_active())},"
+                        f"{eng.srv.max_staleness_seen()}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_combo_delay_tau_checkpoint(tmp_path):
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    init_file = str(tmp_path / "i")
+    ck = str(tmp_path / "c.ckpt")
+    out_file = str(tmp_path / "o")
+    mp.spawn(_combo_rank_main, args=(init_file, ck, out_file), nprocs=WORLD,
+             join=True)
+    with open(out_file) as f:
+        k, rejected, delay_on, max_stale = map(int, f.read().split(","))
+    assert k >= 800
+    assert delay_on == 1
+    state = load_checkpoint(ck)
+    assert set(state["alpha"].keys()) == {0, 1, 2, 3}
+    assert state["k"] >= 200
