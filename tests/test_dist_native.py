@@ -111,7 +111,7 @@ def test_delay_model_matches_python_injector():
                         inj.delay_ms(wid, rk), (P, coeff, wid, rk)
 
 
-def _tau_rank_main(rank, init_file, out_file):
+def _tau_rank_main(rank, init_file, out_file, algo):
     dist.init_process_group("gloo", init_method=f"file://{init_file}",
                             rank=rank, world_size=WORLD)
     try:
@@ -120,10 +120,11 @@ def _tau_rank_main(rank, init_file, out_file):
         # reference silently drops over-tau results and requeues,
         # SparkASGDThread.scala:202-205)
         cfg = EngineConfig(d=12, N=160, num_workers=4, num_iterations=25,
-                           gamma=0.2, taw=0, batch_rate=0.3,
-                           bucket_ratio=0.25, printer_freq=1000,
-                           delay_coeff=0.0, seed=7, device="cpu",
-                           sync=False, algo="asgd", snapshot_weights=False)
+                           gamma=0.2 if algo == "asgd" else 0.02, taw=0,
+                           batch_rate=0.3, bucket_ratio=0.25,
+                           printer_freq=1000, delay_coeff=0.0, seed=7,
+                           device="cpu", sync=False, algo=algo,
+                           snapshot_weights=False)
         X, y = synthetic_dense(cfg.N, cfg.d, seed=3)
         M = 2
         workers = []
@@ -143,12 +144,15 @@ def _tau_rank_main(rank, init_file, out_file):
 
 
 @pytest.mark.timeout(300)
-def test_native_dist_tau_zero_torture():
+@pytest.mark.parametrize("algo", ["asgd", "asaga"])
+def test_native_dist_tau_zero_torture(algo):
+    """taw=0 under both staleness encodings (ASGD arrival-clock,
+    ASAGA k-ts): heavy rejection traffic, run still completes."""
     with tempfile.TemporaryDirectory() as td:
         init_file = os.path.join(td, "init")
         out_file = os.path.join(td, "out")
-        mp.spawn(_tau_rank_main, args=(init_file, out_file), nprocs=WORLD,
-                 join=True)
+        mp.spawn(_tau_rank_main, args=(init_file, out_file, algo),
+                 nprocs=WORLD, join=True)
         with open(out_file) as f:
             k, applied, rejected = map(int, f.read().split(","))
         assert k >= 25
