@@ -562,3 +562,48 @@ def test_native_dist_combo_delay_tau_checkpoint(tmp_path):
     state = load_checkpoint(ck)
     assert set(state["alpha"].keys()) == {0, 1, 2, 3}
     assert state["k"] >= 200
+
+
+def _w3_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=3)
+    try:
+        P = 6
+        cfg = EngineConfig(d=20, N=396, num_workers=P, num_iterations=80,
+                           gamma=0.4, taw=2 ** 30, batch_rate=0.3,
+                           bucket_ratio=0.5, printer_freq=1 << 30,
+                           delay_coeff=0.0, seed=42, device="cpu",
+                           sync=False, algo="asgd", snapshot_weights=False)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+        workers = []
+        for j in range(2):
+            wid = rank * 2 + j
+            s, t = row_shards(cfg.N, P)[wid]
+            workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                             X=X[s:t], y=y[s:t]), cfg,
+                                  device=torch.device("cpu")))
+        eng = NativeDistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=120)
+        if rank == 0:
+            obj0 = float(((X @ torch.zeros(cfg.d) - y) ** 2).mean())
+            obj1 = float(((X @ res.w - y) ** 2).mean())
+            with open(out_file, "w") as f:
+                f.write(f"{res.k},{obj0},{obj1},{len(res.waiting_time)}")
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_native_dist_three_ranks():
+    """3 ranks x 2 workers: multi-peer channel fan-out (two distinct peer
+    ranks served concurrently by the C++ server)."""
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "i")
+        out_file = os.path.join(td, "o")
+        mp.spawn(_w3_rank_main, args=(init_file, out_file), nprocs=3,
+                 join=True)
+        with open(out_file) as f:
+            k, obj0, obj1, n_wids = f.read().split(",")
+        assert int(k) >= 80
+        assert float(obj1) < float(obj0)
+        assert int(n_wids) == 6
