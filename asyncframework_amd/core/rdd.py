@@ -245,6 +245,25 @@ class AsyncRDD:
             acc = f(acc, e)
         return acc
 
+    def count(self) -> int:
+        return sum(len(self._eval_partition(pid))
+                   for pid in range(self.getNumPartitions()))
+
+    # -- future-returning actions (the AsyncRDDActions analog, reference
+    #    rdd/AsyncRDDActions.scala:33-137 — Spark's OTHER, pre-existing
+    #    async mechanism; the ASYNC framework itself used mode-1 runJob) --
+    def countAsync(self):
+        return self._root._pool.submit(self.count)
+
+    def collectAsync(self):
+        return self._root._pool.submit(self.collect)
+
+    def foreachAsync(self, f: Callable[[Any], None]):
+        def _run():
+            for e in self.collect():
+                f(e)
+        return self._root._pool.submit(_run)
+
 
 class ASYNCbroadcast:
     """Versioned value store: ``value(index)`` can read an OLDER broadcast's

@@ -202,3 +202,16 @@ def test_mappartitions_with_index():
     out = rdd.mapPartitionsWithIndex(
         lambda pid, elems: [pid] if pid % 2 == 0 else []).collect()
     assert out == [0, 2]
+
+
+def test_async_actions_futures():
+    """AsyncRDDActions analog (reference AsyncRDDActions.scala:33-137):
+    future-returning count/collect/foreach."""
+    rdd, _, _ = _make_points(n=40, P=4)
+    assert rdd.countAsync().result(timeout=30) == 40
+    assert len(rdd.collectAsync().result(timeout=30)) == 40
+    seen = []
+    rdd.map(lambda p: 1).foreachAsync(seen.append).result(timeout=30)
+    assert len(seen) == 40
+    # composes with transforms
+    assert rdd.sample(False, 0.5, 9).countAsync().result(timeout=30) <= 40
