@@ -181,3 +181,18 @@ def test_cli_torchrun_dist_checkpoint_resume(tmp_path):
                           cwd=cwd)
     assert out2.returncode == 0, out2.stderr[-2000:]
     assert out2.stdout.splitlines()[-1] == "finished"
+
+
+def test_cli_main_usage_error():
+    import subprocess
+    import sys
+    cwd = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run([sys.executable, "-m", "asyncframework_amd.cli",
+                          "bogus-driver"], capture_output=True, text=True,
+                         timeout=120, cwd=cwd)
+    assert out.returncode == 2
+    assert "usage:" in out.stderr
+    out2 = subprocess.run([sys.executable, "-m", "asyncframework_amd.cli"],
+                          capture_output=True, text=True, timeout=120,
+                          cwd=cwd)
+    assert out2.returncode == 2
