@@ -398,6 +398,12 @@ class SyncEngine:
             srv.k = k + 1
             if srv.k in self.mark_at:
                 self.marks[srv.k] = time.perf_counter()
+            if (cfg.checkpoint_every > 0 and cfg.checkpoint_path
+                    and srv.k % cfg.checkpoint_every == 0):
+                from .checkpoint import save_checkpoint
+                save_checkpoint(cfg.checkpoint_path, srv,
+                                [ch.worker for ch in self.channels
+                                 if hasattr(ch, "worker")])
         elapsed = srv.elapsed_ms()
         for ch in self.channels:
             ch.dispatch(Dispatch(w=None, stop=True))

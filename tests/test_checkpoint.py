@@ -87,3 +87,29 @@ def test_resume_continues_from_k(tmp_path):
     eng.verbose = False
     res2 = eng.run(max_wall_s=60)
     assert res2.k >= 40
+
+
+def test_sync_engine_checkpoints(tmp_path):
+    """The sync engines honor checkpoint_every too (the async hook's
+    counterpart; reference has no checkpointing at all)."""
+    import torch
+
+    from asyncframework_amd.data.synthetic import synthetic_dense
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.engine.local import SyncEngine
+    from asyncframework_amd.run import build_dense_workers
+    ck = str(tmp_path / "sync.ckpt")
+    cfg = EngineConfig(d=8, N=80, num_workers=2, num_iterations=30,
+                       gamma=0.2, batch_rate=0.3, bucket_ratio=1.0,
+                       printer_freq=1000, delay_coeff=0.0, seed=1,
+                       device="cpu", sync=True, algo="asgd",
+                       snapshot_weights=False, checkpoint_path=ck,
+                       checkpoint_every=10)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=2)
+    eng = SyncEngine(cfg, build_dense_workers(cfg, X, y))
+    eng.verbose = False
+    eng.run(max_wall_s=60)
+    state = load_checkpoint(ck)
+    assert state["k"] % 10 == 0 and state["k"] >= 10
+    assert torch.is_tensor(state["w"])
