@@ -342,7 +342,9 @@ class SyncEngine:
             ch.start()
         P = cfg.num_workers
         t_start = time.perf_counter()
-        for k in range(cfg.num_iterations):
+        # resume-aware: a restored server continues from its k (fresh runs
+        # start at 0); the step-size schedule gamma/sqrt(k+1) continues too
+        for k in range(srv.k, cfg.num_iterations):
             if max_wall_s and time.perf_counter() - t_start > max_wall_s:
                 break
             self.delay.maybe_activate(k)

@@ -113,3 +113,29 @@ def test_sync_engine_checkpoints(tmp_path):
     state = load_checkpoint(ck)
     assert state["k"] % 10 == 0 and state["k"] >= 10
     assert torch.is_tensor(state["w"])
+
+
+def test_sync_engine_resume_continues_from_k(tmp_path):
+    import torch
+
+    from asyncframework_amd.data.synthetic import synthetic_dense
+    from asyncframework_amd.engine.checkpoint import load_checkpoint
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.run import build_dense_workers, run_engine
+    ck = str(tmp_path / "s.ckpt")
+    kw = dict(d=8, N=80, num_workers=2, num_iterations=30, gamma=0.2,
+              batch_rate=0.3, bucket_ratio=1.0, printer_freq=1000,
+              delay_coeff=0.0, seed=1, device="cpu", sync=True, algo="asgd",
+              snapshot_weights=False)
+    cfg = EngineConfig(**kw, checkpoint_path=ck, checkpoint_every=10)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=2)
+    run_engine(cfg, build_dense_workers(cfg, X, y), verbose=False)
+    state = load_checkpoint(ck)
+
+    # resume with the budget equal to the checkpointed k: zero new rounds,
+    # weights unchanged
+    cfg2 = EngineConfig(**{**kw, "num_iterations": state["k"]})
+    res2, _ = run_engine(cfg2, build_dense_workers(cfg2, X, y),
+                         verbose=False, resume_from=ck)
+    assert res2.k == state["k"]
+    assert torch.equal(res2.w, state["w"])
