@@ -85,6 +85,9 @@ def run_engine(cfg: EngineConfig, workers: List[Worker],
     ``resume_from`` restores a checkpoint (threads engine only)."""
     if engine == "native":
         assert not resume_from, "resume is a threads-engine feature"
+        assert not (cfg.checkpoint_every > 0 and cfg.checkpoint_path), \
+            ("checkpointing with the single-GPU native engine is not wired "
+             "— use --engine threads (the dist engines support it)")
         from .engine.native import NativeLocalEngine
         assert not cfg.sync, "native engine is async-only"
         neng = NativeLocalEngine(cfg, [w.shard for w in workers],
