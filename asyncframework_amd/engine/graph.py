@@ -48,6 +48,12 @@ class GraphEngine:
         from .. import _hip_core  # mandatory native path
         self._core = _hip_core
         assert device.type == "cuda"
+        assert not (cfg.algo == "asaga"
+                    and cfg.history_placement == "host"), \
+            "host-spill history needs the threads engine"
+        assert cfg.delay_coeff == 0.0, \
+            ("the graph engine is a single-worker device loop — straggler "
+             "injection needs the threads/native/dist engines")
         self.cfg = cfg
         self.shard = shard
         self.device = device
