@@ -238,3 +238,60 @@ def test_dist_multi_worker_per_rank_gloo(algo, tmp_path):
     assert float(obj1) < float(obj0)
     assert int(applied) >= 60
     assert int(n_wids) == 4
+
+
+def _pyfail_rank_main(rank, init_file, out_file):
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=WORLD)
+    cfg = EngineConfig(d=16, N=400, num_workers=4, num_iterations=60,
+                       gamma=0.3, taw=2 ** 30, batch_rate=0.3,
+                       bucket_ratio=1.0, printer_freq=1 << 30,
+                       delay_coeff=0.0, seed=42, device="cpu", sync=False,
+                       algo="asgd", snapshot_weights=False,
+                       worker_timeout_s=0.5)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=11)
+    M = 2
+    workers = []
+    for j in range(M):
+        wid = rank * M + j
+        s, t = row_shards(cfg.N, 4)[wid]
+        workers.append(Worker(wid, Shard(row_start=s, n_rows=t - s,
+                                         X=X[s:t], y=y[s:t]), cfg,
+                              device=torch.device("cpu")))
+    if rank == 0:
+        eng = DistEngine(cfg, workers, torch.device("cpu"))
+        res = eng.run(verbose=False, max_wall_s=60)
+        with open(out_file, "w") as f:
+            f.write(f"{res.k}")
+        os._exit(0)
+    else:
+        import threading as _th
+
+        from asyncframework_amd.engine.dist import (_recv,
+                                                    remote_worker_loop)
+        from asyncframework_amd.engine.messages import HDR
+        eng = DistEngine(cfg, workers, torch.device("cpu"))
+
+        def black_hole():
+            buf = torch.zeros(cfg.d + HDR, dtype=torch.float32)
+            _recv(buf, 0, eng.pair_groups[3])
+            import time as _t
+            _t.sleep(3600)
+
+        _th.Thread(target=black_hole, daemon=True).start()
+        remote_worker_loop(workers[0], cfg, eng.pair_groups[2],
+                           torch.device("cpu"))
+        dist.barrier()
+        os._exit(0)
+
+
+@pytest.mark.timeout(300)
+def test_dist_failure_detection_python_engine(tmp_path):
+    """The Python dist engine (the default N>1 path) also survives a lost
+    remote worker at full quorum: the reaper shrinks the gate."""
+    init_file = str(tmp_path / "i")
+    out_file = str(tmp_path / "o")
+    mp.spawn(_pyfail_rank_main, args=(init_file, out_file), nprocs=WORLD,
+             join=True)
+    with open(out_file) as f:
+        assert int(f.read()) >= 60
