@@ -137,12 +137,13 @@ class AsyncEngine:
         checkpoint — same model as the in-process snapshot, which also
         observes workers mid-round)."""
         remotes = [(wid, ch) for wid, ch in enumerate(self.channels)
-                   if hasattr(ch, "request_alpha") and ch.alpha_rows > 0]
+                   if hasattr(ch, "request_alpha") and ch.alpha_rows > 0
+                   and wid not in self.dead]  # a dead peer never answers
         for _, ch in remotes:
             ch.request_alpha()
         out = {}
         for wid, ch in remotes:
-            t = ch.wait_alpha()
+            t = ch.wait_alpha(timeout=15.0)
             if t is not None:
                 out[wid] = t
         return out
