@@ -71,7 +71,17 @@ def load_checkpoint(path: str) -> Dict:
 
 def restore(server: Server, workers: Optional[List[Worker]], state: Dict
             ) -> None:
-    """Load a snapshot back into a server + workers (devices preserved)."""
+    """Load a snapshot back into a server + workers (devices preserved).
+    The worker-pool shape must match: SAGA history tables are keyed by
+    worker id over a fixed sharding, so resuming under a different
+    ``num_workers`` would silently mis-assign history."""
+    ck_P = state.get("cfg", {}).get("num_workers")
+    if ck_P is not None and ck_P != server.cfg.num_workers:
+        raise ValueError(
+            f"checkpoint was taken with num_workers={ck_P}, cannot resume "
+            f"with num_workers={server.cfg.num_workers}")
+    if state.get("cfg", {}).get("d") not in (None, server.cfg.d):
+        raise ValueError("checkpoint dimensionality mismatch")
     server.k = int(state["k"])
     server.AC.setCurrentTime(int(state["current_time"]))
     server.w.copy_(state["w"].to(server.w.device))

@@ -139,3 +139,28 @@ def test_sync_engine_resume_continues_from_k(tmp_path):
                          verbose=False, resume_from=ck)
     assert res2.k == state["k"]
     assert torch.equal(res2.w, state["w"])
+
+
+def test_restore_rejects_mismatched_worker_pool(tmp_path):
+    import pytest as _pt
+    import torch
+
+    from asyncframework_amd.data.synthetic import synthetic_dense
+    from asyncframework_amd.engine.checkpoint import (load_checkpoint,
+                                                      restore,
+                                                      save_checkpoint)
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.engine.server import Server
+    cfg = EngineConfig(d=8, N=80, num_workers=4, algo="asaga")
+    srv = Server(cfg, device=torch.device("cpu"))
+    p = str(tmp_path / "c.ckpt")
+    save_checkpoint(p, srv)
+    state = load_checkpoint(p)
+    cfg2 = EngineConfig(d=8, N=80, num_workers=2, algo="asaga")
+    srv2 = Server(cfg2, device=torch.device("cpu"))
+    with _pt.raises(ValueError, match="num_workers"):
+        restore(srv2, [], state)
+    cfg3 = EngineConfig(d=16, N=80, num_workers=4, algo="asaga")
+    srv3 = Server(cfg3, device=torch.device("cpu"))
+    with _pt.raises(ValueError):
+        restore(srv3, [], state)
