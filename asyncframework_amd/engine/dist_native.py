@@ -188,6 +188,13 @@ class NativeDistEngine:
         if resume_from:
             from .checkpoint import load_checkpoint
             state = load_checkpoint(resume_from)
+            ck_P = state.get("cfg", {}).get("num_workers")
+            if ck_P is not None and ck_P != cfg.num_workers:
+                raise ValueError(
+                    f"checkpoint was taken with num_workers={ck_P}, cannot "
+                    f"resume with num_workers={cfg.num_workers}")
+            if state.get("cfg", {}).get("d") not in (None, cfg.d):
+                raise ValueError("checkpoint dimensionality mismatch")
             k0 = int(state["k"])
             clock0 = int(state["current_time"])
             w0 = state["w"].to(self.device)
