@@ -77,7 +77,8 @@ def load_libsvm(path: str, n_features: Optional[int] = None,
         X = np.zeros((n, d), dtype=np.float32)
         for r in range(n):
             s, t = indptr_a[r], indptr_a[r + 1]
-            X[r, cols_a[s:t]] = vals_a[s:t]
+            # accumulate duplicates (same as the native path's np.add.at)
+            np.add.at(X[r], cols_a[s:t], vals_a[s:t])
         return torch.from_numpy(X).to(dev).to(dtype), torch.from_numpy(y).to(dev)
     return (torch.from_numpy(indptr_a).to(dev),
             torch.from_numpy(cols_a).to(dev),
