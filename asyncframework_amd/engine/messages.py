@@ -57,6 +57,11 @@ def _hdr_tensor(vals) -> torch.Tensor:
 
 
 def pack_dispatch(buf: torch.Tensor, d: int, msg: Dispatch) -> None:
+    # the header is float32 on the wire: integers are exact only to 2^24
+    # (~16.7M updates — far above any BASELINE run length; fail loudly
+    # rather than corrupt clocks silently if that is ever exceeded)
+    assert msg.ts < (1 << 24) and msg.k_submit < (1 << 24), \
+        "wire header exceeds float32 integer range"
     if msg.w is not None:
         buf[:d].copy_(msg.w.to(buf.dtype))
     # ONE host->device copy for the header (per-element writes into a GPU

@@ -378,7 +378,8 @@ class DistServer {
         m = s.q.front();
         s.q.pop_front();
       }
-      // pack (engine/messages.py layout)
+      // pack (engine/messages.py layout; float32 header => integers
+      // exact to 2^24, same bound the Python packer asserts)
       if (m.w.defined()) buf.narrow(0, 0, cfg_.d).copy_(m.w);
       hdr_host[H_TS] = (float)m.ts;
       hdr_host[H_K] = (float)m.k;
