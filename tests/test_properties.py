@@ -54,8 +54,8 @@ def test_row_shards_partition_property(n, p):
 
 @settings(max_examples=60, deadline=None)
 @given(d=st.integers(1, 64),
-       ts=st.integers(0, 2 ** 24),
-       k=st.integers(0, 2 ** 24),
+       ts=st.integers(0, 2 ** 24 - 1),
+       k=st.integers(0, 2 ** 24 - 1),
        accept=st.booleans(),
        stop=st.booleans(),
        snap=st.integers(0, 2),
