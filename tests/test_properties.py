@@ -111,3 +111,25 @@ def test_delay_nonnegative_and_deterministic_property(p, coeff, seed, k):
         d1 = inj.delay_ms(wid, k)
         assert d1 >= 0.0
         assert d1 == inj.delay_ms(wid, k)
+
+
+@settings(max_examples=40, deadline=None)
+@given(n=st.integers(1, 300), p=st.integers(1, 8),
+       frac=st.floats(0.0, 1.0), seed=st.integers(0, 2 ** 31 - 1))
+def test_asyncrdd_lineage_properties(n, p, frac, seed):
+    """Verb-layer lineage laws: map composes elementwise, sample is a
+    deterministic order-preserving subset, count is map-invariant."""
+    from asyncframework_amd.core.rdd import AsyncRDD
+    p = min(p, n)
+    data = list(range(n))
+    parts = [data[i * n // p:(i + 1) * n // p] for i in range(p)]
+    rdd = AsyncRDD(parts)
+    assert rdd.collect() == data
+    assert rdd.map(lambda x: x * 3).collect() == [x * 3 for x in data]
+    s1 = rdd.sample(False, frac, seed).collect()
+    s2 = rdd.sample(False, frac, seed).collect()
+    assert s1 == s2                      # deterministic
+    assert s1 == [x for x in data if x in set(s1)]  # order-preserving subset
+    assert rdd.map(lambda x: None).count() == n
+    assert rdd.filter(lambda x: x % 2 == 0).count() == len(
+        [x for x in data if x % 2 == 0])
