@@ -133,3 +133,19 @@ def test_asyncrdd_lineage_properties(n, p, frac, seed):
     assert rdd.map(lambda x: None).count() == n
     assert rdd.filter(lambda x: x % 2 == 0).count() == len(
         [x for x in data if x % 2 == 0])
+
+
+@settings(max_examples=40, deadline=None)
+@given(seed=st.integers(0, 2 ** 63 - 1), round_k=st.integers(0, 2 ** 20),
+       start=st.integers(0, 9999), n_rows=st.integers(1, 500),
+       rate=st.floats(0.0, 1.0))
+def test_mask_window_consistency_unaligned(seed, round_k, start, n_rows,
+                                           rate):
+    """The numpy mask is window-consistent for ANY start (the verb layer
+    uses arbitrary partition offsets; only the HIP kernel requires
+    4-aligned shard starts)."""
+    part = bernoulli_mask(seed=seed, round_k=round_k, row_start=start,
+                          n_rows=n_rows, rate=rate)
+    full = bernoulli_mask(seed=seed, round_k=round_k, row_start=0,
+                          n_rows=start + n_rows, rate=rate)
+    assert np.array_equal(full[start:start + n_rows], part)
