@@ -45,15 +45,16 @@ class GraphEngine:
 
     def __init__(self, cfg: EngineConfig, shard: Shard,
                  device: torch.device, unroll: Optional[int] = None):
-        from .. import _hip_core  # mandatory native path
-        self._core = _hip_core
-        assert device.type == "cuda"
+        # config-compat asserts first (testable without a GPU)
         assert not (cfg.algo == "asaga"
                     and cfg.history_placement == "host"), \
             "host-spill history needs the threads engine"
         assert cfg.delay_coeff == 0.0, \
             ("the graph engine is a single-worker device loop — straggler "
              "injection needs the threads/native/dist engines")
+        from .. import _hip_core  # mandatory native path
+        self._core = _hip_core
+        assert device.type == "cuda"
         self.cfg = cfg
         self.shard = shard
         self.device = device

@@ -134,3 +134,20 @@ def test_synthetic_csr_deterministic():
     b = synthetic_csr(200, 64, seed=3)
     for t1, t2 in zip(a, b):
         assert torch.equal(t1, t2)
+
+
+def test_graph_engine_rejects_unsupported_configs():
+    """Unsupported combos fail LOUDLY at construction (no silent
+    feature-dropping): host-spill history and straggler injection belong
+    to the threads/native engines."""
+    from asyncframework_amd.engine.graph import GraphEngine
+    from asyncframework_amd.engine.worker import Shard
+    sh = Shard(row_start=0, n_rows=8, X=torch.zeros(8, 4),
+               y=torch.zeros(8))
+    cfg = EngineConfig(d=4, N=8, num_workers=1, algo="asaga",
+                       history_placement="host")
+    with pytest.raises(AssertionError, match="host-spill"):
+        GraphEngine(cfg, sh, torch.device("cpu"))
+    cfg2 = EngineConfig(d=4, N=8, num_workers=1, delay_coeff=1.0)
+    with pytest.raises(AssertionError, match="straggler"):
+        GraphEngine(cfg2, sh, torch.device("cpu"))
