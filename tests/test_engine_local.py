@@ -136,3 +136,28 @@ def test_opt_vars_snapshots_recorded():
     # initial + one per printer hit (k=0,10,20,30)
     assert len(res.opt_vars) == 1 + 4
     assert res.opt_vars[0][0] == 0
+
+
+def test_waiting_time_and_optvars_bookkeeping():
+    """The reference's run bookkeeping: optVars timestamps are monotone,
+    waiting times are non-negative, observed staleness is non-negative."""
+    from asyncframework_amd.data.synthetic import synthetic_dense
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.engine.local import AsyncEngine
+    from asyncframework_amd.run import build_dense_workers
+    cfg = EngineConfig(d=12, N=240, num_workers=4, num_iterations=80,
+                       gamma=0.2, taw=2 ** 30, batch_rate=0.2,
+                       bucket_ratio=0.5, printer_freq=20, delay_coeff=0.0,
+                       seed=3, device="cpu", snapshot_weights=True)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=9)
+    eng = AsyncEngine(cfg, build_dense_workers(cfg, X, y))
+    eng.verbose = False
+    res = eng.run(max_wall_s=120)
+    assert res.k >= 80
+    ts = [t for (t, _) in res.opt_vars]
+    assert ts == sorted(ts) and ts[0] == 0
+    assert len(res.opt_vars) >= 80 // 20  # one per printer_freq + initial
+    assert all(v >= 0 for v in res.waiting_time.values())
+    assert set(res.waiting_time.keys()) == {0, 1, 2, 3}
+    assert all(s >= 0 for s in res.staleness_seen)
+    assert res.applied + res.rejected == len(res.staleness_seen)
