@@ -151,3 +151,18 @@ def test_graph_engine_rejects_unsupported_configs():
     cfg2 = EngineConfig(d=4, N=8, num_workers=1, delay_coeff=1.0)
     with pytest.raises(AssertionError, match="straggler"):
         GraphEngine(cfg2, sh, torch.device("cpu"))
+
+
+def test_hip_adapter_guard_rails():
+    """The GPU adapter validates its inputs loudly (CPU tensors, misaligned
+    shard starts) — importable and checkable without a GPU since the .so
+    cross-compiles."""
+    pytest.importorskip("asyncframework_amd._hip_core")
+    from asyncframework_amd.ops import hip as hip_ops
+    X = torch.zeros(8, 4)
+    y = torch.zeros(8)
+    w = torch.zeros(4)
+    out = torch.zeros(4)
+    with pytest.raises(AssertionError, match="CUDA"):
+        hip_ops.grad_dense(X, y, w, out, seed=1, round_k=0, row_start=0,
+                           rate=0.5, obj=0)
