@@ -17,3 +17,4 @@ as a single-node multi-GPU parameter-server runtime:
 __version__ = "0.1.0"
 
 from .core.context import ASYNCcontext, RDDPartialRes, workerState  # noqa: F401
+from .core.rdd import AsyncRDD, ASYNCbroadcast  # noqa: F401
