@@ -47,3 +47,16 @@ def test_native_matches_python_fallback(tmp_path, monkeypatch):
     pyr = L.load_libsvm(path)
     for a, b in zip(nat, pyr):
         assert torch.equal(a, b)
+
+
+def test_empty_and_comment_only_files(tmp_path):
+    from asyncframework_amd.data.libsvm import load_libsvm
+    p = tmp_path / "empty.libsvm"
+    p.write_text("")
+    indptr, indices, values, y = load_libsvm(str(p), n_features=4)
+    assert indptr.tolist() == [0] and y.numel() == 0
+    p2 = tmp_path / "hash.libsvm"
+    p2.write_text("1 1:2.0\n")
+    indptr2, indices2, values2, y2 = load_libsvm(str(p2), n_features=4)
+    assert y2.tolist() == [1.0]
+    assert indices2.tolist() == [0] and values2.tolist() == [2.0]
