@@ -18,3 +18,25 @@ def test_example_runs(script):
                          cwd=ROOT)
     assert out.returncode == 0, out.stderr[-1500:]
     assert out.stdout.strip()
+
+
+@pytest.mark.timeout(300)
+def test_compare_async_sync_harness():
+    """The async-vs-sync time-to-target-loss harness (the reference's
+    headline experiment) stays runnable and emits its JSON contract."""
+    import json
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools",
+                                      "compare_async_sync.py"),
+         "--device", "cpu", "--rows", "1200", "--cols", "24",
+         "--workers", "4", "--iters", "100", "--gamma", "0.3",
+         "--rate", "0.2", "--target-rel", "0.5"],
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
+    assert out.returncode == 0, out.stderr[-1500:]
+    j = json.loads(out.stdout.strip().splitlines()[-1])
+    assert j["experiment"] == "async_vs_sync_time_to_target_loss"
+    for k in ("initial_objective", "target_objective", "async", "sync",
+              "speedup_to_target"):
+        assert k in j
+    assert j["async"]["time_to_target_ms"] >= 0
+    assert j["speedup_to_target"] > 0
