@@ -24,3 +24,4 @@ for w in 2 4 8; do
   [ "$ENGINES" != native ] && run "$w" python
   [ "$ENGINES" != python ] && run "$w" native
 done
+true
