@@ -74,3 +74,26 @@ def test_worker_state_aggregates():
     assert ac.STAT[0].getMaxStaleness() == 7
     ac.STAT[2].updateNumTasks(3)
     assert ac.STAT[2].getNumTasks() == 3
+
+
+def test_server_stat_running_average():
+    """on_completion maintains the per-worker running average task time
+    (reference workerState avg + mergeResult refresh, RDD.scala:1148-1153)."""
+    import torch
+
+    from asyncframework_amd.engine.config import EngineConfig
+    from asyncframework_amd.engine.messages import WorkerResult
+    from asyncframework_amd.engine.server import Server
+    cfg = EngineConfig(d=4, N=8, num_workers=2)
+    srv = Server(cfg, device=torch.device("cpu"))
+    g = torch.zeros(4)
+    srv.on_completion(WorkerResult(worker_id=0, g=g, ts=0, k_submit=0,
+                                   nrows=1, elapsed_ms=100.0))
+    assert srv.AC.STAT[0].getAverageTaskTime() == 100
+    srv.on_completion(WorkerResult(worker_id=0, g=g, ts=1, k_submit=1,
+                                   nrows=1, elapsed_ms=200.0))
+    assert srv.AC.STAT[0].getAverageTaskTime() == 150
+    assert srv.AC.STAT[0].getNumTasks() == 2
+    assert srv.AC.getCurrentTime() == 2
+    # second worker untouched
+    assert srv.AC.STAT[1].getNumTasks() == 0
