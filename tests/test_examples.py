@@ -40,3 +40,23 @@ def test_compare_async_sync_harness():
         assert k in j
     assert j["async"]["time_to_target_ms"] >= 0
     assert j["speedup_to_target"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_plot_loss_workflow(tmp_path):
+    """The reference's figure workflow: driver log -> error-vs-time plot."""
+    log = tmp_path / "run.log"
+    out = subprocess.run(
+        [sys.executable, "-m", "asyncframework_amd.cli", "asgd-thread",
+         "synthetic", "synthetic", "16", "200", "2", "40", "0.5",
+         "1000000", "0.3", "0.5", "10", "0", "42"],
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
+    assert out.returncode == 0
+    log.write_text(out.stdout)
+    png = tmp_path / "fig.png"
+    out2 = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "plot_loss.py"),
+         str(log), "-o", str(png)],
+        capture_output=True, text=True, timeout=240, cwd=ROOT)
+    assert out2.returncode == 0, out2.stderr[-800:]
+    assert png.exists() and png.stat().st_size > 5000
