@@ -166,3 +166,19 @@ def test_native_extension_is_loaded():
     import asyncframework_amd._hip_core as core
     assert getattr(core, "__hip__", False)
     assert "asyncframework_amd" in core.__file__
+
+
+@pytest.mark.parametrize("objective", ["lsq", "logistic"])
+def test_objective_sweep_gpu_matches_cpu_fp64(objective):
+    """K7 (the GEMM-shaped objective sweep) goes through the library GEMM
+    (hipBLASLt via torch.matmul) — its GPU result must match the CPU fp64
+    reference within GEMM tolerance."""
+    _require_hip()
+    X, y, _ = _dense(4096, 256, seed=5)
+    W = torch.randn(3, 256, device="cuda") * 0.1
+    got = ops.objective_sweep(X, y, W, objective).cpu()
+    want = torch_ref.objective_sweep(X.cpu(), y.cpu(), W.cpu(), objective)
+    assert torch.allclose(got, want, rtol=1e-4, atol=1e-6), (got, want)
+    # bf16 X: looser tolerance, same ballpark
+    got16 = ops.objective_sweep(X.to(torch.bfloat16), y, W, objective).cpu()
+    assert torch.allclose(got16, want, rtol=5e-2), (got16, want)
