@@ -33,13 +33,27 @@ def _now_us() -> float:
 
 
 class Tracer:
-    """Collects Trace Event Format events (ph 'X' complete / 'i' instant)."""
+    """Collects Trace Event Format events (ph 'X' complete / 'i' instant).
 
-    def __init__(self, path: str = ""):
+    Bounded: beyond ``max_events`` (default 2M, ~hundreds of MB of JSON)
+    new events are counted but dropped — a process that traces but never
+    flushes (e.g. a worker rank in a week-long run) must not leak without
+    bound. The saved document records how many were dropped."""
+
+    def __init__(self, path: str = "", max_events: int = 2_000_000):
         self.path = path
         self.enabled = True
+        self.max_events = max_events
+        self.dropped = 0
         self._events: List[Dict[str, Any]] = []
         self._lock = threading.Lock()
+
+    def _append(self, ev: Dict[str, Any]) -> None:
+        with self._lock:
+            if len(self._events) >= self.max_events:
+                self.dropped += 1
+                return
+            self._events.append(ev)
 
     # pid = engine role, tid = worker id / server lane
     def complete(self, name: str, tid: int, t0_us: float, dur_us: float,
@@ -49,8 +63,7 @@ class Tracer:
               "pid": pid, "tid": tid}
         if args:
             ev["args"] = args
-        with self._lock:
-            self._events.append(ev)
+        self._append(ev)
 
     def instant(self, name: str, tid: int,
                 args: Optional[Dict[str, Any]] = None,
@@ -61,8 +74,7 @@ class Tracer:
               "pid": pid, "tid": tid}
         if args:
             ev["args"] = args
-        with self._lock:
-            self._events.append(ev)
+        self._append(ev)
 
     def monotonic_s_to_us(self, t_s: float) -> float:
         """Convert a CLOCK_MONOTONIC-epoch stamp in seconds (C++
@@ -77,6 +89,8 @@ class Tracer:
         with self._lock:
             doc = {"traceEvents": list(self._events),
                    "displayTimeUnit": "ms"}
+            if self.dropped:
+                doc["droppedEvents"] = self.dropped
         with open(path, "w") as f:
             json.dump(doc, f)
         return path

@@ -71,3 +71,18 @@ def test_trace_off_costs_nothing_and_engine_still_runs():
     eng = AsyncEngine(cfg, build_dense_workers(cfg, X, y))
     eng.verbose = False
     assert eng.run(max_wall_s=60).k >= 10
+
+
+def test_tracer_event_cap():
+    tr = trace.Tracer(max_events=5)
+    for i in range(9):
+        tr.instant("e", 0)
+    assert len(tr._events) == 5 and tr.dropped == 4
+    import json
+    import tempfile
+
+    with tempfile.NamedTemporaryFile(suffix=".json") as f:
+        tr.save(f.name)
+        doc = json.load(open(f.name))
+    assert doc["droppedEvents"] == 4
+    assert len(doc["traceEvents"]) == 5
