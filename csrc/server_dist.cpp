@@ -452,7 +452,7 @@ class DistServer {
     dead_[wid] = 0;  // a late result resurrects a declared-dead worker
     const bool accept =
         cfg_.asaga ? (k_ - ts) <= cfg_.taw : staleness <= cfg_.taw;
-    if (cfg_.trace)
+    if (cfg_.trace && trace_ev_.size() < kTraceCap)
       trace_ev_.push_back({t_now, (int32_t)wid, accept ? (int8_t)1
                                                        : (int8_t)2,
                            (int32_t)k_, (int32_t)staleness});
@@ -541,7 +541,7 @@ class DistServer {
       waiting_ms_[wid] += (int64_t)((t_now - prev_fin) * 1000.0);
       submit_t_[wid] = t_now;
       avail_[wid] = 0;
-      if (cfg_.trace)
+      if (cfg_.trace && trace_ev_.size() < kTraceCap)
         trace_ev_.push_back({t_now, (int32_t)wid, (int8_t)0, (int32_t)k_,
                              0});
       DispatchMsg m;
@@ -627,6 +627,9 @@ class DistServer {
   int64_t threads_exited_ = 0;
   std::condition_variable threads_exit_cv_;
   std::string channel_error_;
+  // bound the buffer like utils/trace.py's Tracer (long traced runs must
+  // not grow without bound; ~2M events = a plottable trace)
+  static constexpr size_t kTraceCap = 2'000'000;
   struct TraceEv {
     double ts;
     int32_t wid;
