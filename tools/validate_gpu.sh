@@ -2,7 +2,9 @@
 # One-shot GPU validation (single MI355X box) — designed for:
 #   /usr/local/graft/bin/gpurun --timeout 2000 -- 'bash tools/validate_gpu.sh'
 # Writes per-step logs under gpurun_out/validate/; prints a PASS/FAIL table.
-# Budget: ~12-18 min of box time.
+# Budget: ~10-20 min typical; each step is individually timeout-bounded
+# (worst case ~60 min if everything hangs to its limit — set gpurun
+# --timeout accordingly, e.g. 2400).
 set -u
 OUT=gpurun_out/validate
 mkdir -p "$OUT"
