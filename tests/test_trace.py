@@ -86,3 +86,16 @@ def test_tracer_event_cap():
         doc = json.load(open(f.name))
     assert doc["droppedEvents"] == 4
     assert len(doc["traceEvents"]) == 5
+
+
+def test_stop_trace_inactive_returns_none():
+    assert trace.stop_trace() is None
+
+
+def test_start_trace_replaces_active(tmp_path):
+    t1 = trace.start_trace(str(tmp_path / "a.json"))
+    t2 = trace.start_trace(str(tmp_path / "b.json"))
+    assert trace.get_tracer() is t2 and t1 is not t2
+    p = trace.stop_trace()
+    assert p and p.endswith("b.json")
+    assert trace.get_tracer() is None
