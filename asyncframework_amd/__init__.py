@@ -14,7 +14,30 @@ as a single-node multi-GPU parameter-server runtime:
   baseline) with the reference's 13-positional-arg CLI and stdout contract.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
+
+
+def _ensure_native_fresh() -> None:
+    """Provenance guard (round-1 postmortem: a stale prebuilt .so shipped a
+    deadlocked engine). When running from a source checkout, verify the
+    in-tree extensions were built from the sources on disk and force-rebuild
+    on mismatch BEFORE anything imports them. Cheap (hashes ~100 KB) when
+    everything matches; no-op outside a checkout."""
+    import importlib.util
+    from pathlib import Path
+    root = Path(__file__).resolve().parent.parent
+    bh = root / "build_hip.py"
+    if not (root / "csrc").is_dir() or not bh.exists():
+        return
+    spec = importlib.util.spec_from_file_location("build_hip", bh)
+    mod = importlib.util.module_from_spec(spec)
+    import sys
+    sys.modules.setdefault("build_hip", mod)
+    spec.loader.exec_module(mod)
+    mod.ensure_fresh()
+
+
+_ensure_native_fresh()
 
 from .core.context import ASYNCcontext, RDDPartialRes, workerState  # noqa: F401
 from .core.rdd import AsyncRDD, ASYNCbroadcast  # noqa: F401

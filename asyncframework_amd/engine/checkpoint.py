@@ -66,7 +66,10 @@ def save_checkpoint(path: str, server: Server,
 
 
 def load_checkpoint(path: str) -> Dict:
-    return torch.load(path, map_location="cpu", weights_only=False)
+    # weights_only=True: the schema is tensors + primitives only, so there
+    # is no reason to allow arbitrary unpickling (a hostile checkpoint file
+    # could otherwise execute code at load).
+    return torch.load(path, map_location="cpu", weights_only=True)
 
 
 def restore(server: Server, workers: Optional[List[Worker]], state: Dict
@@ -87,6 +90,20 @@ def restore(server: Server, workers: Optional[List[Worker]], state: Dict
     server.w.copy_(state["w"].to(server.w.device))
     if state.get("alpha_bar") is not None and server.alpha_bar is not None:
         server.alpha_bar.copy_(state["alpha_bar"].to(server.alpha_bar.device))
+    missing = []
     for wk in workers or []:
-        if wk.alpha is not None and wk.id in state["alpha"]:
+        if wk.alpha is None:
+            continue
+        if wk.id in state["alpha"]:
             wk.alpha.copy_(state["alpha"][wk.id].to(wk.alpha.device))
+        else:
+            missing.append(wk.id)
+    if missing:
+        # a checkpoint taken while a remote peer's snap sideband timed out
+        # has no table for that worker; resuming silently with zeroed SAGA
+        # history would corrupt the run's semantics — be loud about it.
+        import warnings
+        warnings.warn(
+            f"checkpoint has no SAGA history table for worker(s) {missing}; "
+            "their alpha stays as-is (zeros on a fresh engine)",
+            RuntimeWarning, stacklevel=2)

@@ -146,6 +146,12 @@ class AsyncEngine:
             t = ch.wait_alpha(timeout=15.0)
             if t is not None:
                 out[wid] = t
+            else:
+                import warnings
+                warnings.warn(
+                    f"worker {wid}: SAGA history snapshot timed out; the "
+                    "checkpoint will have no alpha table for it (restore() "
+                    "warns on resume)", RuntimeWarning, stacklevel=2)
         return out
 
     def _reap_dead_workers(self):

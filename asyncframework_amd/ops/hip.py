@@ -15,6 +15,25 @@ import torch
 from .. import _hip_core  # in-tree .so — ImportError here is intentional
 
 
+def _assert_provenance() -> None:
+    """Backstop for the __init__ guard: the loaded binary's embedded source
+    hash must match the csrc/ sources next to it (catches a manually swapped
+    .so; the round-1 stale-binary hole)."""
+    import sys
+    bh = sys.modules.get("build_hip")
+    if bh is None:  # not running from a source checkout
+        return
+    expect = bh.src_hash()
+    got = getattr(_hip_core, "__src_hash__", "unstamped")
+    if got != expect:
+        raise ImportError(
+            f"_hip_core.so provenance mismatch: binary built from {got}, "
+            f"sources hash to {expect}; run `python build_hip.py --force`")
+
+
+_assert_provenance()
+
+
 def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 

@@ -307,14 +307,35 @@ def run_dist(args, device, rank, world):
     dist.destroy_process_group()
 
 
+def _start_heartbeat(period_s: float = 30.0):
+    """stderr heartbeat so a driver watching the process can tell a slow
+    bench from a wedged one (round-1: a deadlocked engine was silent for
+    30 min). The native engine additionally hard-aborts via its C++ stall
+    watchdog (csrc/engine_native.cpp) if no update lands for 60 s."""
+    import threading
+    t0 = time.perf_counter()
+
+    def beat():
+        while True:
+            time.sleep(period_s)
+            print(f"[bench heartbeat] alive t={time.perf_counter() - t0:.0f}s",
+                  file=sys.stderr, flush=True)
+
+    th = threading.Thread(target=beat, daemon=True)
+    th.start()
+
+
 def main():
     args = parse_args()
+    _start_heartbeat()
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
 
     if args.device:
         device = torch.device(args.device)
+        if device.type == "cuda":
+            torch.cuda.set_device(device)
     elif torch.cuda.is_available():
         device = torch.device(f"cuda:{local_rank}")
         torch.cuda.set_device(device)

@@ -217,4 +217,11 @@ PYBIND11_MODULE(_hip_core, m) {
   register_native_engine(m);
 
   m.attr("__hip__") = true;
+// source-provenance stamp: build_hip.py passes -DASYNCAMD_SRC_HASH=<sha256
+// of csrc sources>; the Python loader refuses a binary whose stamp doesn't
+// match the sources on disk (round-1 stale-binary postmortem).
+#ifndef ASYNCAMD_SRC_HASH
+#define ASYNCAMD_SRC_HASH "unstamped"
+#endif
+  m.attr("__src_hash__") = ASYNCAMD_SRC_HASH;
 }

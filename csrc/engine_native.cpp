@@ -136,6 +136,7 @@ struct EngineCfg {
   long calib_window = 0;
   long mark_lo = -1, mark_hi = -1;
   double max_wall_s = 3600.0;
+  double stall_s = 60.0;     // watchdog: abort if no completion for this long
   long snap_every = 0;       // optVars cadence (reference printer_freq)
   uintptr_t snap_ring = 0;   // [snap_cap][d] device ring
   long snap_cap = 0;
@@ -318,23 +319,17 @@ struct NativeEngine {
     const int staleness = clock - wk.ts;  // arrival-clock staleness
     clock += 1;
     max_staleness_seen = std::max<long>(max_staleness_seen, staleness);
-    const bool accept = (cfg.algo == 1) ? (k - wk.k_submit) <= cfg.taw
+    // ASGD: completion-clock staleness <= taw (SparkASGDThread.scala:172).
+    // ASAGA: k - ts <= taw with ts = the arrival clock at submit
+    // (SparkASAGAThread.scala:191, raw-ts packing RDD.scala:1333) — matches
+    // server.py accepts() and csrc/server_dist.cpp, NOT the applied-update
+    // counter at submit (k_submit), which would diverge once rejections occur.
+    const bool accept = (cfg.algo == 1) ? (k - wk.ts) <= cfg.taw
                                         : staleness <= cfg.taw;
     if (accept) {
       if (k < cfg.calib_window) {
         cul_time_ms += (t_now - wk.submit_t) * 1000.0;
         cul_count += 1;
-      }
-      // optVars snapshot (reference SparkASGDThread.scala:195-198: at the
-      // pre-increment k, every printer_freq applied updates)
-      if (cfg.snap_every > 0 && k % cfg.snap_every == 0 &&
-          (long)snap_ms.size() < cfg.snap_cap) {
-        HIP_CHECK(hipMemcpyAsync(
-            (void*)(cfg.snap_ring + (uintptr_t)snap_ms.size() *
-                                        (size_t)cfg.d * 4),
-            (const void*)w, (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
-            sstream));
-        snap_ms.push_back((t_now - run_t0) * 1000.0);
       }
       const double par_recs = cfg.rate * (double)cfg.N / cfg.P;
       if (cfg.algo == 1) {
@@ -351,6 +346,19 @@ struct NativeEngine {
       }
       HIP_CHECK(hipGetLastError());
       HIP_CHECK(hipEventRecord(update_ev, sstream));
+      // optVars snapshot (reference SparkASGDThread.scala:195-198: at the
+      // pre-increment k, every printer_freq applied updates) — enqueued on
+      // sstream AFTER the update kernel so the logged iterate includes this
+      // round's update, like the reference and the Python engines.
+      if (cfg.snap_every > 0 && k % cfg.snap_every == 0 &&
+          (long)snap_ms.size() < cfg.snap_cap) {
+        HIP_CHECK(hipMemcpyAsync(
+            (void*)(cfg.snap_ring + (uintptr_t)snap_ms.size() *
+                                        (size_t)cfg.d * 4),
+            (const void*)w, (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
+            sstream));
+        snap_ms.push_back((t_now - run_t0) * 1000.0);
+      }
       k += 1;
       applied += 1;
       if (k == cfg.mark_lo) {
@@ -405,21 +413,47 @@ struct NativeEngine {
         dispatch(wid, t0);
       }
     }
+    double last_progress = t0;
     while (k < cfg.iters) {
       const double t_now = now_s();
       if (t_now - t0 > cfg.max_wall_s) break;
-      // release due delayed dispatches
-      while (!delayed.empty() && delayed.front().first <= t_now) {
-        const int wid = delayed.front().second;
-        delayed.pop_front();
-        dispatch(wid, t_now);
+      // stall watchdog: a wedged loop (lost completion, bad event, driver
+      // hang) must fail loudly in seconds, not spin GPU-idle to max_wall_s
+      // (the round-1 failure mode). Delayed dispatches push the deadline.
+      double stall_deadline = last_progress + cfg.stall_s;
+      for (auto& dw : delayed)
+        stall_deadline = std::max(stall_deadline, dw.first + cfg.stall_s);
+      if (t_now > stall_deadline) {
+        int busy_n = 0;
+        for (auto& wk : ws) busy_n += wk.busy ? 1 : 0;
+        throw std::runtime_error(
+            "native engine stalled: no completion for " +
+            std::to_string(cfg.stall_s) + " s at k=" + std::to_string(k) +
+            " (busy=" + std::to_string(busy_n) +
+            ", delayed=" + std::to_string(delayed.size()) +
+            ", pending=" + std::to_string(pendingq.size()) + ")");
+      }
+      // release due delayed dispatches (due times are NOT monotone across
+      // the deque — straggler draws differ per worker — so scan all)
+      for (size_t di = 0; di < delayed.size();) {
+        if (delayed[di].first <= t_now) {
+          const int wid = delayed[di].second;
+          delayed.erase(delayed.begin() + (long)di);
+          dispatch(wid, t_now);
+        } else {
+          ++di;
+        }
       }
       // poll completions
       bool any = false;
       for (int i = 0; i < cfg.P && k < cfg.iters; ++i) {
         WorkerBuf& wk = ws[i];
+        // NB round-1 regression: a `submit_t > finish_t` guard here was
+        // permanently false after dispatch() seeded finish_t = t_now on the
+        // first round (BENCH_r01 30-min 0%-GPU hang). !busy alone gates
+        // polling; busy is only set between dispatch and completion.
         if (!wk.busy) continue;
-        if (wk.submit_t > wk.finish_t && wk.done) {
+        if (wk.done) {
           if (t_now < wk.poll_after) continue;
           const hipError_t q = hipEventQuery(wk.done);
           if (q == hipSuccess) {
@@ -434,7 +468,10 @@ struct NativeEngine {
           }
         }
       }
-      if (any) maybe_dispatch_pending(now_s());
+      if (any) {
+        last_progress = t_now;
+        maybe_dispatch_pending(now_s());
+      }
     }
     HIP_CHECK(hipStreamSynchronize(sstream));
     const double t1 = now_s();
@@ -485,6 +522,8 @@ void register_native_engine(py::module_& m) {
         cfg.mark_lo = py::cast<long>(c["mark_lo"]);
         cfg.mark_hi = py::cast<long>(c["mark_hi"]);
         cfg.max_wall_s = py::cast<double>(c["max_wall_s"]);
+        if (c.contains("stall_s"))
+          cfg.stall_s = py::cast<double>(c["stall_s"]);
         cfg.snap_every = py::cast<long>(c["snap_every"]);
         cfg.snap_ring = py::cast<uintptr_t>(c["snap_ring"]);
         cfg.snap_cap = py::cast<long>(c["snap_cap"]);
