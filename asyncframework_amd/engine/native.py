@@ -108,12 +108,15 @@ class NativeLocalEngine:
         return res
 
     def bench(self, warmup: int, steps: int,
-              max_wall_s: float = 1800.0) -> Tuple[float, Dict]:
+              max_wall_s: float = 1800.0,
+              snapshot_every: int = 0) -> Tuple[float, Dict]:
         """Exactly-K-steps contract: the native loop stamps wall times when
-        k crosses warmup and warmup+steps (after a server-stream sync)."""
+        k crosses warmup and warmup+steps (after a server-stream sync).
+        snapshot_every > 0 also records the optVars loss curve (a 3 KB D2D
+        copy on the server stream every that many applied updates)."""
         res = self.run(num_iterations=warmup + steps + 1,
                        mark_lo=warmup, mark_hi=warmup + steps,
-                       max_wall_s=max_wall_s)
+                       max_wall_s=max_wall_s, snapshot_every=snapshot_every)
         t0, t1 = res["mark_lo_t"], res["mark_hi_t"]
         if not (t1 > t0 > 0):
             raise RuntimeError(f"native bench marks missing: {res}")
