@@ -64,7 +64,7 @@ class NativeLocalEngine:
                 idx_out = torch.zeros(cap, dtype=torch.int32, device=device)
                 e_out = torch.zeros(cap, dtype=torch.float32, device=device)
                 wd.update(alpha=alpha.data_ptr(), idx_out=idx_out.data_ptr(),
-                          e_out=e_out.data_ptr())
+                          e_out=e_out.data_ptr(), saga_cap=cap)
                 keep += [alpha, idx_out, e_out]
                 self.alpha_tables.append(alpha)
             self._keep.extend(keep)

@@ -29,6 +29,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include "multi_update.h"
+
 #include <algorithm>
 #include <chrono>
 #include <cmath>
@@ -61,6 +63,11 @@ void launch_saga_update(float*, const float*, float*, float, float, float,
 void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
 void launch_sgd_update_zero(float*, float*, float, float, int, hipStream_t);
 void launch_saga_update_zero(float*, float*, float*, float, float, float,
+                             int, hipStream_t);
+void launch_multi_update(float*, float* const*, float* const*, float*, float,
+                         float, float, int, const MultiUpdateArgs*,
+                         hipStream_t);
+void launch_saga_commit_devn(float*, const int*, const float*, const int*,
                              int, hipStream_t);
 }
 
@@ -116,7 +123,7 @@ struct WorkerBuf {
   int ts = 0;        // arrival clock at dispatch
   long k_submit = 0; // round index at dispatch
   bool pending_commit = false;
-  int pending_n = 0;
+  int saga_cap = 0;  // staging capacity (commit grid bound, SAGA only)
   double submit_t = 0, finish_t = 0, waiting_ms = 0;
   double poll_after = 0;  // don't hipEventQuery before this time
   bool g_dirty = false;   // gradient buffer holds a rejected round's sums
@@ -171,6 +178,10 @@ struct NativeEngine {
   double ewma_round_s = 0;  // poll gating
   std::vector<double> snap_ms;  // host stamps for the optVars snapshots
   double run_t0 = 0;
+  // batched-update machinery
+  float** g_tab_dev = nullptr;     // device table: worker id -> g pointer
+  float** wbuf_tab_dev = nullptr;  // device table: worker id -> wbuf pointer
+  double inv_batch = 0, inv_N = 0;
 
   void init_stragglers() {
     straggler_kind.assign(cfg.P, 0);
@@ -258,28 +269,37 @@ struct NativeEngine {
     HIP_CHECK(hipGetLastError());
   }
 
-  void dispatch(int wid, double t_now) {
+  // Common dispatch tail: SAGA commit of the previous accepted round (count
+  // read on-device — no D2H sync), waiting-time bookkeeping, grad launch
+  // ordered after the latest applied update, completion event.
+  // ``copy_w``: true -> snapshot w with a hipMemcpyAsync on the worker
+  // stream (first dispatch / delayed release); false -> the batched update
+  // kernel already wrote this worker's wbuf on sstream (the common path).
+  void dispatch_impl(int wid, double t_now, bool copy_w) {
     WorkerBuf& wk = ws[wid];
     // accept-gated SAGA history commit from the worker's previous round
-    if (cfg.algo == 1 && wk.pending_commit && wk.pending_n > 0) {
-      launch_saga_commit((float*)wk.alpha, (const int*)wk.idx_out,
-                         (const float*)wk.e_out, wk.pending_n, wk.stream);
+    if (cfg.algo == 1 && wk.pending_commit) {
+      launch_saga_commit_devn((float*)wk.alpha, (const int*)wk.idx_out,
+                              (const float*)wk.e_out,
+                              (const int*)(wk.ctr + 4), wk.saga_cap,
+                              wk.stream);
       HIP_CHECK(hipGetLastError());
     }
     wk.pending_commit = false;
-    wk.pending_n = 0;
     if (wk.finish_t == 0) wk.finish_t = t_now;  // first dispatch: no wait
     wk.waiting_ms += (t_now - wk.finish_t) * 1000.0;
     wk.submit_t = t_now;
     wk.busy = true;
     wk.ts = clock;
     wk.k_submit = k;
-    // versioned weights: snapshot w on the worker stream, ordered after the
-    // latest applied update (ASYNCbroadcast semantics)
+    // versioned weights: the worker computes against the w it was
+    // dispatched with (ASYNCbroadcast semantics) — ordered after the
+    // latest applied update either way
     HIP_CHECK(hipStreamWaitEvent(wk.stream, update_ev, 0));
-    HIP_CHECK(hipMemcpyAsync((void*)wk.wbuf, (const void*)w,
-                             (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
-                             wk.stream));
+    if (copy_w)
+      HIP_CHECK(hipMemcpyAsync((void*)wk.wbuf, (const void*)w,
+                               (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
+                               wk.stream));
     launch_grad(wk, wk.k_submit + 1);  // reference seed+k+1
     HIP_CHECK(hipEventRecord(wk.done, wk.stream));
     // poll gating: querying every busy event each pass costs ~1-2 us per
@@ -287,7 +307,12 @@ struct NativeEngine {
     wk.poll_after = t_now + 0.3 * ewma_round_s;
   }
 
-  void maybe_dispatch_pending(double t_now) {
+  void dispatch(int wid, double t_now) { dispatch_impl(wid, t_now, true); }
+
+  // Pop every pending worker that may dispatch now: stragglers move to the
+  // delayed queue with a due time, the rest land in ``ready`` (dispatched
+  // after the batch flush writes their weight snapshots).
+  void collect_ready(double t_now, std::vector<int>& ready) {
     if (pendingq.empty()) return;
     if (available() < gate()) return;
     // delay calibration activation (reference :247-252)
@@ -304,12 +329,15 @@ struct NativeEngine {
         ws[wid].busy = true;  // occupied while "straggling"
         delayed.emplace_back(t_now + dly, wid);
       } else {
-        dispatch(wid, t_now);
+        ready.push_back(wid);
       }
     }
   }
 
-  void on_completion(int wid, double t_now) {
+  // Completion bookkeeping only (no launches): arrival clock, staleness,
+  // EWMA, tau filter, requeue. The caller batches accepted gradients into
+  // one multi_update kernel per poll sweep.
+  bool book_completion(int wid, double t_now) {
     WorkerBuf& wk = ws[wid];
     wk.busy = false;
     wk.finish_t = t_now;
@@ -328,61 +356,83 @@ struct NativeEngine {
                                         : staleness <= cfg.taw;
     if (accept) {
       if (k < cfg.calib_window) {
-        cul_time_ms += (t_now - wk.submit_t) * 1000.0;
+        cul_time_ms += rt * 1000.0;
         cul_count += 1;
-      }
-      const double par_recs = cfg.rate * (double)cfg.N / cfg.P;
-      if (cfg.algo == 1) {
-        launch_saga_update_zero((float*)w, (float*)wk.g, (float*)alpha_bar,
-                                (float)cfg.gamma, (float)(1.0 / par_recs),
-                                (float)(1.0 / cfg.N), cfg.d, sstream);
-        wk.pending_commit = true;
-        wk.pending_n = -1;  // resolved lazily from ctr at commit time
-      } else {
-        const double gamma_k =
-            cfg.gamma / std::sqrt((double)(k / cfg.P + 1));
-        launch_sgd_update_zero((float*)w, (float*)wk.g, (float)gamma_k,
-                               (float)(1.0 / par_recs), cfg.d, sstream);
-      }
-      HIP_CHECK(hipGetLastError());
-      HIP_CHECK(hipEventRecord(update_ev, sstream));
-      // optVars snapshot (reference SparkASGDThread.scala:195-198: at the
-      // pre-increment k, every printer_freq applied updates) — enqueued on
-      // sstream AFTER the update kernel so the logged iterate includes this
-      // round's update, like the reference and the Python engines.
-      if (cfg.snap_every > 0 && k % cfg.snap_every == 0 &&
-          (long)snap_ms.size() < cfg.snap_cap) {
-        HIP_CHECK(hipMemcpyAsync(
-            (void*)(cfg.snap_ring + (uintptr_t)snap_ms.size() *
-                                        (size_t)cfg.d * 4),
-            (const void*)w, (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
-            sstream));
-        snap_ms.push_back((t_now - run_t0) * 1000.0);
-      }
-      k += 1;
-      applied += 1;
-      if (k == cfg.mark_lo) {
-        HIP_CHECK(hipStreamSynchronize(sstream));
-        mark_lo_t = now_s();
-      }
-      if (k == cfg.mark_hi) {
-        HIP_CHECK(hipStreamSynchronize(sstream));
-        mark_hi_t = now_s();
       }
     } else {
       rejected += 1;
       wk.pending_commit = false;
-      wk.g_dirty = true;  // dispatch() re-zeroes before the next round
+      wk.g_dirty = true;  // dispatch re-zeroes before the next round
     }
     pendingq.push_back(wid);
+    return accept;
   }
 
-  void resolve_saga_commit_count(WorkerBuf& wk) {
-    // read the device sampled-count for the staged commit (host copy; only
-    // on accept, once per accepted SAGA round)
-    int n = 0;
-    HIP_CHECK(hipMemcpy(&n, (void*)(wk.ctr + 4), 4, hipMemcpyDeviceToHost));
-    wk.pending_n = n;
+  // -- batched update assembly (one kernel per poll sweep; elementwise-
+  //    sequential application inside the kernel makes the result identical
+  //    to launching the per-round update kernels back to back) ------------
+  MultiUpdateArgs mu{};
+
+  void flush_batch(const std::vector<int>* snap_targets) {
+    size_t si = 0;
+    const size_t total = snap_targets ? snap_targets->size() : 0;
+    bool launched = false;
+    while (true) {
+      mu.m = 0;
+      while (si < total && mu.m < MU_MAX)
+        mu.sw[mu.m++] = (*snap_targets)[si++];
+      if (mu.n == 0 && mu.m == 0) break;
+      mu.algo = cfg.algo;
+      launch_multi_update((float*)w, (float* const*)g_tab_dev,
+                          (float* const*)wbuf_tab_dev, (float*)alpha_bar,
+                          (float)cfg.gamma, (float)inv_batch, (float)inv_N,
+                          cfg.d, &mu, sstream);
+      HIP_CHECK(hipGetLastError());
+      launched = true;
+      mu.n = 0;
+      if (si >= total) break;
+    }
+    mu.m = 0;
+    if (launched) HIP_CHECK(hipEventRecord(update_ev, sstream));
+  }
+
+  // Append one accepted gradient; handles snapshot/mark boundaries so the
+  // exactly-K-steps bench contract and the printer_freq optVars cadence
+  // keep per-update precision despite batching.
+  void append_accepted(int wid, double t_now) {
+    WorkerBuf& wk = ws[wid];
+    const bool snap_now =
+        cfg.snap_every > 0 && k % cfg.snap_every == 0 &&
+        (long)snap_ms.size() < cfg.snap_cap;
+    if (cfg.algo == 0) {
+      const double gamma_k = cfg.gamma / std::sqrt((double)(k / cfg.P + 1));
+      mu.scale[mu.n] = (float)(gamma_k * inv_batch);
+    }
+    mu.gw[mu.n] = wid;
+    mu.n += 1;
+    if (cfg.algo == 1) wk.pending_commit = true;
+    k += 1;
+    applied += 1;
+    if (snap_now || k == cfg.mark_lo || k == cfg.mark_hi || mu.n == MU_MAX)
+      flush_batch(nullptr);
+    if (snap_now) {
+      // optVars snapshot (reference SparkASGDThread.scala:195-198): after
+      // the update that landed at the pre-increment printer_freq multiple
+      HIP_CHECK(hipMemcpyAsync(
+          (void*)(cfg.snap_ring + (uintptr_t)snap_ms.size() *
+                                      (size_t)cfg.d * 4),
+          (const void*)w, (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
+          sstream));
+      snap_ms.push_back((t_now - run_t0) * 1000.0);
+    }
+    if (k == cfg.mark_lo) {
+      HIP_CHECK(hipStreamSynchronize(sstream));
+      mark_lo_t = now_s();
+    }
+    if (k == cfg.mark_hi) {
+      HIP_CHECK(hipStreamSynchronize(sstream));
+      mark_hi_t = now_s();
+    }
   }
 
   struct Result {
@@ -400,6 +450,21 @@ struct NativeEngine {
       HIP_CHECK(hipStreamCreateWithFlags(&wk.stream, hipStreamNonBlocking));
       HIP_CHECK(hipEventCreateWithFlags(&wk.done, hipEventDisableTiming));
     }
+    inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
+    inv_N = 1.0 / (double)cfg.N;
+    {  // device pointer tables for the batched update kernel
+      std::vector<float*> hg(cfg.P), hw(cfg.P);
+      for (int i = 0; i < cfg.P; ++i) {
+        hg[i] = (float*)ws[i].g;
+        hw[i] = (float*)ws[i].wbuf;
+      }
+      HIP_CHECK(hipMalloc(&g_tab_dev, cfg.P * sizeof(float*)));
+      HIP_CHECK(hipMalloc(&wbuf_tab_dev, cfg.P * sizeof(float*)));
+      HIP_CHECK(hipMemcpy(g_tab_dev, hg.data(), cfg.P * sizeof(float*),
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpy(wbuf_tab_dev, hw.data(), cfg.P * sizeof(float*),
+                          hipMemcpyHostToDevice));
+    }
     init_stragglers();
     for (int i = 0; i < cfg.P; ++i) pendingq.push_back(i);
     const double t0 = now_s();
@@ -414,6 +479,8 @@ struct NativeEngine {
       }
     }
     double last_progress = t0;
+    std::vector<int> ready;
+    ready.reserve(cfg.P);
     while (k < cfg.iters) {
       const double t_now = now_s();
       if (t_now - t0 > cfg.max_wall_s) break;
@@ -444,7 +511,9 @@ struct NativeEngine {
           ++di;
         }
       }
-      // poll completions
+      // poll completions; accepted gradients accumulate into one batched
+      // update kernel per sweep (append_accepted flushes early at snapshot
+      // and bench-mark boundaries to keep per-update precision)
       bool any = false;
       for (int i = 0; i < cfg.P && k < cfg.iters; ++i) {
         WorkerBuf& wk = ws[i];
@@ -457,11 +526,8 @@ struct NativeEngine {
           if (t_now < wk.poll_after) continue;
           const hipError_t q = hipEventQuery(wk.done);
           if (q == hipSuccess) {
-            // SAGA: fetch the staged count before deciding (cheap, 4 B)
-            const bool will_commit = cfg.algo == 1;
-            on_completion(i, now_s());
-            if (will_commit && wk.pending_commit)
-              resolve_saga_commit_count(wk);
+            const double tc = now_s();
+            if (book_completion(i, tc)) append_accepted(i, tc);
             any = true;
           } else if (q != hipErrorNotReady) {
             HIP_CHECK(q);
@@ -470,7 +536,14 @@ struct NativeEngine {
       }
       if (any) {
         last_progress = t_now;
-        maybe_dispatch_pending(now_s());
+        ready.clear();
+        collect_ready(now_s(), ready);
+        // one kernel: apply the sweep's remaining accepted gradients AND
+        // write the post-batch w into the redispatching workers' snapshot
+        // buffers (replaces a 3 KB hipMemcpyAsync per dispatch)
+        flush_batch(&ready);
+        const double td = now_s();
+        for (int wid : ready) dispatch_impl(wid, td, false);
       }
     }
     HIP_CHECK(hipStreamSynchronize(sstream));
@@ -494,6 +567,9 @@ struct NativeEngine {
     }
     HIP_CHECK(hipEventDestroy(update_ev));
     HIP_CHECK(hipStreamDestroy(sstream));
+    HIP_CHECK(hipFree(g_tab_dev));
+    HIP_CHECK(hipFree(wbuf_tab_dev));
+    g_tab_dev = wbuf_tab_dev = nullptr;
     return out;
   }
 };
@@ -549,6 +625,7 @@ void register_native_engine(py::module_& m) {
             wk.alpha = py::cast<uintptr_t>(wd["alpha"]);
             wk.idx_out = py::cast<uintptr_t>(wd["idx_out"]);
             wk.e_out = py::cast<uintptr_t>(wd["e_out"]);
+            wk.saga_cap = py::cast<int>(wd["saga_cap"]);
           }
           eng.ws.push_back(wk);
         }

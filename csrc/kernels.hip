@@ -805,6 +805,60 @@ __global__ void saga_commit_kernel(float* __restrict__ alpha,
   if (i < n) alpha[idx[i]] = e[i];
 }
 
+// Commit with the staged count read on-device (n_dev = &ctr[1], written by
+// the SAGA gradient kernel): removes the native engine's per-accept 4-byte
+// D2H sync. Grid is sized for the staging capacity; excess blocks exit on
+// the bound check.
+__global__ void saga_commit_devn_kernel(float* __restrict__ alpha,
+                                        const int* __restrict__ idx,
+                                        const float* __restrict__ e,
+                                        const int* __restrict__ n_dev) {
+  const int n = *n_dev;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) alpha[idx[i]] = e[i];
+}
+
+// ---------------------------------------------------------- batched update
+// (native engine): ONE kernel applies a whole sweep of accepted gradients —
+// elementwise-sequentially, so the result is bit-identical to launching the
+// per-round update kernels back to back — zeroes the consumed gradient
+// accumulators, and writes the post-batch w into the weight-snapshot
+// buffers of the workers being redispatched (the versioned-broadcast copy
+// that dispatch() otherwise does with a separate hipMemcpyAsync each).
+// Pointer tables (g_tab/wbuf_tab, one slot per worker) live on the device;
+// the per-batch worker ids + ASGD step scales ride in the kernel-arg block
+// (<= 4 KB), so a batch costs exactly one launch and no H2D staging.
+#include "multi_update.h"
+
+__global__ __launch_bounds__(256) void multi_update_kernel(
+    float* __restrict__ w, float* const* __restrict__ g_tab,
+    float* const* __restrict__ wbuf_tab, float* __restrict__ alpha_bar,
+    float gamma, float inv_batch, float inv_N, int d, MultiUpdateArgs a) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < d;
+       i += gridDim.x * blockDim.x) {
+    float wi = w[i];
+    if (a.algo == 0) {
+      for (int j = 0; j < a.n; ++j) {
+        float* g = g_tab[a.gw[j]];
+        wi -= a.scale[j] * g[i];
+        g[i] = 0.f;
+      }
+    } else {
+      float ab = alpha_bar[i];
+      for (int j = 0; j < a.n; ++j) {
+        float* g = g_tab[a.gw[j]];
+        const float gi = g[i];
+        wi -= gamma * (inv_batch * gi + ab);
+        ab += inv_N * gi;
+        g[i] = 0.f;
+      }
+      alpha_bar[i] = ab;
+    }
+    w[i] = wi;
+    for (int j = 0; j < a.m; ++j) wbuf_tab[a.sw[j]][i] = wi;
+  }
+}
+
 // Fused reduce+update (overlap graph mode): one kernel sums the per-block
 // partial slabs AND applies the ASGD update. Each block reads k at entry
 // (k is stable for the whole kernel); the last-finishing block (atomic
@@ -1204,6 +1258,24 @@ void launch_saga_update_fused(float* w, float* g, float* alpha_bar,
   hipLaunchKernelGGL(saga_update_fused_kernel, dim3(1), dim3(1024), 0,
                      stream, w, g, alpha_bar, k_dev, gamma, inv_batch, inv_N,
                      d);
+}
+
+void launch_multi_update(float* w, float* const* g_tab,
+                         float* const* wbuf_tab, float* alpha_bar,
+                         float gamma, float inv_batch, float inv_N, int d,
+                         const MultiUpdateArgs* a, hipStream_t stream) {
+  int grid = (d + 255) / 256;
+  if (grid > 1024) grid = 1024;
+  hipLaunchKernelGGL(multi_update_kernel, dim3(grid), dim3(256), 0, stream,
+                     w, g_tab, wbuf_tab, alpha_bar, gamma, inv_batch, inv_N,
+                     d, *a);
+}
+
+void launch_saga_commit_devn(float* alpha, const int* idx, const float* e,
+                             const int* n_dev, int cap, hipStream_t stream) {
+  const int grid = (cap + 255) / 256;
+  hipLaunchKernelGGL(saga_commit_devn_kernel, dim3(grid > 0 ? grid : 1),
+                     dim3(256), 0, stream, alpha, idx, e, n_dev);
 }
 
 }  // extern "C"
