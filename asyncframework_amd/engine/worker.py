@@ -67,9 +67,20 @@ class Worker:
         if self.cfg.history_placement == "host" and self.is_cuda:
             self.alpha = torch.zeros(n, dtype=torch.float32,
                                      device="cpu").pin_memory()
-            # staging buffer for the per-round pinned-host -> HBM gather
+            # device staging: stale except at the round's sampled rows,
+            # which spill_refresh re-gathers from the pinned master each
+            # round (SURVEY §7.3 — only ~rate of the table moves per round)
             self._alpha_dev_buf = torch.zeros(n, dtype=torch.float32,
                                               device=self.device)
+            rate = self.cfg.batch_rate
+            cap = n if rate >= 1.0 else min(n, int(rate * n * 2) + 4096)
+            self._spill_cap = cap
+            self._spill_rows = torch.empty(cap, dtype=torch.int32,
+                                           device=self.device)
+            self._spill_y = torch.empty(cap, dtype=torch.float32,
+                                        device=self.device)
+            self._spill_cnt = torch.zeros(1, dtype=torch.int32,
+                                          device=self.device)
         else:
             self.alpha = torch.zeros(n, dtype=torch.float32, device=self.device)
 
@@ -139,11 +150,24 @@ class Worker:
         alpha = self._alpha_device()
         host_spill = alpha.device.type == "cpu" and self.is_cuda
         if host_spill:
-            # Spill path: stage the shard's history to HBM for the round via
-            # pinned-host hipMemcpyAsync into a preallocated buffer
-            # (BASELINE config 5). The kernel sees a device-resident table;
-            # accepted commits scatter back to the pinned table.
-            self._alpha_dev_buf.copy_(alpha, non_blocking=True)
+            # Spill path (BASELINE config 5): the master table stays pinned
+            # in host DRAM; refresh ONLY the round's sampled entries into
+            # the device staging table (mask-keyed gather — the sample set
+            # is deterministic from the Philox key, so scan_rows computes
+            # it on-device before the gradient kernel needs it). Fully
+            # async; no whole-table copy, no host synchronization.
+            # ASYNCAMD_SPILL_FULLCOPY=1 restores the round-1 full-copy
+            # behavior for A/B.
+            import os
+            if os.environ.get("ASYNCAMD_SPILL_FULLCOPY") == "1":
+                self._alpha_dev_buf.copy_(alpha, non_blocking=True)
+            else:
+                ops.spill_refresh(self._alpha_dev_buf, alpha, sh.y,
+                                  self._spill_rows, self._spill_y,
+                                  self._spill_cnt, self._spill_cap,
+                                  seed=cfg.seed, round_k=round_key,
+                                  row_start=sh.row_start,
+                                  rate=cfg.batch_rate)
             alpha_dev = self._alpha_dev_buf
         else:
             alpha_dev = alpha
@@ -168,7 +192,10 @@ class Worker:
         if idx is None or int(idx.numel()) == 0:
             return
         if self.alpha.device.type == "cpu" and self.is_cuda:
-            self.alpha[idx.cpu()] = e.cpu()
+            # kernel scatter into the pinned master (device-writable):
+            # async on the worker stream, ordered before the next round's
+            # spill_refresh gather on the same stream
+            ops.saga_commit_pinned(self.alpha, idx, e)
         else:
             ops.saga_commit(self.alpha, idx, e)
 

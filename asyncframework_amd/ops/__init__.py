@@ -162,6 +162,29 @@ def saga_commit(alpha: torch.Tensor, idx: torch.Tensor, e: torch.Tensor) -> None
     alpha[idx] = e.to(alpha.dtype)
 
 
+def saga_commit_pinned(alpha_pinned: torch.Tensor, idx: torch.Tensor,
+                       e: torch.Tensor) -> None:
+    """Spill-path commit: kernel scatter into the pinned-host master table
+    (GPU-only; the CPU engines keep their tables device/host-resident)."""
+    mod = _require_hip("saga_commit_pinned")
+    if mod is not None:
+        mod.saga_commit_pinned(alpha_pinned, idx, e)
+        return
+    alpha_pinned[idx.cpu().long()] = e.cpu().to(alpha_pinned.dtype)
+
+
+def spill_refresh(alpha_dev, alpha_pinned, y, rowlist, ylist, cnt, cap, *,
+                  seed, round_k, row_start, rate) -> None:
+    """Spill-path staging refresh (scan mask + gather sampled entries)."""
+    mod = _require_hip("spill_refresh")
+    if mod is not None:
+        mod.spill_refresh(alpha_dev, alpha_pinned, y, rowlist, ylist, cnt,
+                          cap, seed=seed, round_k=round_k,
+                          row_start=row_start, rate=rate)
+        return
+    alpha_dev.copy_(alpha_pinned, non_blocking=True)  # debug fallback
+
+
 def sgd_update(w: torch.Tensor, g: torch.Tensor, gamma_k: float,
                inv_batch: float) -> None:
     """Fused weight update (kernel K5)."""

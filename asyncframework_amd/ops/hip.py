@@ -140,6 +140,35 @@ def saga_commit(alpha, idx, e) -> None:
                           int(idx32.numel()), _stream())
 
 
+def saga_commit_pinned(alpha_pinned, idx, e) -> None:
+    """Commit staged history scalars into the pinned-host master table by
+    kernel scatter (ROCm pinned memory is device-writable): replaces the
+    round-1 spill path's two synchronous D2H copies + host scatter."""
+    assert alpha_pinned.is_pinned() and alpha_pinned.dtype == torch.float32
+    idx32 = idx.to(torch.int32) if idx.dtype != torch.int32 else idx
+    _hip_core.saga_commit(alpha_pinned.data_ptr(), idx32.data_ptr(),
+                          e.data_ptr(), int(idx32.numel()), _stream())
+
+
+def spill_refresh(alpha_dev, alpha_pinned, y, rowlist, ylist, cnt, cap, *,
+                  seed, round_k, row_start, rate) -> None:
+    """Host-spill α staging refresh (BASELINE config 5): scan
+    the round's Philox mask (same key the gradient kernel will use) into a
+    device row list, then gather ONLY those entries from the pinned master
+    into the device staging table — ~1% of the table instead of all of it,
+    fully async on the current stream."""
+    _chk(alpha_dev, "alpha_dev", torch.float32)
+    assert alpha_pinned.is_pinned() and alpha_pinned.dtype == torch.float32
+    cnt.zero_()
+    _hip_core.scan_rows(y.data_ptr(), rowlist.data_ptr(), ylist.data_ptr(),
+                        cnt.data_ptr(), 0, int(y.numel()), int(seed),
+                        int(round_k) & 0xFFFFFFFF, int(row_start),
+                        float(rate), _stream())
+    _hip_core.alpha_gather(alpha_dev.data_ptr(), alpha_pinned.data_ptr(),
+                           rowlist.data_ptr(), cnt.data_ptr(), int(cap),
+                           _stream())
+
+
 def sgd_update(w, g, gamma_k, inv_batch) -> None:
     _chk(w, "w", torch.float32)
     _chk(g, "g", torch.float32)

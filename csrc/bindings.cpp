@@ -50,6 +50,8 @@ void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
 void launch_saga_update(float*, const float*, float*, float, float, float,
                         int, hipStream_t);
 void launch_saga_commit(float*, const int*, const float*, int, hipStream_t);
+void launch_alpha_gather(float*, const float*, const int*, const int*, int,
+                         hipStream_t);
 void launch_sgd_update_fused(float*, float*, int*, float, float, int, int,
                              hipStream_t);
 void launch_saga_update_fused(float*, float*, float*, int*, float, float,
@@ -193,6 +195,15 @@ PYBIND11_MODULE(_hip_core, m) {
                        (hipStream_t)stream);
     check(hipGetLastError(), "saga_commit launch");
   });
+
+  m.def("alpha_gather",
+        [](uintptr_t a_dst, uintptr_t a_src, uintptr_t rows,
+           uintptr_t count_dev, int cap, uintptr_t stream) {
+          launch_alpha_gather((float*)a_dst, (const float*)a_src,
+                              (const int*)rows, (const int*)count_dev, cap,
+                              (hipStream_t)stream);
+          check(hipGetLastError(), "alpha_gather launch");
+        });
 
   m.def("sgd_update_fused",
         [](uintptr_t w, uintptr_t g, uintptr_t k_dev, float gamma,

@@ -805,6 +805,25 @@ __global__ void saga_commit_kernel(float* __restrict__ alpha,
   if (i < n) alpha[idx[i]] = e[i];
 }
 
+// Host-spill refresh (BASELINE config 5): pull ONLY the round's sampled
+// history scalars from the pinned-host master table into the device
+// staging table (a_dst[rows[j]] = a_src[rows[j]]). a_src is a pinned host
+// pointer — ROCm pinned allocations are device-visible, so the gather is
+// ~n_sampled coalesced-issue 4 B reads over the host link instead of the
+// whole-table hipMemcpy the round-1 path did. rows/count come from
+// scan_rows_kernel with the same Philox key the gradient kernel uses.
+__global__ void alpha_gather_kernel(float* __restrict__ a_dst,
+                                    const float* __restrict__ a_src,
+                                    const int* __restrict__ rows,
+                                    const int* __restrict__ count_dev) {
+  const int n = *count_dev;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    const int r = rows[i];
+    a_dst[r] = a_src[r];
+  }
+}
+
 // Commit with the staged count read on-device (n_dev = &ctr[1], written by
 // the SAGA gradient kernel): removes the native engine's per-accept 4-byte
 // D2H sync. Grid is sized for the staging capacity; excess blocks exit on
@@ -1276,6 +1295,13 @@ void launch_saga_commit_devn(float* alpha, const int* idx, const float* e,
   const int grid = (cap + 255) / 256;
   hipLaunchKernelGGL(saga_commit_devn_kernel, dim3(grid > 0 ? grid : 1),
                      dim3(256), 0, stream, alpha, idx, e, n_dev);
+}
+
+void launch_alpha_gather(float* a_dst, const float* a_src, const int* rows,
+                         const int* count_dev, int cap, hipStream_t stream) {
+  const int grid = (cap + 255) / 256;
+  hipLaunchKernelGGL(alpha_gather_kernel, dim3(grid > 0 ? grid : 1),
+                     dim3(256), 0, stream, a_dst, a_src, rows, count_dev);
 }
 
 }  // extern "C"

@@ -182,3 +182,42 @@ def test_objective_sweep_gpu_matches_cpu_fp64(objective):
     # bf16 X: looser tolerance, same ballpark
     got16 = ops.objective_sweep(X.to(torch.bfloat16), y, W, objective).cpu()
     assert torch.allclose(got16, want, rtol=5e-2), (got16, want)
+
+
+def test_spill_refresh_gathers_exactly_the_mask():
+    """Host-spill staging refresh: the device staging table must receive the
+    pinned master's values at EXACTLY the round's Philox-sampled rows (the
+    same mask the gradient kernel computes) and nothing else. Also proves
+    ROCm pinned host memory is device-readable (the design assumption)."""
+    _require_hip()
+    n, seed, rk, rs, rate = 50_000, 42, 7, 0, 0.02
+    master = (torch.arange(n, dtype=torch.float32) + 1.0).pin_memory()
+    a_dev = torch.zeros(n, dtype=torch.float32, device="cuda")
+    y = torch.zeros(n, dtype=torch.float32, device="cuda")
+    cap = int(rate * n * 2) + 4096
+    rows = torch.empty(cap, dtype=torch.int32, device="cuda")
+    ylist = torch.empty(cap, dtype=torch.float32, device="cuda")
+    cnt = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ops.spill_refresh(a_dev, master, y, rows, ylist, cnt, cap,
+                      seed=seed, round_k=rk, row_start=rs, rate=rate)
+    torch.cuda.synchronize()
+    mask = torch.from_numpy(bernoulli_mask(seed, rk, rs, n, rate))
+    got = a_dev.cpu()
+    assert int(cnt.item()) == int(mask.sum())
+    assert torch.equal(got[mask], master[mask])
+    assert torch.all(got[~mask] == 0)
+
+
+def test_saga_commit_pinned_scatters_to_host():
+    """Device-kernel scatter into the pinned-host master table (replaces the
+    round-1 two-sync-D2H commit)."""
+    _require_hip()
+    n = 10_000
+    master = torch.zeros(n, dtype=torch.float32).pin_memory()
+    idx = torch.tensor([3, 9999, 512, 7], dtype=torch.int32, device="cuda")
+    e = torch.tensor([1.5, -2.0, 3.25, 0.5], device="cuda")
+    ops.saga_commit_pinned(master, idx, e)
+    torch.cuda.synchronize()
+    want = torch.zeros(n)
+    want[idx.cpu().long()] = e.cpu()
+    assert torch.equal(master, want)
