@@ -7,7 +7,7 @@ buffer (the versioned-broadcast semantic) and a gradient accumulator; the
 C++ loop does dispatch -> event poll -> tau filter -> fused update ->
 quorum-gated redispatch, including the reference's straggler model and
 calibration. Use for GPU multi-worker configs (the threaded Python engine
-stays for CPU tests, host-spill history, and as the semantics oracle)."""
+stays for CPU tests and as the semantics oracle)."""
 
 from __future__ import annotations
 
@@ -28,8 +28,6 @@ class NativeLocalEngine:
         from .. import _hip_core
         self._core = _hip_core
         assert device.type == "cuda", "native engine is GPU-only"
-        assert cfg.history_placement != "host", \
-            "host-spill history uses the threaded engine"
         self.cfg = cfg
         self.device = device
         self.shards = shards
@@ -57,16 +55,37 @@ class NativeLocalEngine:
                 wd.update(X=sh.X.data_ptr())
                 wd["x_is_bf16"] = 1 if sh.X.dtype == torch.bfloat16 else 0
             if cfg.algo == "asaga":
-                alpha = torch.zeros(sh.n_rows, dtype=torch.float32,
-                                    device=device)
                 cap = sh.n_rows if cfg.batch_rate >= 1.0 else min(
                     sh.n_rows, int(cfg.batch_rate * sh.n_rows * 3) + 4096)
                 idx_out = torch.zeros(cap, dtype=torch.int32, device=device)
                 e_out = torch.zeros(cap, dtype=torch.float32, device=device)
+                if cfg.history_placement == "host":
+                    # spill mode (BASELINE config 5): master table pinned in
+                    # host DRAM; the device table becomes staging that the
+                    # C++ loop refreshes per round via scan_rows +
+                    # alpha_gather (mask-keyed, ~rate of the table)
+                    master = torch.zeros(sh.n_rows, dtype=torch.float32,
+                                         device="cpu").pin_memory()
+                    alpha = torch.zeros(sh.n_rows, dtype=torch.float32,
+                                        device=device)
+                    srows = torch.empty(cap, dtype=torch.int32,
+                                        device=device)
+                    sylist = torch.empty(cap, dtype=torch.float32,
+                                         device=device)
+                    scnt = torch.zeros(1, dtype=torch.int32, device=device)
+                    wd.update(alpha_host=master.data_ptr(),
+                              srows=srows.data_ptr(),
+                              sylist=sylist.data_ptr(),
+                              scnt=scnt.data_ptr())
+                    keep += [master, srows, sylist, scnt]
+                    self.alpha_tables.append(master)
+                else:
+                    alpha = torch.zeros(sh.n_rows, dtype=torch.float32,
+                                        device=device)
+                    self.alpha_tables.append(alpha)
                 wd.update(alpha=alpha.data_ptr(), idx_out=idx_out.data_ptr(),
                           e_out=e_out.data_ptr(), saga_cap=cap)
                 keep += [alpha, idx_out, e_out]
-                self.alpha_tables.append(alpha)
             self._keep.extend(keep)
             self._bufs.append(wd)
 

@@ -138,11 +138,14 @@ def objective_sweep(X: torch.Tensor, y: torch.Tensor, W: torch.Tensor,
     T = W.shape[0]
     N = X.shape[0]
     out = torch.zeros(T, dtype=torch.float64, device=X.device)
-    Wt = W.to(X.device)
+    # bf16 X: cast the SMALL operand (W, [T,d]) once and run a bf16 MFMA
+    # GEMM instead of converting every X chunk up to fp32 (the conversion
+    # kernel costs as much as the GEMM itself — profiles/r02 MFMA capture)
+    Wt = W.to(X.device, X.dtype if X.dtype == torch.bfloat16 else W.dtype)
     for s in range(0, N, batch_rows):
         Xb = X[s:s + batch_rows]
         yb = y[s:s + batch_rows].float()
-        Z = (Xb.to(Wt.dtype) @ Wt.t()).float()  # [B, T] — library GEMM
+        Z = (Xb @ Wt.t()).float()  # [B, T] — library GEMM (MFMA on gfx950)
         if objective == "lsq":
             out += ((Z - yb[:, None]) ** 2).double().sum(dim=0)
         else:

@@ -59,8 +59,8 @@ MODELS = {
                                delay_coeff=1.0, workers=8),
     "asaga-mnist8m-hostspill": dict(rows=8_100_000, cols=784, rate=0.01,
                                     algo="asaga", sync=False, dtype="bf16",
-                                    sparse=False, engine="threads",
-                                    history="host", workers=8),
+                                    sparse=False, engine="native",
+                                    history="host", workers=32),
 }
 
 

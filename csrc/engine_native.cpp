@@ -69,6 +69,10 @@ void launch_multi_update(float*, float* const*, float* const*, float*, float,
                          hipStream_t);
 void launch_saga_commit_devn(float*, const int*, const float*, const int*,
                              int, hipStream_t);
+void launch_scan_rows(const float*, int*, float*, int*, const int*, long,
+                      uint64_t, uint32_t, uint64_t, double, hipStream_t);
+void launch_alpha_gather(float*, const float*, const int*, const int*, int,
+                         hipStream_t);
 }
 
 namespace {
@@ -112,6 +116,10 @@ struct WorkerBuf {
   // device pointers supplied by Python (torch tensors kept alive there)
   uintptr_t X = 0, indptr = 0, indices = 0, values = 0, y = 0;
   uintptr_t alpha = 0, idx_out = 0, e_out = 0;  // SAGA staging
+  // host-spill mode (BASELINE config 5): alpha_host = pinned master table
+  // (device-visible); alpha above becomes the device staging table that
+  // scan_rows + alpha_gather refresh at the round's sampled rows
+  uintptr_t alpha_host = 0, srows = 0, sylist = 0, scnt = 0;
   uintptr_t wbuf = 0, g = 0, ctr = 0;           // snapshot, grad, counters
   long n_rows = 0, row_start = 0;
   bool sparse = false;
@@ -232,6 +240,21 @@ struct NativeEngine {
     }
     if (cfg.algo == 1)
       HIP_CHECK(hipMemsetAsync((void*)wk.ctr, 0, 8, wk.stream));
+    if (cfg.algo == 1 && wk.alpha_host) {
+      // spill refresh: recompute the round's Philox row set on-device and
+      // gather ONLY those entries from the pinned master into the staging
+      // table the gradient kernel reads (ordered after this worker's
+      // commit scatter on the same stream)
+      HIP_CHECK(hipMemsetAsync((void*)wk.scnt, 0, 4, wk.stream));
+      launch_scan_rows((const float*)wk.y, (int*)wk.srows,
+                       (float*)wk.sylist, (int*)wk.scnt, nullptr, wk.n_rows,
+                       cfg.seed, (uint32_t)round_key, (uint64_t)wk.row_start,
+                       cfg.rate, wk.stream);
+      launch_alpha_gather((float*)wk.alpha, (const float*)wk.alpha_host,
+                          (const int*)wk.srows, (const int*)wk.scnt,
+                          wk.saga_cap, wk.stream);
+      HIP_CHECK(hipGetLastError());
+    }
     if (cfg.algo == 1) {
       if (wk.sparse)
         launch_saga_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
@@ -278,8 +301,10 @@ struct NativeEngine {
   void dispatch_impl(int wid, double t_now, bool copy_w) {
     WorkerBuf& wk = ws[wid];
     // accept-gated SAGA history commit from the worker's previous round
+    // (spill mode: scatter straight into the pinned master table)
     if (cfg.algo == 1 && wk.pending_commit) {
-      launch_saga_commit_devn((float*)wk.alpha, (const int*)wk.idx_out,
+      float* dst = (float*)(wk.alpha_host ? wk.alpha_host : wk.alpha);
+      launch_saga_commit_devn(dst, (const int*)wk.idx_out,
                               (const float*)wk.e_out,
                               (const int*)(wk.ctr + 4), wk.saga_cap,
                               wk.stream);
@@ -626,6 +651,12 @@ void register_native_engine(py::module_& m) {
             wk.idx_out = py::cast<uintptr_t>(wd["idx_out"]);
             wk.e_out = py::cast<uintptr_t>(wd["e_out"]);
             wk.saga_cap = py::cast<int>(wd["saga_cap"]);
+            if (wd.contains("alpha_host")) {
+              wk.alpha_host = py::cast<uintptr_t>(wd["alpha_host"]);
+              wk.srows = py::cast<uintptr_t>(wd["srows"]);
+              wk.sylist = py::cast<uintptr_t>(wd["sylist"]);
+              wk.scnt = py::cast<uintptr_t>(wd["scnt"]);
+            }
           }
           eng.ws.push_back(wk);
         }
