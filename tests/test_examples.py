@@ -29,7 +29,7 @@ def test_compare_async_sync_harness():
         [sys.executable, os.path.join(ROOT, "tools",
                                       "compare_async_sync.py"),
          "--device", "cpu", "--rows", "1200", "--cols", "24",
-         "--workers", "4", "--iters", "100", "--gamma", "0.3",
+         "--workers", "4", "--iters-async", "100", "--gamma-async", "0.3",
          "--rate", "0.2", "--target-rel", "0.5"],
         capture_output=True, text=True, timeout=240, cwd=ROOT)
     assert out.returncode == 0, out.stderr[-1500:]
