@@ -81,7 +81,7 @@ def parse_args():
     p.add_argument("--algo", default="", choices=["", "asgd", "asaga"])
     p.add_argument("--device", default=None, help="override (cpu for debug)")
     p.add_argument("--engine", default="",
-                   choices=["", "graph", "threads", "native"])
+                   choices=["", "graph", "threads", "native", "resident"])
     p.add_argument("--workers", type=int, default=0,
                    help="logical workers (threads/native engines)")
     p.add_argument("--dist-engine", default="",
@@ -214,9 +214,10 @@ def run_single(args, device):
                                dtype=dt, device=device,
                                objective=args.objective)
 
-    if args.engine == "native" and device.type == "cuda":
-        # C++ event-loop engine: multi-worker async, zero Python per round
-        from asyncframework_amd.engine.native import NativeLocalEngine
+    if args.engine in ("native", "resident") and device.type == "cuda":
+        # native: C++ event-loop engine (multi-worker async, zero Python
+        # per round). resident: the whole loop in ONE persistent kernel
+        # (zero host API calls per round; dense only).
         shards = []
         for s, t in row_shards(args.rows, n_workers):
             if args.sparse:
@@ -231,7 +232,12 @@ def run_single(args, device):
                 X, y = data
                 shards.append(Shard(row_start=s, n_rows=t - s, X=X[s:t],
                                     y=y[s:t]))
-        neng = NativeLocalEngine(cfg, shards, device)
+        if args.engine == "resident":
+            from asyncframework_amd.engine.resident import ResidentEngine
+            neng = ResidentEngine(cfg, shards, device)
+        else:
+            from asyncframework_amd.engine.native import NativeLocalEngine
+            neng = NativeLocalEngine(cfg, shards, device)
         elapsed, res = neng.bench(args.warmup, args.steps,
                                   snapshot_every=snap_cadence(args))
         tt = to_target_summary(args, data, res.get("opt_vars"))

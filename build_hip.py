@@ -30,8 +30,9 @@ HIPCC = os.environ.get("HIPCC", "hipcc")
 
 SOURCES = [ROOT / "csrc" / "kernels.hip", ROOT / "csrc" / "bindings.cpp",
            ROOT / "csrc" / "libsvm_parser.cpp",
-           ROOT / "csrc" / "engine_native.cpp"]
-HEADERS = [ROOT / "csrc" / "philox.h"]
+           ROOT / "csrc" / "engine_native.cpp",
+           ROOT / "csrc" / "engine_resident.hip"]
+HEADERS = [ROOT / "csrc" / "philox.h", ROOT / "csrc" / "multi_update.h"]
 DIST_SRC = ROOT / "csrc" / "server_dist.cpp"
 DIST_DIR = ROOT / "asyncframework_amd" / "_dist_build"
 PROV = ROOT / "asyncframework_amd" / "_hip_core.provenance.json"

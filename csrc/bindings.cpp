@@ -19,6 +19,7 @@ namespace py = pybind11;
 
 void register_libsvm(py::module_& m);  // csrc/libsvm_parser.cpp
 void register_native_engine(py::module_& m);  // csrc/engine_native.cpp
+void register_resident_engine(py::module_& m);  // csrc/engine_resident.hip
 
 extern "C" {
 void launch_grad_dense(const void*, const float*, const float*, float*,
@@ -226,6 +227,7 @@ PYBIND11_MODULE(_hip_core, m) {
 
   register_libsvm(m);
   register_native_engine(m);
+  register_resident_engine(m);
 
   m.attr("__hip__") = true;
 // source-provenance stamp: build_hip.py passes -DASYNCAMD_SRC_HASH=<sha256

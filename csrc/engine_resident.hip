@@ -1,0 +1,774 @@
+// Device-resident async parameter-server engine: the WHOLE bounded-
+// staleness loop (dispatch -> versioned weight snapshot -> Philox-sampled
+// gradient -> tau filter -> update -> quorum-gated redispatch, plus the
+// straggler model and optVars snapshots) runs inside ONE persistent HIP
+// kernel. Zero host involvement between launch and completion.
+//
+// Motivation (measured, profiles/r02): the host-driven native engine is
+// bound by ~6-12 us of HIP API time per update (launch + event per worker
+// round) — the GPU itself is >95% idle at the flagship config. Moving the
+// control plane on-device removes that wall entirely; the new bound is the
+// server block's apply loop (~1-2 us/update of HBM traffic on d=784).
+//
+// Topology (gfx950: 256 CUs, 64-wide waves):
+//   block 0                      = the parameter server (single-writer on w,
+//                                  matching the reference's updater thread,
+//                                  SparkASGDThread.scala:153-226)
+//   blocks 1 + w*G .. w*G+G      = worker group w (one logical partition,
+//                                  reference Executor/TaskRunner)
+// All blocks are co-resident (grid <= max occupancy, checked at launch);
+// communication is device-scope acquire/release atomics on go/done
+// counters. Every spin loop checks a realtime deadline and aborts the
+// kernel rather than hanging the GPU.
+//
+// Semantics preserved exactly (same contracts as csrc/engine_native.cpp):
+//   * versioned broadcast: the SERVER copies w into the worker's wbuf
+//     before setting go (single-writer => the snapshot is a consistent
+//     iterate; ASYNCbroadcast.scala:21-27 semantics)
+//   * sampling: Philox(seed, k_submit+1, row_start+row) — identical keying
+//     to the host engines and utils/philox.py (reference seed+k+1)
+//   * tau filter: ASGD staleness<=taw on the arrival clock; ASAGA k-ts<=taw
+//     (SparkASGDThread.scala:172 / SparkASAGAThread.scala:191)
+//   * ASGD step gamma/sqrt(k/P+1) with Scala int division (:190)
+//   * quorum gate floor(P*bucket_ratio) (:233-237), straggler model
+//     (:124-141) with on-device calibration from accepted round times
+//   * SAGA: worker-resident history, commit-on-accept at next dispatch
+//     (ScalarMap merge inside the tau test, SparkASAGAThread.scala:206-208)
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#include "philox.h"
+
+#define RES_BLOCK 256
+#define RES_MAXP 64
+
+namespace {
+
+struct ResidentArgs {
+  float* w;                 // [d] master weights
+  // per-worker descriptor table, 8 u64 each:
+  //   [0]=X ptr, [1]=y ptr, [2]=wbuf ptr, [3]=g ptr, [4]=n_rows,
+  //   [5]=row_start, [6]=alpha ptr (SAGA), [7]=alpha_stage ptr (unused)
+  const unsigned long long* desc;
+  float* alpha_bar;         // [d] (SAGA)
+  float* snap_ring;         // [snap_cap][d] optVars ring (may be null)
+  unsigned long long* snap_cycles;  // [snap_cap] timestamps (cycles)
+  // control arrays, one slot per worker
+  unsigned int* go_round;   // server release-stores; workers acquire-spin
+  unsigned int* go_key;     // Philox round key for the round
+  unsigned int* go_flags;   // bit0: commit staged SAGA scalars (accept_prev)
+  unsigned int* done_round; // worker group release-stores on completion
+  unsigned int* done_ctr;   // intra-group arrival counter
+  unsigned long long* out;  // results, see OUT_* below
+  long long N;
+  int d, P, G;
+  long long iters;
+  float gamma;
+  unsigned int thresh;      // Philox Bernoulli threshold (rate)
+  unsigned long long seed;
+  long long taw;
+  int gate;
+  int x_is_bf16;
+  int algo;                 // 0 asgd, 1 asaga
+  int objective;            // 0 lsq, 1 logistic
+  double coeff;             // straggler model (0 off, -1 cloud, >0 worker 0)
+  long long calib_window;
+  double cycles_per_ms;
+  unsigned long long deadline_cycles;  // abort watchdog (absolute)
+  long long mark_lo, mark_hi;
+  long long snap_every, snap_cap;
+  float inv_batch, inv_N;
+};
+
+enum {
+  OUT_K = 0, OUT_APPLIED, OUT_REJECTED, OUT_MAXSTALE, OUT_MARKLO_C,
+  OUT_MARKHI_C, OUT_ABORT, OUT_T0_C, OUT_TEND_C, OUT_SNAPN, OUT_N
+};
+
+__device__ __forceinline__ unsigned long long realtime() {
+  return __builtin_amdgcn_s_memrealtime();
+}
+
+__device__ __forceinline__ unsigned int load_acq(const unsigned int* p) {
+  return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ void store_rel(unsigned int* p, unsigned int v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = ((unsigned int)u) << 16;
+  return c.f;
+}
+
+// ---------------------------------------------------------------- worker
+
+// One worker-group block's share of a round: scan its slice of the shard
+// with the round's Philox key, accumulate e_r * x_r of sampled rows into
+// its LDS slab, then flush the slab into the worker's global g with
+// device-scope atomics. SAGA: also gather alpha, emit new scalars into the
+// staging region (g buffer tail), commit previous round's on go_flags bit0.
+template <typename XT>
+__device__ void worker_round(const ResidentArgs& a, int w, int b,
+                             unsigned int key, unsigned int flags,
+                             float* lds_g) {
+  const unsigned long long* D = a.desc + (size_t)w * 8;
+  const XT* X = (const XT*)D[0];
+  const float* y = (const float*)D[1];
+  const float* wbuf = (const float*)D[2];
+  float* g = (float*)D[3];
+  const long long n_rows = (long long)D[4];
+  const long long row_start = (long long)D[5];
+  float* alpha = (float*)D[6];      // SAGA only
+  const int d = a.d;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  constexpr int WAVES = RES_BLOCK / 64;
+
+  for (int j = tid; j < d; j += RES_BLOCK) lds_g[j] = 0.f;
+  __syncthreads();
+
+  // static row partition: block b owns rows [b*slice, min(n, (b+1)*slice)),
+  // 4-aligned so the 4-row Philox blocks stay within one owner
+  const long long slice =
+      ((n_rows + a.G - 1) / a.G + 3) & ~3ll;
+  const long long r0 = (long long)b * slice;
+  const long long r1 = (r0 + slice < n_rows) ? r0 + slice : n_rows;
+
+  // each wave walks 256-row chunks of the slice; within a chunk each lane
+  // evaluates ONE Philox block (4 consecutive rows)
+  for (long long base = r0 + (long long)wave * 256; base < r1;
+       base += (long long)WAVES * 256) {
+    const long long row4 = base + (long long)lane * 4;
+    if (row4 >= r1) continue;
+    const uint4 rnd =
+        philox_block4(a.seed, key, (unsigned long long)(row_start + row4) / 4);
+    const unsigned int rv[4] = {rnd.x, rnd.y, rnd.z, rnd.w};
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const long long r = row4 + q;
+      if (r >= r1 || rv[q] >= a.thresh) continue;
+      // sampled row r: whole-lane sequential dot + axpy (rows are sparse
+      // in the sample; lanes work independent rows => high MLP)
+      const XT* xr = X + (size_t)r * d;
+      float dot = 0.f;
+      for (int j = 0; j < d; ++j) {
+        const float xv = a.x_is_bf16 ? bf16_to_f32(((const unsigned short*)xr)[j])
+                                     : ((const float*)xr)[j];
+        dot += xv * wbuf[j];
+      }
+      float e;
+      if (a.objective == 1)
+        e = 1.f / (1.f + __expf(-dot)) - y[r];
+      else
+        e = dot - y[r];
+      if (a.algo == 1) {
+        // SAGA: corrected gradient (e - alpha_r) * x, new scalar = e
+        const float corr = e - alpha[r];
+        for (int j = 0; j < d; ++j) {
+          const float xv = a.x_is_bf16
+                               ? bf16_to_f32(((const unsigned short*)xr)[j])
+                               : ((const float*)xr)[j];
+          atomicAdd(&lds_g[j], corr * xv);
+        }
+        // stage the new scalar: alpha commit is accept-gated, so write to
+        // a per-row staging value == e; commit copies it in next round
+        ((float*)D[7])[r] = e;
+      } else {
+        for (int j = 0; j < d; ++j) {
+          const float xv = a.x_is_bf16
+                               ? bf16_to_f32(((const unsigned short*)xr)[j])
+                               : ((const float*)xr)[j];
+          atomicAdd(&lds_g[j], e * xv);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int j = tid; j < d; j += RES_BLOCK)
+    if (lds_g[j] != 0.f) atomicAdd(&g[j], lds_g[j]);
+
+  // SAGA commit of THIS round's scalars happens at the worker's NEXT
+  // dispatch (accept-gated): flags bit0 says the previous round was
+  // accepted => fold staged scalars of the previous round's sampled rows.
+  // (Handled before the gradient at round start; see commit_pass.)
+  (void)flags;
+}
+
+// SAGA commit pass: before computing round `key`, fold the PREVIOUS
+// accepted round's staged scalars (identified by re-running its Philox
+// mask) into alpha. prev_key == 0 means nothing to commit.
+__device__ void commit_pass(const ResidentArgs& a, int w, int b,
+                            unsigned int prev_key) {
+  const unsigned long long* D = a.desc + (size_t)w * 8;
+  const long long n_rows = (long long)D[4];
+  const long long row_start = (long long)D[5];
+  float* alpha = (float*)D[6];
+  const float* stage = (const float*)D[7];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  constexpr int WAVES = RES_BLOCK / 64;
+  const long long slice = ((n_rows + a.G - 1) / a.G + 3) & ~3ll;
+  const long long r0 = (long long)b * slice;
+  const long long r1 = (r0 + slice < n_rows) ? r0 + slice : n_rows;
+  for (long long base = r0 + (long long)wave * 256; base < r1;
+       base += (long long)WAVES * 256) {
+    const long long row4 = base + (long long)lane * 4;
+    if (row4 >= r1) continue;
+    const uint4 rnd = philox_block4(a.seed, prev_key,
+                                    (unsigned long long)(row_start + row4) / 4);
+    const unsigned int rv[4] = {rnd.x, rnd.y, rnd.z, rnd.w};
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const long long r = row4 + q;
+      if (r < r1 && rv[q] < a.thresh) alpha[r] = stage[r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------- server
+
+// Per-worker server state lives in LDS (P <= RES_MAXP). Thread 0 runs the
+// control logic; decisions land in LDS; the whole block executes the
+// vector ops (apply update, zero g, copy snapshot) in parallel.
+struct ServerState {
+  unsigned int round_no[RES_MAXP];
+  unsigned int busy[RES_MAXP];
+  unsigned int prev_key[RES_MAXP];     // last accepted round key (SAGA)
+  unsigned int last_accept[RES_MAXP];
+  int ts[RES_MAXP];                    // arrival clock at dispatch
+  long long ksub[RES_MAXP];
+  unsigned long long submit_c[RES_MAXP], finish_c[RES_MAXP];
+  unsigned long long due_c[RES_MAXP];  // straggler release time (0 = none)
+  int straggler_kind[RES_MAXP];
+  // persistent pending queue (a gate hold must NOT drop requeued workers;
+  // ring, thread-0 mutated only)
+  int pendq[RES_MAXP];
+  int pq_head, pq_n;
+  // control scalars (thread 0 writes, block reads after syncthreads)
+  int op;          // 0 none, 1 apply+zero, 2 zero only, 3 snap, 5 dispatch
+  int op_w;
+  float op_scale;
+  int ndis;        // uniform dispatch count for the redispatch pass
+  long long k;
+  int clock_;
+  long long applied, rejected, max_stale;
+  double cul_ms;
+  long long cul_n;
+  double avg_delay_ms;
+  int delay_flag;
+  int stop;
+};
+
+__device__ __forceinline__ double dev_uniform01(unsigned long long seed,
+                                                unsigned int round_k,
+                                                unsigned int stream) {
+  // counters (0, 0, round, stream) == host uniform01 in the host engines
+  uint32_t c0 = 0, c1 = 0, c2 = round_k, c3 = stream;
+  uint32_t k0 = (uint32_t)(seed & 0xFFFFFFFFull);
+  uint32_t k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    uint64_t p0 = 0xD2511F53ull * (uint64_t)c0;
+    uint64_t p1 = 0xCD9E8D57ull * (uint64_t)c2;
+    uint32_t hi0 = (uint32_t)(p0 >> 32), lo0 = (uint32_t)p0;
+    uint32_t hi1 = (uint32_t)(p1 >> 32), lo1 = (uint32_t)p1;
+    c0 = hi1 ^ c1 ^ k0;
+    c1 = lo1;
+    c2 = hi0 ^ c3 ^ k1;
+    c3 = lo0;
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  return c0 / 4294967296.0;
+}
+
+__device__ double delay_ms_for_dev(const ResidentArgs& a,
+                                   const ServerState& st, int wid,
+                                   long long round_k) {
+  if (!st.delay_flag || a.coeff == 0.0) return 0.0;
+  if (a.coeff != -1.0) {
+    if (wid == 0 && a.coeff > 0)
+      return round(a.coeff * st.avg_delay_ms);
+    return 0.0;
+  }
+  if (st.straggler_kind[wid] == 2)
+    return round((dev_uniform01(a.seed, (unsigned)round_k, wid) * 7.5 + 2.5) *
+                 st.avg_delay_ms);
+  if (st.straggler_kind[wid] == 1)
+    return round((dev_uniform01(a.seed, (unsigned)round_k, wid) + 1.5) *
+                 st.avg_delay_ms);
+  return 0.0;
+}
+
+__device__ void server_block(const ResidentArgs& a) {
+  __shared__ ServerState st;
+  const int tid = threadIdx.x;
+  const int d = a.d;
+  if (tid == 0) {
+    st.k = 0;
+    st.clock_ = 0;
+    st.applied = st.rejected = 0;
+    st.max_stale = -1;
+    st.cul_ms = 0;
+    st.cul_n = 0;
+    st.avg_delay_ms = 0;
+    st.delay_flag = 0;
+    st.stop = 0;
+    st.pq_head = 0;
+    st.pq_n = 0;
+    for (int i = 0; i < a.P; ++i) {
+      st.round_no[i] = 0;
+      st.busy[i] = 0;
+      st.prev_key[i] = 0;
+      st.last_accept[i] = 1;
+      st.ts[i] = 0;
+      st.ksub[i] = 0;
+      st.submit_c[i] = st.finish_c[i] = 0;
+      st.due_c[i] = 0;
+      st.straggler_kind[i] = 0;
+    }
+    // reference straggler sets (SparkASGDThread.scala:124-141)
+    const int length = (int)llround(0.25 * a.P);
+    const int length_normal = (int)llround(0.8 * length);
+    const int length_longtail = length - length_normal;
+    for (int c = 0; c < length; ++c) {
+      const int idx = c * 4;
+      if (idx < a.P)
+        st.straggler_kind[idx] = (c < length_longtail) ? 2 : 1;
+    }
+    a.out[OUT_T0_C] = realtime();
+  }
+  __syncthreads();
+
+  // first dispatch: all workers, gate ignored (reference k==0 path)
+  for (int w = 0; w < a.P; ++w) {
+    const float* src = a.w;
+    float* dst = (float*)(a.desc[(size_t)w * 8 + 2]);
+    for (int j = tid; j < d; j += RES_BLOCK) dst[j] = src[j];
+    __syncthreads();
+    if (tid == 0) {
+      st.ts[w] = st.clock_;
+      st.ksub[w] = st.k;
+      st.busy[w] = 1;
+      st.submit_c[w] = realtime();
+      a.go_key[w] = (unsigned int)(st.ksub[w] + 1);
+      a.go_flags[w] = 0;
+      st.round_no[w] += 1;
+      __threadfence();
+      store_rel(&a.go_round[w], st.round_no[w]);
+    }
+    __syncthreads();
+  }
+
+  __shared__ int s_done;
+
+  while (true) {
+    if (tid == 0) {
+      s_done = 0;
+      const unsigned long long now_c = realtime();
+      if (st.k >= a.iters) s_done = 1;
+      if (now_c > a.deadline_cycles) {
+        a.out[OUT_ABORT] = 1;
+        s_done = 1;
+      }
+    }
+    __syncthreads();
+    if (s_done) break;
+
+    // ---- completion sweep (thread 0 decides; block applies) ----
+    for (int w = 0; w < a.P; ++w) {
+      __syncthreads();
+      if (tid == 0) {
+        st.op = 0;
+        if (st.busy[w] && st.due_c[w] != 0 && realtime() >= st.due_c[w]) {
+          // straggler release: back to pending for an immediate dispatch
+          st.due_c[w] = 0;
+          st.busy[w] = 0;
+          st.pendq[(st.pq_head + st.pq_n) % RES_MAXP] = w;
+          st.pq_n += 1;
+        } else if (st.busy[w] && st.due_c[w] == 0 &&
+                   load_acq(&a.done_round[w]) == st.round_no[w] &&
+                   st.k < a.iters) {
+          const unsigned long long now_c = realtime();
+          st.busy[w] = 0;
+          st.finish_c[w] = now_c;
+          const int staleness = st.clock_ - st.ts[w];
+          st.clock_ += 1;
+          if (staleness > st.max_stale) st.max_stale = staleness;
+          const bool accept = (a.algo == 1)
+                                  ? (st.k - st.ts[w]) <= a.taw
+                                  : staleness <= a.taw;
+          st.pendq[(st.pq_head + st.pq_n) % RES_MAXP] = w;
+          st.pq_n += 1;
+          if (accept) {
+            if (st.k < a.calib_window) {
+              st.cul_ms +=
+                  (double)(now_c - st.submit_c[w]) / a.cycles_per_ms;
+              st.cul_n += 1;
+            }
+            const double gamma_k =
+                a.gamma / sqrt((double)(st.k / a.P + 1));
+            st.op = 1;
+            st.op_w = w;
+            st.op_scale = (a.algo == 1) ? a.gamma
+                                        : (float)(gamma_k * a.inv_batch);
+            st.prev_key[w] = (unsigned int)(st.ksub[w] + 1);
+            st.last_accept[w] = 1;
+          } else {
+            st.rejected += 1;
+            st.last_accept[w] = 0;
+            st.op = 2;
+            st.op_w = w;
+          }
+        }
+      }
+      __syncthreads();
+      if (st.op == 1) {
+        // apply + zero (single-writer on w; elementwise-sequential order
+        // identical to the host engines)
+        float* g = (float*)(a.desc[(size_t)st.op_w * 8 + 3]);
+        if (a.algo == 1) {
+          for (int j = tid; j < d; j += RES_BLOCK) {
+            const float gj = g[j];
+            a.w[j] -= a.gamma * (a.inv_batch * gj + a.alpha_bar[j]);
+            a.alpha_bar[j] += a.inv_N * gj;
+            g[j] = 0.f;
+          }
+        } else {
+          for (int j = tid; j < d; j += RES_BLOCK) {
+            a.w[j] -= st.op_scale * g[j];
+            g[j] = 0.f;
+          }
+        }
+        __syncthreads();
+        if (tid == 0) {
+          // optVars snapshot at the pre-increment printer_freq multiple,
+          // after applying (reference :195-198)
+          if (a.snap_every > 0 && st.k % a.snap_every == 0 &&
+              (long long)a.out[OUT_SNAPN] < a.snap_cap) {
+            st.op = 3;  // block copies below
+          } else {
+            st.op = 0;
+          }
+        }
+        __syncthreads();
+        if (st.op == 3) {
+          const long long si = (long long)a.out[OUT_SNAPN];
+          float* dst = a.snap_ring + (size_t)si * d;
+          for (int j = tid; j < d; j += RES_BLOCK) dst[j] = a.w[j];
+          __syncthreads();
+          if (tid == 0) {
+            a.snap_cycles[si] = realtime();
+            a.out[OUT_SNAPN] = (unsigned long long)(si + 1);
+          }
+        }
+        if (tid == 0) {
+          st.k += 1;
+          st.applied += 1;
+          if (st.k == a.mark_lo) a.out[OUT_MARKLO_C] = realtime();
+          if (st.k == a.mark_hi) a.out[OUT_MARKHI_C] = realtime();
+        }
+      } else if (st.op == 2) {
+        float* g = (float*)(a.desc[(size_t)st.op_w * 8 + 3]);
+        for (int j = tid; j < d; j += RES_BLOCK) g[j] = 0.f;
+        __syncthreads();
+      }
+      __syncthreads();
+    }
+
+    // ---- redispatch pass (quorum gate + straggler model) ----
+    __syncthreads();
+    if (tid == 0) {
+      st.ndis = 0;
+      int avail = 0;
+      for (int i = 0; i < a.P; ++i)
+        if (!st.busy[i]) ++avail;
+      if (st.pq_n > 0 && avail >= a.gate) st.ndis = st.pq_n;
+      if (!st.delay_flag && st.k > a.calib_window) {
+        if (st.cul_n > 0) st.avg_delay_ms = st.cul_ms / (double)st.cul_n;
+        st.delay_flag = 1;
+      }
+    }
+    __syncthreads();
+    const int ndis = st.ndis;
+    for (int i = 0; i < ndis; ++i) {
+      __syncthreads();
+      int w = -1;
+      if (tid == 0) {
+        st.op = 0;
+        st.op_w = st.pendq[st.pq_head];
+        st.pq_head = (st.pq_head + 1) % RES_MAXP;
+        st.pq_n -= 1;
+        const double dly = delay_ms_for_dev(a, st, st.op_w, st.k);
+        if (dly > 0) {
+          st.busy[st.op_w] = 1;
+          st.due_c[st.op_w] =
+              realtime() + (unsigned long long)(dly * a.cycles_per_ms);
+        } else {
+          st.op = 5;  // snapshot + go
+        }
+      }
+      __syncthreads();
+      w = st.op_w;
+      if (st.op == 5) {
+        float* dst = (float*)(a.desc[(size_t)w * 8 + 2]);
+        for (int j = tid; j < d; j += RES_BLOCK) dst[j] = a.w[j];
+        __syncthreads();
+        if (tid == 0) {
+          st.ts[w] = st.clock_;
+          st.ksub[w] = st.k;
+          st.busy[w] = 1;
+          st.due_c[w] = 0;
+          st.submit_c[w] = realtime();
+          a.go_key[w] = (unsigned int)(st.ksub[w] + 1);
+          // SAGA: bit0 = commit previous accepted round's scalars;
+          // bits 1..31 carry the previous key
+          a.go_flags[w] =
+              (a.algo == 1 && st.last_accept[w] && st.prev_key[w])
+                  ? ((st.prev_key[w] << 1) | 1u)
+                  : 0u;
+          st.round_no[w] += 1;
+          __threadfence();
+          store_rel(&a.go_round[w], st.round_no[w]);
+        }
+        __syncthreads();
+      }
+    }
+    __builtin_amdgcn_s_sleep(8);
+  }
+
+  // stop: release workers
+  __syncthreads();
+  if (tid == 0) {
+    __threadfence();
+    for (int w = 0; w < a.P; ++w) store_rel(&a.go_round[w], 0xFFFFFFFFu);
+    a.out[OUT_K] = (unsigned long long)st.k;
+    a.out[OUT_APPLIED] = (unsigned long long)st.applied;
+    a.out[OUT_REJECTED] = (unsigned long long)st.rejected;
+    a.out[OUT_MAXSTALE] = (unsigned long long)(st.max_stale + 1);  // -1-safe
+    a.out[OUT_TEND_C] = realtime();
+  }
+}
+
+// ---------------------------------------------------------------- kernel
+
+template <typename XT>
+__global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
+    ResidentArgs a) {
+  if (blockIdx.x == 0) {
+    server_block(a);
+    return;
+  }
+  extern __shared__ float lds_g[];
+  const int w = (blockIdx.x - 1) / a.G;
+  const int b = (blockIdx.x - 1) % a.G;
+  unsigned int r_local = 0;
+  while (true) {
+    // acquire-spin for the next round (throttled; deadline-guarded)
+    unsigned int r;
+    while (true) {
+      r = load_acq(&a.go_round[w]);
+      if (r != r_local) break;
+      if (realtime() > a.deadline_cycles) return;
+      __builtin_amdgcn_s_sleep(16);
+    }
+    if (r == 0xFFFFFFFFu) return;  // stop
+    r_local = r;
+    const unsigned int key = a.go_key[w];
+    const unsigned int flags = a.go_flags[w];
+    if (a.algo == 1 && (flags & 1u)) commit_pass(a, w, b, flags >> 1);
+    worker_round<XT>(a, w, b, key, flags, lds_g);
+    // group arrival: last block publishes done with release semantics
+    __threadfence();
+    const unsigned int arrived =
+        __hip_atomic_fetch_add(&a.done_ctr[w], 1u, __ATOMIC_ACQ_REL,
+                               __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived + 1 == (unsigned int)a.G) {
+      __hip_atomic_store(&a.done_ctr[w], 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      store_rel(&a.done_round[w], r_local);
+    }
+  }
+}
+
+// realtime-clock calibration helper: one thread writes s_memrealtime
+__global__ void read_realtime_kernel(unsigned long long* out) {
+  if (threadIdx.x == 0) *out = __builtin_amdgcn_s_memrealtime();
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------- binding
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <chrono>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+#define RHIP_CHECK(x)                                                      \
+  do {                                                                     \
+    hipError_t _e = (x);                                                   \
+    if (_e != hipSuccess)                                                  \
+      throw std::runtime_error(std::string("resident engine: ") +          \
+                               hipGetErrorString(_e) + " @ " #x);          \
+  } while (0)
+
+double wall_s() {
+  using clk = std::chrono::steady_clock;
+  return std::chrono::duration<double>(clk::now().time_since_epoch()).count();
+}
+
+// (cycles_at_t0, cycles_per_ms) of the GPU's constant realtime counter
+std::pair<unsigned long long, double> calibrate_realtime(hipStream_t s) {
+  unsigned long long* dbuf = nullptr;
+  RHIP_CHECK(hipMalloc(&dbuf, 2 * sizeof(unsigned long long)));
+  hipLaunchKernelGGL(read_realtime_kernel, dim3(1), dim3(64), 0, s, dbuf);
+  RHIP_CHECK(hipStreamSynchronize(s));
+  const double w0 = wall_s();
+  std::this_thread::sleep_for(std::chrono::milliseconds(60));
+  hipLaunchKernelGGL(read_realtime_kernel, dim3(1), dim3(64), 0, s,
+                     dbuf + 1);
+  RHIP_CHECK(hipStreamSynchronize(s));
+  const double w1 = wall_s();
+  unsigned long long c[2];
+  RHIP_CHECK(hipMemcpy(c, dbuf, sizeof(c), hipMemcpyDeviceToHost));
+  RHIP_CHECK(hipFree(dbuf));
+  const double cpm = (double)(c[1] - c[0]) / ((w1 - w0) * 1000.0);
+  return {c[1], cpm};
+}
+
+}  // namespace
+
+void register_resident_engine(py::module_& m) {
+  m.def("resident_run", [](py::dict c) {
+    ResidentArgs a{};
+    a.w = (float*)py::cast<uintptr_t>(c["w"]);
+    a.desc = (const unsigned long long*)py::cast<uintptr_t>(c["desc"]);
+    a.alpha_bar = (float*)py::cast<uintptr_t>(c["alpha_bar"]);
+    a.snap_ring = (float*)py::cast<uintptr_t>(c["snap_ring"]);
+    a.snap_cycles =
+        (unsigned long long*)py::cast<uintptr_t>(c["snap_cycles"]);
+    a.go_round = (unsigned int*)py::cast<uintptr_t>(c["go_round"]);
+    a.go_key = (unsigned int*)py::cast<uintptr_t>(c["go_key"]);
+    a.go_flags = (unsigned int*)py::cast<uintptr_t>(c["go_flags"]);
+    a.done_round = (unsigned int*)py::cast<uintptr_t>(c["done_round"]);
+    a.done_ctr = (unsigned int*)py::cast<uintptr_t>(c["done_ctr"]);
+    a.out = (unsigned long long*)py::cast<uintptr_t>(c["out"]);
+    a.N = py::cast<long long>(c["N"]);
+    a.d = py::cast<int>(c["d"]);
+    a.P = py::cast<int>(c["P"]);
+    a.G = py::cast<int>(c["G"]);
+    a.iters = py::cast<long long>(c["iters"]);
+    a.gamma = (float)py::cast<double>(c["gamma"]);
+    const double rate = py::cast<double>(c["rate"]);
+    if (rate >= 1.0)
+      throw std::runtime_error("resident engine: rate >= 1 unsupported");
+    a.thresh = philox_threshold(rate);
+    a.seed = py::cast<uint64_t>(c["seed"]);
+    a.taw = py::cast<long long>(c["taw"]);
+    a.gate = py::cast<int>(c["gate"]);
+    a.x_is_bf16 = py::cast<int>(c["x_is_bf16"]);
+    a.algo = py::cast<int>(c["algo"]);
+    a.objective = py::cast<int>(c["objective"]);
+    a.coeff = py::cast<double>(c["coeff"]);
+    a.calib_window = py::cast<long long>(c["calib_window"]);
+    a.mark_lo = py::cast<long long>(c["mark_lo"]);
+    a.mark_hi = py::cast<long long>(c["mark_hi"]);
+    a.snap_every = py::cast<long long>(c["snap_every"]);
+    a.snap_cap = py::cast<long long>(c["snap_cap"]);
+    a.inv_batch = (float)py::cast<double>(c["inv_batch"]);
+    a.inv_N = (float)py::cast<double>(c["inv_N"]);
+    const double max_wall_s = py::cast<double>(c["max_wall_s"]);
+    if (a.P < 1 || a.P > RES_MAXP)
+      throw std::runtime_error("resident engine: P must be in [1, 64]");
+
+    py::gil_scoped_release rel;
+    hipStream_t stream;
+    RHIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    auto [c_now, cpm] = calibrate_realtime(stream);
+    a.cycles_per_ms = cpm;
+    a.deadline_cycles =
+        c_now + (unsigned long long)(max_wall_s * 1000.0 * cpm);
+
+    const int grid = 1 + a.P * a.G;
+    const size_t lds = (size_t)a.d * sizeof(float);
+    int nblk = 0;
+    if (a.x_is_bf16)
+      RHIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &nblk, (const void*)resident_engine_kernel<unsigned short>,
+          RES_BLOCK, lds));
+    else
+      RHIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &nblk, (const void*)resident_engine_kernel<float>, RES_BLOCK,
+          lds));
+    hipDeviceProp_t prop;
+    RHIP_CHECK(hipGetDeviceProperties(&prop, 0));
+    const int max_resident = nblk * prop.multiProcessorCount;
+    if (grid > max_resident)
+      throw std::runtime_error(
+          "resident engine: grid " + std::to_string(grid) +
+          " exceeds co-resident capacity " + std::to_string(max_resident) +
+          " (persistent kernel would deadlock); lower P or G");
+    if (a.x_is_bf16)
+      hipLaunchKernelGGL(resident_engine_kernel<unsigned short>, dim3(grid),
+                         dim3(RES_BLOCK), lds, stream, a);
+    else
+      hipLaunchKernelGGL(resident_engine_kernel<float>, dim3(grid),
+                         dim3(RES_BLOCK), lds, stream, a);
+    RHIP_CHECK(hipGetLastError());
+    // host-side watchdog: poll with a margin past the device deadline
+    const double host_deadline = wall_s() + max_wall_s + 10.0;
+    while (true) {
+      const hipError_t q = hipStreamQuery(stream);
+      if (q == hipSuccess) break;
+      if (q != hipErrorNotReady) RHIP_CHECK(q);
+      if (wall_s() > host_deadline) {
+        hipStreamDestroy(stream);
+        throw std::runtime_error(
+            "resident engine: kernel exceeded host deadline (device "
+            "watchdog failed to fire)");
+      }
+      std::this_thread::sleep_for(std::chrono::microseconds(200));
+    }
+    RHIP_CHECK(hipStreamDestroy(stream));
+
+    unsigned long long out_h[OUT_N];
+    RHIP_CHECK(hipMemcpy(out_h, (const void*)a.out, sizeof(out_h),
+                         hipMemcpyDeviceToHost));
+    py::gil_scoped_acquire acq;
+    py::dict r;
+    r["k"] = (long long)out_h[OUT_K];
+    r["applied"] = (long long)out_h[OUT_APPLIED];
+    r["rejected"] = (long long)out_h[OUT_REJECTED];
+    r["max_staleness"] = (long long)out_h[OUT_MAXSTALE] - 1;
+    r["aborted"] = (bool)out_h[OUT_ABORT];
+    const double t0c = (double)out_h[OUT_T0_C];
+    r["elapsed_ms"] = ((double)out_h[OUT_TEND_C] - t0c) / cpm;
+    r["mark_lo_t"] = out_h[OUT_MARKLO_C]
+                         ? ((double)out_h[OUT_MARKLO_C] - t0c) / cpm / 1e3
+                         : 0.0;
+    r["mark_hi_t"] = out_h[OUT_MARKHI_C]
+                         ? ((double)out_h[OUT_MARKHI_C] - t0c) / cpm / 1e3
+                         : 0.0;
+    r["snap_n"] = (long long)out_h[OUT_SNAPN];
+    r["cycles_per_ms"] = cpm;
+    r["t0_cycles"] = (unsigned long long)out_h[OUT_T0_C];
+    return r;
+  });
+}

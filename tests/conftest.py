@@ -12,7 +12,7 @@ def pytest_configure(config):
 # remaining GPU tests from the driver).
 _GPU_FILE_ORDER = [
     "test_gpu_kernels", "test_philox", "test_gpu_graph", "test_gpu_native",
-    "test_gpu_engine", "test_dist_native",
+    "test_gpu_resident", "test_gpu_engine", "test_dist_native",
 ]
 
 

@@ -1,0 +1,141 @@
+"""Device-resident persistent-kernel engine (csrc/engine_resident.hip):
+P=1 equivalence against the sequential reference, multiworker async runs,
+SAGA history, tau filter, straggler model, optVars snapshots."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from asyncframework_amd.data.shard import row_shards
+from asyncframework_amd.data.synthetic import synthetic_dense
+from asyncframework_amd.engine.config import EngineConfig
+from asyncframework_amd.engine.resident import ResidentEngine
+from asyncframework_amd.engine.worker import Shard
+from asyncframework_amd.ops import torch_ref
+from asyncframework_amd.utils.philox import bernoulli_mask
+
+
+def _cfg(**kw):
+    base = dict(d=64, N=20_000, num_workers=1, num_iterations=60, gamma=0.3,
+                taw=1 << 30, batch_rate=0.05, bucket_ratio=0.5,
+                printer_freq=1 << 30, delay_coeff=0.0, seed=42,
+                device="cuda:0", snapshot_weights=False)
+    base.update(kw)
+    return EngineConfig(**base)
+
+
+def _shards(cfg, X, y):
+    return [Shard(row_start=s, n_rows=t - s, X=X[s:t], y=y[s:t])
+            for s, t in row_shards(cfg.N, cfg.num_workers)]
+
+
+def _eng(cfg, X, y, G=8):
+    return ResidentEngine(cfg, _shards(cfg, X, y), torch.device("cuda:0"),
+                          blocks_per_worker=G)
+
+
+def test_resident_p1_matches_sequential_ref():
+    """P=1 is sequential: the persistent kernel must reproduce the exact
+    host-side reference iteration (same Philox keys seed+k+1)."""
+    cfg = _cfg()
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=1, device="cuda:0")
+    eng = _eng(cfg, X, y)
+    res = eng.run(max_wall_s=120)
+    assert res["k"] == cfg.num_iterations
+    w = torch.zeros(cfg.d, device="cuda:0")
+    for k in range(cfg.num_iterations):
+        mask = torch.from_numpy(
+            bernoulli_mask(cfg.seed, k + 1, 0, cfg.N, cfg.batch_rate)).cuda()
+        g, _ = torch_ref.grad_dense(X.float(), y, w, mask, cfg.objective)
+        gamma_k = cfg.gamma / math.sqrt(k // cfg.num_workers + 1)
+        w -= gamma_k * g / cfg.par_recs
+    rel = float((eng.w - w).norm() / (w.norm() + 1e-12))
+    assert rel < 1e-4, rel
+
+
+def test_resident_multiworker_async():
+    cfg = _cfg(num_workers=8, num_iterations=2000, bucket_ratio=0.5,
+               N=80_000)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=2, device="cuda:0")
+    eng = _eng(cfg, X, y, G=4)
+    res = eng.run(max_wall_s=120)
+    assert res["k"] >= cfg.num_iterations
+    assert res["applied"] == res["k"]
+    obj0 = float(((X.float() @ torch.zeros(cfg.d, device="cuda:0") - y) ** 2
+                  ).mean())
+    obj1 = float(((X.float() @ eng.w - y) ** 2).mean())
+    assert obj1 < obj0
+
+
+def test_resident_bf16_flagship_shape():
+    cfg = _cfg(d=784, N=200_000, num_workers=8, num_iterations=1000,
+               gamma=0.5, batch_rate=0.01, dtype="bf16")
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=3, dtype=torch.bfloat16,
+                           device="cuda:0")
+    eng = _eng(cfg, X, y, G=4)
+    res = eng.run(max_wall_s=120)
+    assert res["k"] >= cfg.num_iterations
+    obj0 = float(((X.float() @ torch.zeros(cfg.d, device="cuda:0") - y) ** 2
+                  ).mean())
+    obj1 = float(((X.float() @ eng.w - y) ** 2).mean())
+    assert obj1 < obj0
+
+
+def test_resident_asaga_matches_native_p1():
+    """SAGA on the resident engine vs the host native engine at P=1:
+    identical accept/commit sequences => near-identical results (tolerance
+    covers the LDS-atomic accumulation order)."""
+    from asyncframework_amd.engine.native import NativeLocalEngine
+    cfg = _cfg(algo="asaga", gamma=0.05, num_iterations=80)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=4, device="cuda:0")
+    r_eng = _eng(cfg, X, y)
+    r_res = r_eng.run(max_wall_s=120)
+    n_eng = NativeLocalEngine(cfg, _shards(cfg, X, y),
+                              torch.device("cuda:0"))
+    n_res = n_eng.run(max_wall_s=120)
+    assert r_res["k"] == n_res["k"] == cfg.num_iterations
+    rel = float((r_eng.w - n_eng.w).norm() / (n_eng.w.norm() + 1e-12))
+    assert rel < 1e-4, rel
+    ra = r_eng.alpha_tables[0]
+    na = n_eng.alpha_tables[0]
+    assert int((ra != 0).sum()) > 0
+    assert torch.allclose(ra, na, atol=1e-5)
+
+
+def test_resident_tau_filter_rejects():
+    """tau=0 with many workers: stale arrivals must be rejected (the
+    torture-tier semantic: observed staleness > 0 => rejections occur)."""
+    cfg = _cfg(num_workers=8, num_iterations=800, taw=0, N=80_000,
+               bucket_ratio=0.25)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=5, device="cuda:0")
+    eng = _eng(cfg, X, y, G=4)
+    res = eng.run(max_wall_s=120)
+    assert res["k"] >= cfg.num_iterations
+    assert res["rejected"] > 0
+    assert res["max_staleness"] > 0
+
+
+def test_resident_straggler_model_runs():
+    cfg = _cfg(num_workers=8, num_iterations=600, N=80_000,
+               delay_coeff=-1.0, calib_factor=5)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=6, device="cuda:0")
+    eng = _eng(cfg, X, y, G=4)
+    res = eng.run(max_wall_s=120)
+    assert res["k"] >= cfg.num_iterations
+
+
+def test_resident_snapshots_and_marks():
+    cfg = _cfg(num_workers=4, num_iterations=400, N=40_000)
+    X, y = synthetic_dense(cfg.N, cfg.d, seed=7, device="cuda:0")
+    eng = _eng(cfg, X, y, G=4)
+    elapsed, res = eng.bench(100, 200, snapshot_every=50)
+    assert elapsed > 0
+    ov = res["opt_vars"]
+    assert len(ov) >= 3
+    ts = [t for t, _ in ov]
+    assert ts == sorted(ts)
+    # the last snapshot differs from w0 (updates actually recorded)
+    assert float(ov[-1][1].norm()) > 0
