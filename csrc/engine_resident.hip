@@ -353,6 +353,7 @@ __device__ void server_block(const ResidentArgs& a) {
     const float* src = a.w;
     float* dst = (float*)(a.desc[(size_t)w * 8 + 2]);
     for (int j = tid; j < d; j += RES_BLOCK) dst[j] = src[j];
+    __threadfence();  // every thread publishes ITS OWN snapshot writes
     __syncthreads();
     if (tid == 0) {
       st.ts[w] = st.clock_;
@@ -521,6 +522,7 @@ __device__ void server_block(const ResidentArgs& a) {
       if (st.op == 5) {
         float* dst = (float*)(a.desc[(size_t)w * 8 + 2]);
         for (int j = tid; j < d; j += RES_BLOCK) dst[j] = a.w[j];
+        __threadfence();  // publish snapshot + this sweep's g zeroing
         __syncthreads();
         if (tid == 0) {
           st.ts[w] = st.clock_;
@@ -568,34 +570,57 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     return;
   }
   extern __shared__ float lds_g[];
+  __shared__ unsigned int s_go[3];  // round, key, flags (LDS broadcast)
   const int w = (blockIdx.x - 1) / a.G;
   const int b = (blockIdx.x - 1) % a.G;
+  const int tid = threadIdx.x;
   unsigned int r_local = 0;
   while (true) {
-    // acquire-spin for the next round (throttled; deadline-guarded)
-    unsigned int r;
-    while (true) {
-      r = load_acq(&a.go_round[w]);
-      if (r != r_local) break;
-      if (realtime() > a.deadline_cycles) return;
-      __builtin_amdgcn_s_sleep(16);
+    // ONE thread acquire-spins (65k threads hammering agent-scope acquires
+    // would storm the L2s with invalidates); the round record is broadcast
+    // through LDS. Throttled and deadline-guarded.
+    if (tid == 0) {
+      unsigned int r;
+      while (true) {
+        r = load_acq(&a.go_round[w]);
+        if (r != r_local) break;
+        if (realtime() > a.deadline_cycles) {
+          r = 0xFFFFFFFFu;
+          break;
+        }
+        __builtin_amdgcn_s_sleep(32);
+      }
+      s_go[0] = r;
+      if (r != 0xFFFFFFFFu) {
+        s_go[1] = a.go_key[w];
+        s_go[2] = a.go_flags[w];
+      }
     }
-    if (r == 0xFFFFFFFFu) return;  // stop
+    __syncthreads();
+    const unsigned int r = s_go[0];
+    if (r == 0xFFFFFFFFu) return;  // stop (or deadline abort)
     r_local = r;
-    const unsigned int key = a.go_key[w];
-    const unsigned int flags = a.go_flags[w];
+    const unsigned int key = s_go[1];
+    const unsigned int flags = s_go[2];
+    __syncthreads();
     if (a.algo == 1 && (flags & 1u)) commit_pass(a, w, b, flags >> 1);
     worker_round<XT>(a, w, b, key, flags, lds_g);
-    // group arrival: last block publishes done with release semantics
+    // group arrival: every thread fences ITS OWN global writes, then one
+    // thread per block counts the arrival; the group's last block
+    // publishes done with release semantics
     __threadfence();
-    const unsigned int arrived =
-        __hip_atomic_fetch_add(&a.done_ctr[w], 1u, __ATOMIC_ACQ_REL,
-                               __HIP_MEMORY_SCOPE_AGENT);
-    if (arrived + 1 == (unsigned int)a.G) {
-      __hip_atomic_store(&a.done_ctr[w], 0u, __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      store_rel(&a.done_round[w], r_local);
+    __syncthreads();
+    if (tid == 0) {
+      const unsigned int arrived =
+          __hip_atomic_fetch_add(&a.done_ctr[w], 1u, __ATOMIC_ACQ_REL,
+                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (arrived + 1 == (unsigned int)a.G) {
+        __hip_atomic_store(&a.done_ctr[w], 0u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        store_rel(&a.done_round[w], r_local);
+      }
     }
+    __syncthreads();
   }
 }
 
