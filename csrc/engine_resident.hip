@@ -96,14 +96,19 @@ __device__ __forceinline__ unsigned int load_acq(const unsigned int* p) {
   return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
 }
 
-// Spin-friendly flag read: RELAXED agent-scope atomic loads read the
-// coherent point WITHOUT the cache invalidation an acquire implies — a
-// chip full of spinning blocks doing acquire-loads storms every XCD's L2
-// with invalidates (measured: ~60x throughput loss). Poll relaxed, then
-// issue ONE acquire fence when the flag actually changed, before touching
-// the data it publishes.
-__device__ __forceinline__ unsigned int load_rlx(const unsigned int* p) {
-  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+// Spin-friendly flag read. Two traps measured on gfx950:
+//  * acquire-loads invalidate the XCD L2 on EVERY poll — a chip full of
+//    spinning blocks storms the caches (~60x throughput loss);
+//  * plain relaxed atomic LOADS can keep hitting a stale local L2 line
+//    (no invalidate => a flag write from another XCD is only observed on
+//    eviction, ~ms later).
+// A relaxed atomic RMW (fetch_add 0) always resolves at the coherent
+// point with no cache-wide side effects — the correct poll primitive.
+// Follow a positive poll with ONE acquire fence before touching the data
+// the flag publishes.
+__device__ __forceinline__ unsigned int load_rlx(unsigned int* p) {
+  return __hip_atomic_fetch_add(p, 0u, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
 }
 
 __device__ __forceinline__ void acq_fence() {
