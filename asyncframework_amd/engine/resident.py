@@ -36,6 +36,8 @@ class ResidentEngine:
             "host-spill history -> native engine"
         assert 1 <= cfg.num_workers <= 64
         assert cfg.batch_rate < 1.0
+        assert cfg.d <= 2048, \
+            "resident engine caps d at 2048 (register-cached x chunks)"
         self.cfg = cfg
         self.device = device
         self.G = blocks_per_worker

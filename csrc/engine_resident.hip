@@ -132,6 +132,9 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 // its LDS slab, then flush the slab into the worker's global g with
 // device-scope atomics. SAGA: also gather alpha, emit new scalars into the
 // staging region (g buffer tail), commit previous round's on go_flags bit0.
+#define RES_ROWCAP 1024  // sampled-row list per scan window (= window size)
+#define RES_MAXNV 32     // register-cached x chunks => d <= 64*32 = 2048
+
 template <typename XT>
 __device__ void worker_round(const ResidentArgs& a, int w, int b,
                              unsigned int key, unsigned int flags,
@@ -150,81 +153,80 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
   const int wave = tid >> 6;
   constexpr int WAVES = RES_BLOCK / 64;
 
+  __shared__ int rows_s[RES_ROWCAP];
+  __shared__ int qn_s;
+
   for (int j = tid; j < d; j += RES_BLOCK) lds_g[j] = 0.f;
-  __syncthreads();
 
   // static row partition: block b owns rows [b*slice, min(n, (b+1)*slice)),
   // 4-aligned so the 4-row Philox blocks stay within one owner
-  const long long slice =
-      ((n_rows + a.G - 1) / a.G + 3) & ~3ll;
+  const long long slice = ((n_rows + a.G - 1) / a.G + 3) & ~3ll;
   const long long r0 = (long long)b * slice;
   const long long r1 = (r0 + slice < n_rows) ? r0 + slice : n_rows;
+  const int nv = (d + 63) / 64;
 
-  // each wave walks 256-row chunks of the slice; within a chunk each lane
-  // evaluates ONE Philox block (4 consecutive rows)
-  for (long long base = r0 + (long long)wave * 256; base < r1;
-       base += (long long)WAVES * 256) {
+  // windows of WAVES*256 rows: scan (lane = one Philox block of 4 rows,
+  // sampled ids appended to the LDS list), then process WAVE-PER-ROW with
+  // coalesced loads and the x chunk held in registers between the dot and
+  // the rank-1 accumulate (single-lane row processing was ~0.3 ms/row:
+  // divergent scalar loads on a dependent FMA chain)
+  for (long long cb = r0; cb < r1; cb += (long long)WAVES * 256) {
+    __syncthreads();
+    if (tid == 0) qn_s = 0;
+    __syncthreads();
+    const long long base = cb + (long long)wave * 256;
     const long long row4 = base + (long long)lane * 4;
-    if (row4 >= r1) continue;
-    const uint4 rnd =
-        philox_block4(a.seed, key, (unsigned long long)(row_start + row4) / 4);
-    const unsigned int rv[4] = {rnd.x, rnd.y, rnd.z, rnd.w};
+    if (row4 < r1) {
+      const uint4 rnd = philox_block4(
+          a.seed, key, (unsigned long long)(row_start + row4) / 4);
+      const unsigned int rv[4] = {rnd.x, rnd.y, rnd.z, rnd.w};
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const long long r = row4 + q;
-      if (r >= r1 || rv[q] >= a.thresh) continue;
-      // sampled row r: whole-lane sequential dot + axpy (rows are sparse
-      // in the sample; lanes work independent rows => high MLP)
-      const XT* xr = X + (size_t)r * d;
-      float dot = 0.f;
-      for (int j = 0; j < d; ++j) {
-        const float xv = a.x_is_bf16 ? bf16_to_f32(((const unsigned short*)xr)[j])
-                                     : ((const float*)xr)[j];
-        dot += xv * wbuf[j];
+      for (int q = 0; q < 4; ++q) {
+        const long long r = row4 + q;
+        if (r < r1 && rv[q] < a.thresh)
+          rows_s[atomicAdd(&qn_s, 1)] = (int)r;
       }
+    }
+    __syncthreads();
+    const int qn = qn_s;
+    for (int i = wave; i < qn; i += WAVES) {
+      const int r = rows_s[i];
+      const XT* xr = X + (size_t)r * d;
+      float xv[RES_MAXNV];
+      float p = 0.f;
+#pragma unroll 4
+      for (int t = 0; t < nv; ++t) {
+        const int j = t * 64 + lane;
+        float v = 0.f;
+        if (j < d)
+          v = a.x_is_bf16 ? bf16_to_f32(((const unsigned short*)xr)[j])
+                          : ((const float*)xr)[j];
+        xv[t] = v;
+        p += v * (j < d ? wbuf[j] : 0.f);
+      }
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) p += __shfl_xor(p, off);
       float e;
       if (a.objective == 1)
-        e = 1.f / (1.f + __expf(-dot)) - y[r];
+        e = 1.f / (1.f + __expf(-p)) - y[r];
       else
-        e = dot - y[r];
-      // phase-offset the slab accumulate: the lanes run j in lockstep, so
-      // without an offset all 64 hit the SAME LDS address every step (a
-      // 64-way serialized atomic per element); 16 phases cut that 16x
-      const int phase = (d >= 16) ? (lane & 15) * (d >> 4) : 0;
+        e = p - y[r];
+      float scale = e;
       if (a.algo == 1) {
         // SAGA: corrected gradient (e - alpha_r) * x, new scalar = e
-        const float corr = e - alpha[r];
-        int j = phase;
-        for (int it = 0; it < d; ++it) {
-          const float xv = a.x_is_bf16
-                               ? bf16_to_f32(((const unsigned short*)xr)[j])
-                               : ((const float*)xr)[j];
-          atomicAdd(&lds_g[j], corr * xv);
-          if (++j == d) j = 0;
-        }
-        // stage the new scalar: alpha commit is accept-gated, so write to
-        // a per-row staging value == e; commit copies it in next round
-        ((float*)D[7])[r] = e;
-      } else {
-        int j = phase;
-        for (int it = 0; it < d; ++it) {
-          const float xv = a.x_is_bf16
-                               ? bf16_to_f32(((const unsigned short*)xr)[j])
-                               : ((const float*)xr)[j];
-          atomicAdd(&lds_g[j], e * xv);
-          if (++j == d) j = 0;
-        }
+        scale = e - alpha[r];
+        if (lane == 0) ((float*)D[7])[r] = e;  // accept-gated staging
+      }
+#pragma unroll 4
+      for (int t = 0; t < nv; ++t) {
+        const int j = t * 64 + lane;
+        if (j < d) atomicAdd(&lds_g[j], scale * xv[t]);
       }
     }
   }
   __syncthreads();
   for (int j = tid; j < d; j += RES_BLOCK)
     if (lds_g[j] != 0.f) atomicAdd(&g[j], lds_g[j]);
-
-  // SAGA commit of THIS round's scalars happens at the worker's NEXT
-  // dispatch (accept-gated): flags bit0 says the previous round was
-  // accepted => fold staged scalars of the previous round's sampled rows.
-  // (Handled before the gradient at round start; see commit_pass.)
   (void)flags;
 }
 
