@@ -163,7 +163,6 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
   const long long slice = ((n_rows + a.G - 1) / a.G + 3) & ~3ll;
   const long long r0 = (long long)b * slice;
   const long long r1 = (r0 + slice < n_rows) ? r0 + slice : n_rows;
-  const int nv = (d + 63) / 64;
 
   // windows of WAVES*256 rows: scan (lane = one Philox block of 4 rows,
   // sampled ids appended to the LDS list), then process WAVE-PER-ROW with
@@ -189,20 +188,32 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
     }
     __syncthreads();
     const int qn = qn_s;
+    const int nfull = d >> 6;        // unconditional 64-wide chunks — a
+    const int tail = d - (nfull << 6);  // per-element j<d check makes hipcc
+                                        // drain vmcnt at every load
     for (int i = wave; i < qn; i += WAVES) {
       const int r = rows_s[i];
       const XT* xr = X + (size_t)r * d;
       float xv[RES_MAXNV];
       float p = 0.f;
 #pragma unroll 4
-      for (int t = 0; t < nv; ++t) {
+      for (int t = 0; t < nfull; ++t) {
         const int j = t * 64 + lane;
+        const float v = a.x_is_bf16
+                            ? bf16_to_f32(((const unsigned short*)xr)[j])
+                            : ((const float*)xr)[j];
+        xv[t] = v;
+        p += v * wbuf[j];
+      }
+      if (tail) {
         float v = 0.f;
-        if (j < d)
+        const int j = nfull * 64 + lane;
+        if (lane < tail) {
           v = a.x_is_bf16 ? bf16_to_f32(((const unsigned short*)xr)[j])
                           : ((const float*)xr)[j];
-        xv[t] = v;
-        p += v * (j < d ? wbuf[j] : 0.f);
+          p += v * wbuf[j];
+        }
+        xv[nfull] = v;
       }
 #pragma unroll
       for (int off = 32; off > 0; off >>= 1) p += __shfl_xor(p, off);
@@ -218,10 +229,10 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
         if (lane == 0) ((float*)D[7])[r] = e;  // accept-gated staging
       }
 #pragma unroll 4
-      for (int t = 0; t < nv; ++t) {
-        const int j = t * 64 + lane;
-        if (j < d) atomicAdd(&lds_g[j], scale * xv[t]);
-      }
+      for (int t = 0; t < nfull; ++t)
+        atomicAdd(&lds_g[t * 64 + lane], scale * xv[t]);
+      if (tail && lane < tail)
+        atomicAdd(&lds_g[nfull * 64 + lane], scale * xv[nfull]);
     }
   }
   __syncthreads();

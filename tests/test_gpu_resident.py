@@ -104,7 +104,8 @@ def test_resident_asaga_matches_native_p1():
     ra = r_eng.alpha_tables[0]
     na = n_eng.alpha_tables[0]
     assert int((ra != 0).sum()) > 0
-    assert torch.allclose(ra, na, atol=1e-4)
+    # alpha = e(w at the sampling round): inherits the w drift bound above
+    assert torch.allclose(ra, na, atol=5e-3, rtol=1e-3)
 
 
 def test_resident_tau_filter_rejects():
