@@ -75,7 +75,7 @@ class ResidentEngine:
             for name in ("go_round", "go_key", "go_flags", "done_round",
                          "done_ctr")
         }
-        self._out = torch.zeros(10, dtype=torch.int64, device=device)
+        self._out = torch.zeros(16, dtype=torch.int64, device=device)
 
     def run(self, num_iterations: Optional[int] = None,
             mark_lo: int = -1, mark_hi: int = -1,

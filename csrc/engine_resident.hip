@@ -85,7 +85,10 @@ struct ResidentArgs {
 
 enum {
   OUT_K = 0, OUT_APPLIED, OUT_REJECTED, OUT_MAXSTALE, OUT_MARKLO_C,
-  OUT_MARKHI_C, OUT_ABORT, OUT_T0_C, OUT_TEND_C, OUT_SNAPN, OUT_N
+  OUT_MARKHI_C, OUT_ABORT, OUT_T0_C, OUT_TEND_C, OUT_SNAPN,
+  // device-side profiling (cycles / counts)
+  OUT_SRV_LOOPS, OUT_SRV_SWEEP_C, OUT_SRV_DISPATCH_C, OUT_W0_ROUNDS,
+  OUT_W0_GRAD_C, OUT_W0_SPIN_C, OUT_N
 };
 
 __device__ __forceinline__ unsigned long long realtime() {
@@ -423,6 +426,7 @@ __device__ void server_block(const ResidentArgs& a) {
     }
     __syncthreads();
     if (s_done) break;
+    const unsigned long long prof_t0 = (tid == 0) ? realtime() : 0;
 
     // ---- completion sweep (thread 0 decides; block applies) ----
     for (int w = 0; w < a.P; ++w) {
@@ -528,6 +532,8 @@ __device__ void server_block(const ResidentArgs& a) {
     // ---- redispatch pass (quorum gate + straggler model) ----
     __syncthreads();
     if (tid == 0) {
+      a.out[OUT_SRV_LOOPS] += 1;
+      a.out[OUT_SRV_SWEEP_C] += realtime() - prof_t0;
       st.ndis = 0;
       int avail = 0;
       for (int i = 0; i < a.P; ++i)
@@ -539,6 +545,7 @@ __device__ void server_block(const ResidentArgs& a) {
       }
     }
     __syncthreads();
+    const unsigned long long prof_t1 = (tid == 0) ? realtime() : 0;
     const int ndis = st.ndis;
     for (int i = 0; i < ndis; ++i) {
       __syncthreads();
@@ -584,6 +591,7 @@ __device__ void server_block(const ResidentArgs& a) {
         __syncthreads();
       }
     }
+    if (tid == 0) a.out[OUT_SRV_DISPATCH_C] += realtime() - prof_t1;
     __builtin_amdgcn_s_sleep(8);
   }
 
@@ -621,6 +629,7 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     // through LDS. Throttled and deadline-guarded.
     if (tid == 0) {
       unsigned int r;
+      const unsigned long long spin_t0 = realtime();
       while (true) {
         r = load_rlx(&a.go_round[w]);
         if (r != r_local) {
@@ -633,6 +642,8 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
         }
         __builtin_amdgcn_s_sleep(32);
       }
+      if (w == 0 && b == 0)
+        a.out[OUT_W0_SPIN_C] += realtime() - spin_t0;
       s_go[0] = r;
       if (r != 0xFFFFFFFFu) {
         s_go[1] = a.go_key[w];
@@ -646,8 +657,14 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     const unsigned int key = s_go[1];
     const unsigned int flags = s_go[2];
     __syncthreads();
+    const unsigned long long grad_t0 =
+        (tid == 0 && w == 0 && b == 0) ? realtime() : 0;
     if (a.algo == 1 && (flags & 1u)) commit_pass(a, w, b, flags >> 1);
     worker_round<XT>(a, w, b, key, flags, lds_g);
+    if (tid == 0 && w == 0 && b == 0) {
+      a.out[OUT_W0_GRAD_C] += realtime() - grad_t0;
+      a.out[OUT_W0_ROUNDS] += 1;
+    }
     // group arrival: every thread fences ITS OWN global writes, then one
     // thread per block counts the arrival; the group's last block
     // publishes done with release semantics
@@ -837,6 +854,13 @@ void register_resident_engine(py::module_& m) {
     r["snap_n"] = (long long)out_h[OUT_SNAPN];
     r["cycles_per_ms"] = cpm;
     r["t0_cycles"] = (unsigned long long)out_h[OUT_T0_C];
+    // device-side profiling (where the cycles went)
+    r["srv_loops"] = (long long)out_h[OUT_SRV_LOOPS];
+    r["srv_sweep_ms"] = (double)out_h[OUT_SRV_SWEEP_C] / cpm;
+    r["srv_dispatch_ms"] = (double)out_h[OUT_SRV_DISPATCH_C] / cpm;
+    r["w0_rounds"] = (long long)out_h[OUT_W0_ROUNDS];
+    r["w0_grad_ms"] = (double)out_h[OUT_W0_GRAD_C] / cpm;
+    r["w0_spin_ms"] = (double)out_h[OUT_W0_SPIN_C] / cpm;
     return r;
   });
 }
