@@ -26,7 +26,7 @@ _ALGO = {"asgd": 0, "asaga": 1}
 
 class ResidentEngine:
     def __init__(self, cfg: EngineConfig, shards: List[Shard],
-                 device: torch.device, blocks_per_worker: int = 8):
+                 device: torch.device, blocks_per_worker: int = 16):
         from .. import _hip_core
         self._core = _hip_core
         assert device.type == "cuda", "resident engine is GPU-only"
