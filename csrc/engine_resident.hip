@@ -194,48 +194,97 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
     const int nfull = d >> 6;        // unconditional 64-wide chunks — a
     const int tail = d - (nfull << 6);  // per-element j<d check makes hipcc
                                         // drain vmcnt at every load
-    for (int i = wave; i < qn; i += WAVES) {
-      const int r = rows_s[i];
-      const XT* xr = X + (size_t)r * d;
-      float xv[RES_MAXNV];
-      float p = 0.f;
-#pragma unroll 4
+    // R rows in flight per wave: the dot pass issues R independent
+    // coalesced load streams (hides the ~600-cycle HBM latency that made
+    // one-row-at-a-time ~2.5 us/row); the accumulate pass RE-READS x from
+    // L1/L2 (cheaper than register-caching the chunks, which made hipcc
+    // spill the runtime-indexed array to scratch) and does ONE LDS atomic
+    // per chunk for all R rows.
+    constexpr int R = 4;
+    for (int i0 = wave; i0 < qn; i0 += WAVES * R) {
+      const XT* xp[R];
+      float p[R];
+      int nr = 0;
+#pragma unroll
+      for (int u = 0; u < R; ++u) {
+        p[u] = 0.f;
+        const int idx = i0 + u * WAVES;
+        if (idx < qn) {
+          xp[u] = X + (size_t)rows_s[idx] * d;
+          nr = u + 1;
+        } else {
+          xp[u] = X;  // dummy; masked out of p/scale below
+        }
+      }
       for (int t = 0; t < nfull; ++t) {
         const int j = t * 64 + lane;
-        const float v = a.x_is_bf16
-                            ? bf16_to_f32(((const unsigned short*)xr)[j])
-                            : ((const float*)xr)[j];
-        xv[t] = v;
-        p += v * wbuf[j];
-      }
-      if (tail) {
-        float v = 0.f;
-        const int j = nfull * 64 + lane;
-        if (lane < tail) {
-          v = a.x_is_bf16 ? bf16_to_f32(((const unsigned short*)xr)[j])
-                          : ((const float*)xr)[j];
-          p += v * wbuf[j];
-        }
-        xv[nfull] = v;
-      }
+        const float wv = wbuf[j];
 #pragma unroll
-      for (int off = 32; off > 0; off >>= 1) p += __shfl_xor(p, off);
-      float e;
-      if (a.objective == 1)
-        e = 1.f / (1.f + __expf(-p)) - y[r];
-      else
-        e = p - y[r];
-      float scale = e;
-      if (a.algo == 1) {
-        // SAGA: corrected gradient (e - alpha_r) * x, new scalar = e
-        scale = e - alpha[r];
-        if (lane == 0) ((float*)D[7])[r] = e;  // accept-gated staging
+        for (int u = 0; u < R; ++u) {
+          const float v = a.x_is_bf16
+                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
+                              : ((const float*)xp[u])[j];
+          p[u] += v * wv;
+        }
       }
-#pragma unroll 4
-      for (int t = 0; t < nfull; ++t)
-        atomicAdd(&lds_g[t * 64 + lane], scale * xv[t]);
-      if (tail && lane < tail)
-        atomicAdd(&lds_g[nfull * 64 + lane], scale * xv[nfull]);
+      if (tail && lane < tail) {
+        const int j = nfull * 64 + lane;
+        const float wv = wbuf[j];
+#pragma unroll
+        for (int u = 0; u < R; ++u) {
+          const float v = a.x_is_bf16
+                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
+                              : ((const float*)xp[u])[j];
+          p[u] += v * wv;
+        }
+      }
+      float scale[R];
+#pragma unroll
+      for (int u = 0; u < R; ++u) {
+        float pu = p[u];
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) pu += __shfl_xor(pu, off);
+        if (u >= nr) {
+          scale[u] = 0.f;
+          continue;
+        }
+        const int r = rows_s[i0 + u * WAVES];
+        float e;
+        if (a.objective == 1)
+          e = 1.f / (1.f + __expf(-pu)) - y[r];
+        else
+          e = pu - y[r];
+        scale[u] = e;
+        if (a.algo == 1) {
+          // SAGA: corrected gradient (e - alpha_r) * x, new scalar = e
+          scale[u] = e - alpha[r];
+          if (lane == 0) ((float*)D[7])[r] = e;  // accept-gated staging
+        }
+      }
+      for (int t = 0; t < nfull; ++t) {
+        const int j = t * 64 + lane;
+        float acc = 0.f;
+#pragma unroll
+        for (int u = 0; u < R; ++u) {
+          const float v = a.x_is_bf16
+                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
+                              : ((const float*)xp[u])[j];
+          acc += scale[u] * v;
+        }
+        atomicAdd(&lds_g[j], acc);
+      }
+      if (tail && lane < tail) {
+        const int j = nfull * 64 + lane;
+        float acc = 0.f;
+#pragma unroll
+        for (int u = 0; u < R; ++u) {
+          const float v = a.x_is_bf16
+                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
+                              : ((const float*)xp[u])[j];
+          acc += scale[u] * v;
+        }
+        atomicAdd(&lds_g[j], acc);
+      }
     }
   }
   __syncthreads();
