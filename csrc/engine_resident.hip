@@ -679,10 +679,11 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     if (tid == 0) {
       unsigned int r;
       const unsigned long long spin_t0 = realtime();
-      int slp = 2;  // exponential backoff: fast reaction to a quick go,
-                    // low poll pressure when idle (hundreds of spinning
-                    // blocks doing ~1 atomic RMW/us saturate the coherent
-                    // fabric and slow every other memory client)
+      int spins = 0;  // exponential backoff: fast reaction to a quick go,
+                      // low poll pressure when idle (hundreds of spinning
+                      // blocks doing ~1 atomic RMW/us saturate the coherent
+                      // fabric and slow every other memory client).
+                      // s_sleep needs an immediate, hence the ladder.
       while (true) {
         r = load_rlx(&a.go_round[w]);
         if (r != r_local) {
@@ -693,8 +694,13 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
           r = 0xFFFFFFFFu;
           break;
         }
-        __builtin_amdgcn_s_sleep(slp);
-        if (slp < 127) slp <<= 1;
+        if (spins < 4)
+          __builtin_amdgcn_s_sleep(2);
+        else if (spins < 8)
+          __builtin_amdgcn_s_sleep(16);
+        else
+          __builtin_amdgcn_s_sleep(127);
+        ++spins;
       }
       if (w == 0 && b == 0)
         a.out[OUT_W0_SPIN_C] += realtime() - spin_t0;
