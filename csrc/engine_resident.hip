@@ -128,6 +128,15 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
   return c.f;
 }
 
+// compile-time element conversion: a runtime `x_is_bf16 ?` ternary keeps a
+// BRANCH around every load, and branchy loads make the compiler drain
+// vmcnt(0) per element (zero memory-level parallelism — the same trap the
+// r01 pipe kernel hit; verified in this kernel's ISA)
+__device__ __forceinline__ float to_float(float v) { return v; }
+__device__ __forceinline__ float to_float(unsigned short u) {
+  return bf16_to_f32(u);
+}
+
 // ---------------------------------------------------------------- worker
 
 // One worker-group block's share of a round: scan its slice of the shard
@@ -220,23 +229,13 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
         const int j = t * 64 + lane;
         const float wv = wbuf[j];
 #pragma unroll
-        for (int u = 0; u < R; ++u) {
-          const float v = a.x_is_bf16
-                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
-                              : ((const float*)xp[u])[j];
-          p[u] += v * wv;
-        }
+        for (int u = 0; u < R; ++u) p[u] += to_float(xp[u][j]) * wv;
       }
       if (tail && lane < tail) {
         const int j = nfull * 64 + lane;
         const float wv = wbuf[j];
 #pragma unroll
-        for (int u = 0; u < R; ++u) {
-          const float v = a.x_is_bf16
-                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
-                              : ((const float*)xp[u])[j];
-          p[u] += v * wv;
-        }
+        for (int u = 0; u < R; ++u) p[u] += to_float(xp[u][j]) * wv;
       }
       float scale[R];
 #pragma unroll
@@ -265,24 +264,14 @@ __device__ void worker_round(const ResidentArgs& a, int w, int b,
         const int j = t * 64 + lane;
         float acc = 0.f;
 #pragma unroll
-        for (int u = 0; u < R; ++u) {
-          const float v = a.x_is_bf16
-                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
-                              : ((const float*)xp[u])[j];
-          acc += scale[u] * v;
-        }
+        for (int u = 0; u < R; ++u) acc += scale[u] * to_float(xp[u][j]);
         atomicAdd(&lds_g[j], acc);
       }
       if (tail && lane < tail) {
         const int j = nfull * 64 + lane;
         float acc = 0.f;
 #pragma unroll
-        for (int u = 0; u < R; ++u) {
-          const float v = a.x_is_bf16
-                              ? bf16_to_f32(((const unsigned short*)xp[u])[j])
-                              : ((const float*)xp[u])[j];
-          acc += scale[u] * v;
-        }
+        for (int u = 0; u < R; ++u) acc += scale[u] * to_float(xp[u][j]);
         atomicAdd(&lds_g[j], acc);
       }
     }
