@@ -89,6 +89,11 @@ enum {
   // device-side profiling (cycles / counts)
   OUT_SRV_LOOPS, OUT_SRV_SWEEP_C, OUT_SRV_DISPATCH_C, OUT_W0_ROUNDS,
   OUT_W0_GRAD_C, OUT_W0_SPIN_C,
+  // realtime of the last completion processed: on a deadline abort this
+  // separates a LIVE run that ran out of wall budget (e.g. tau=0 rejecting
+  // everything — the host engines return partial results there too) from a
+  // protocol wedge (no completions at all for the final stretch)
+  OUT_LAST_PROG_C,
   // abort diagnostics: per-worker server state, packed (see server_block's
   // abort dump); only written when the deadline watchdog fires
   OUT_DUMP, OUT_N = OUT_DUMP + RES_MAXP
@@ -574,6 +579,7 @@ __device__ void server_block(const ResidentArgs& a) {
         st.nseg += 1;
       }
       st.k = kv;
+      if (st.nacc + st.nrej > 0) a.out[OUT_LAST_PROG_C] = realtime();
     }
     __syncthreads();
     // ONE acquire for all of this sweep's completions, by every thread
@@ -987,6 +993,10 @@ void register_resident_engine(py::module_& m) {
     r["w0_rounds"] = (long long)out_h[OUT_W0_ROUNDS];
     r["w0_grad_ms"] = (double)out_h[OUT_W0_GRAD_C] / cpm;
     r["w0_spin_ms"] = (double)out_h[OUT_W0_SPIN_C] / cpm;
+    r["last_progress_ms"] =
+        out_h[OUT_LAST_PROG_C]
+            ? ((double)out_h[OUT_LAST_PROG_C] - t0c) / cpm
+            : -1.0;
     return r;
   });
 }
