@@ -655,7 +655,12 @@ __device__ void server_block(const ResidentArgs& a) {
       int avail = 0;
       for (int i = 0; i < a.P; ++i)
         if (!st.busy[i]) ++avail;
-      const int n = (st.pq_n > 0 && avail >= a.gate) ? st.pq_n : 0;
+      // no new rounds once k hit the budget (the host engines stop
+      // dispatching there too; an extra tail round would commit one more
+      // SAGA staging batch than the native engine — observed as a flaky
+      // alpha-pattern mismatch at P=1 before this guard)
+      const int n =
+          (st.pq_n > 0 && avail >= a.gate && st.k < a.iters) ? st.pq_n : 0;
       for (int i = 0; i < n; ++i) {
         const int w = st.pendq[st.pq_head];
         st.pq_head = (st.pq_head + 1) % RES_MAXP;
