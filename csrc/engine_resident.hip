@@ -721,9 +721,18 @@ __device__ void server_block(const ResidentArgs& a) {
     __builtin_amdgcn_s_sleep(8);
   }
 
-  // stop: release workers
+  // stop: release workers. SAGA: piggyback a FINAL commit order for
+  // workers whose last accepted round was never followed by a dispatch
+  // (the native engine's tail dispatch commits these; busy workers
+  // already committed at their current dispatch, and straggler-held
+  // workers stay uncommitted there too — matched exactly)
   __syncthreads();
   if (tid == 0) {
+    for (int w = 0; w < a.P; ++w)
+      a.go_flags[w] = (a.algo == 1 && !st.busy[w] && st.last_accept[w] &&
+                       st.prev_key[w])
+                          ? ((st.prev_key[w] << 1) | 1u)
+                          : 0u;
     __threadfence();
     for (int w = 0; w < a.P; ++w) store_rel(&a.go_round[w], 0xFFFFFFFFu);
     a.out[OUT_K] = (unsigned long long)st.k;
@@ -768,9 +777,9 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
           break;
         }
         if (realtime() > a.deadline_cycles) {
-          r = 0xFFFFFFFFu;
-          break;
-        }
+          r = 0xFFFFFFFEu;  // distinct from the server's stop sentinel:
+          break;            // a deadline exit must NOT run the final
+        }                   // commit (go_flags may be stale)
         if (spins < 4)
           __builtin_amdgcn_s_sleep(2);
         else if (spins < 8)
@@ -782,14 +791,18 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
       if (w == 0 && b == 0)
         a.out[OUT_W0_SPIN_C] += realtime() - spin_t0;
       s_go[0] = r;
-      if (r != 0xFFFFFFFFu) {
+      if (r != 0xFFFFFFFEu) {
         s_go[1] = a.go_key[w];
         s_go[2] = a.go_flags[w];
       }
     }
     __syncthreads();
     const unsigned int r = s_go[0];
-    if (r == 0xFFFFFFFFu) return;  // stop (or deadline abort)
+    if (r == 0xFFFFFFFEu) return;  // deadline abort: exit, no commit
+    if (r == 0xFFFFFFFFu) {        // stop: final pending SAGA commit
+      if (a.algo == 1 && (s_go[2] & 1u)) commit_pass(a, w, b, s_go[2] >> 1);
+      return;
+    }
     r_local = r;
     const unsigned int key = s_go[1];
     const unsigned int flags = s_go[2];
