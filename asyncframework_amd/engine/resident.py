@@ -26,7 +26,7 @@ _ALGO = {"asgd": 0, "asaga": 1}
 
 class ResidentEngine:
     def __init__(self, cfg: EngineConfig, shards: List[Shard],
-                 device: torch.device, blocks_per_worker: int = 16):
+                 device: torch.device, blocks_per_worker: int = 0):
         from .. import _hip_core
         self._core = _hip_core
         assert device.type == "cuda", "resident engine is GPU-only"
@@ -40,7 +40,10 @@ class ResidentEngine:
             "resident engine caps d at 2048 (register-cached x chunks)"
         self.cfg = cfg
         self.device = device
-        self.G = blocks_per_worker
+        # G=0: auto-size to fill the chip (256 CUs want >>256 workgroups;
+        # measured: G=31 at P=32 beats G=16 by ~8%). The launch retries
+        # with halved G if occupancy rejects the grid (run() below).
+        self.G = blocks_per_worker or max(2, min(31, 1008 // cfg.num_workers))
         d = cfg.d
         self.w = torch.zeros(d, dtype=torch.float32, device=device)
         self.alpha_bar = torch.zeros(d, dtype=torch.float32, device=device)
@@ -117,7 +120,18 @@ class ResidentEngine:
             **{k: v.data_ptr() for k, v in self._ctl.items()},
         )
         torch.cuda.synchronize()
-        res = self._core.resident_run(conf)
+        while True:
+            try:
+                res = self._core.resident_run(conf)
+                break
+            except RuntimeError as e:
+                # co-residency refusal (grid would not be simultaneously
+                # schedulable => persistent kernel would deadlock): halve G
+                if "co-resident" in str(e) and self.G > 2:
+                    self.G = max(2, self.G // 2)
+                    conf["G"] = self.G
+                    continue
+                raise
         torch.cuda.synchronize()
         res["wall_exhausted"] = False
         if res["aborted"]:
