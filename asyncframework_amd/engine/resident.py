@@ -73,8 +73,11 @@ class ResidentEngine:
                 self._keep += [alpha, stage]
         self._desc = desc.to(device)
         P = cfg.num_workers
+        # one 128-B cache line PER WORKER per flag (RES_CSTRIDE in the
+        # kernel): hundreds of spinning blocks poll these with atomic RMWs,
+        # and packed flags serialize every poll on 1-2 coherent lines
         self._ctl = {
-            name: torch.zeros(P, dtype=torch.int32, device=device)
+            name: torch.zeros(P * 32, dtype=torch.int32, device=device)
             for name in ("go_round", "go_key", "go_flags", "done_round",
                          "done_ctr")
         }
@@ -149,9 +152,9 @@ class ResidentEngine:
                     dict(w=i, busy=v & 1, hold=(v >> 1) & 1,
                          round_no=(v >> 8) & 0xFFFFFFFF,
                          ksub=(v >> 40) & 0xFFFF,
-                         go=int(self._ctl["go_round"][i]),
-                         done=int(self._ctl["done_round"][i]),
-                         ctr=int(self._ctl["done_ctr"][i]))
+                         go=int(self._ctl["go_round"][i * 32]),
+                         done=int(self._ctl["done_round"][i * 32]),
+                         ctr=int(self._ctl["done_ctr"][i * 32]))
                     for i, v in enumerate(dump)]
                 raise RuntimeError(
                     f"resident engine wedged at k={res['k']} (device "

@@ -44,6 +44,11 @@
 
 #define RES_BLOCK 256
 #define RES_MAXP 64
+// control flags are padded one per 128-B cache line: hundreds of idle
+// blocks poll these with atomic RMWs, and when all P flags share 2 lines
+// the coherent point serializes every poll against every other (measured
+// as a ~2x gradient-round slowdown at P=32 vs isolated)
+#define RES_CSTRIDE 32
 
 namespace {
 
@@ -468,12 +473,13 @@ __device__ void server_block(const ResidentArgs& a) {
       st.ksub[w] = st.k;
       st.busy[w] = 1;
       st.submit_c[w] = realtime();
-      a.go_key[w] = (unsigned int)(st.ksub[w] + 1);
-      a.go_flags[w] = 0;
+      a.go_key[w * RES_CSTRIDE] = (unsigned int)(st.ksub[w] + 1);
+      a.go_flags[w * RES_CSTRIDE] = 0;
       st.round_no[w] += 1;
     }
     __threadfence();
-    for (int w = 0; w < a.P; ++w) store_rel(&a.go_round[w], st.round_no[w]);
+    for (int w = 0; w < a.P; ++w)
+      store_rel(&a.go_round[w * RES_CSTRIDE], st.round_no[w]);
   }
   __syncthreads();
 
@@ -503,7 +509,8 @@ __device__ void server_block(const ResidentArgs& a) {
 
     // ---- phase A: parallel completion poll (wave 0, one lane per
     // worker — P serialized thread-0 RMWs were ~0.5 us each) ----
-    if (tid < a.P) st.done_snap[tid] = load_rlx(&a.done_round[tid]);
+    if (tid < a.P)
+      st.done_snap[tid] = load_rlx(&a.done_round[tid * RES_CSTRIDE]);
     __syncthreads();
 
     // ---- phase B: scalar classify (thread 0; builds the op lists) ----
@@ -703,10 +710,10 @@ __device__ void server_block(const ResidentArgs& a) {
           st.busy[w] = 1;
           st.due_c[w] = 0;
           st.submit_c[w] = realtime();
-          a.go_key[w] = (unsigned int)(st.ksub[w] + 1);
+          a.go_key[w * RES_CSTRIDE] = (unsigned int)(st.ksub[w] + 1);
           // SAGA: bit0 = commit previous accepted round's scalars;
           // bits 1..31 carry the previous key
-          a.go_flags[w] =
+          a.go_flags[w * RES_CSTRIDE] =
               (a.algo == 1 && st.last_accept[w] && st.prev_key[w])
                   ? ((st.prev_key[w] << 1) | 1u)
                   : 0u;
@@ -714,7 +721,8 @@ __device__ void server_block(const ResidentArgs& a) {
         }
         __threadfence();  // publish go_key/go_flags before the go stores
         for (int u = 0; u < ndis; ++u)
-          store_rel(&a.go_round[st.dis_w[u]], st.round_no[st.dis_w[u]]);
+          store_rel(&a.go_round[st.dis_w[u] * RES_CSTRIDE],
+                    st.round_no[st.dis_w[u]]);
       }
     }
     if (tid == 0) a.out[OUT_SRV_DISPATCH_C] += realtime() - prof_t1;
@@ -729,12 +737,14 @@ __device__ void server_block(const ResidentArgs& a) {
   __syncthreads();
   if (tid == 0) {
     for (int w = 0; w < a.P; ++w)
-      a.go_flags[w] = (a.algo == 1 && !st.busy[w] && st.last_accept[w] &&
+      a.go_flags[w * RES_CSTRIDE] = (a.algo == 1 && !st.busy[w] &&
+                                     st.last_accept[w] &&
                        st.prev_key[w])
                           ? ((st.prev_key[w] << 1) | 1u)
                           : 0u;
     __threadfence();
-    for (int w = 0; w < a.P; ++w) store_rel(&a.go_round[w], 0xFFFFFFFFu);
+    for (int w = 0; w < a.P; ++w)
+      store_rel(&a.go_round[w * RES_CSTRIDE], 0xFFFFFFFFu);
     a.out[OUT_K] = (unsigned long long)st.k;
     a.out[OUT_APPLIED] = (unsigned long long)st.applied;
     a.out[OUT_REJECTED] = (unsigned long long)st.rejected;
@@ -771,7 +781,7 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
                       // fabric and slow every other memory client).
                       // s_sleep needs an immediate, hence the ladder.
       while (true) {
-        r = load_rlx(&a.go_round[w]);
+        r = load_rlx(&a.go_round[w * RES_CSTRIDE]);
         if (r != r_local) {
           acq_fence();  // one invalidate, only on a real transition
           break;
@@ -792,8 +802,8 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
         a.out[OUT_W0_SPIN_C] += realtime() - spin_t0;
       s_go[0] = r;
       if (r != 0xFFFFFFFEu) {
-        s_go[1] = a.go_key[w];
-        s_go[2] = a.go_flags[w];
+        s_go[1] = a.go_key[w * RES_CSTRIDE];
+        s_go[2] = a.go_flags[w * RES_CSTRIDE];
       }
     }
     __syncthreads();
@@ -822,12 +832,14 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     __syncthreads();
     if (tid == 0) {
       const unsigned int arrived =
-          __hip_atomic_fetch_add(&a.done_ctr[w], 1u, __ATOMIC_ACQ_REL,
+          __hip_atomic_fetch_add(&a.done_ctr[w * RES_CSTRIDE], 1u,
+                                 __ATOMIC_ACQ_REL,
                                  __HIP_MEMORY_SCOPE_AGENT);
       if (arrived + 1 == (unsigned int)a.G) {
-        __hip_atomic_store(&a.done_ctr[w], 0u, __ATOMIC_RELAXED,
+        __hip_atomic_store(&a.done_ctr[w * RES_CSTRIDE], 0u,
+                           __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
-        store_rel(&a.done_round[w], r_local);
+        store_rel(&a.done_round[w * RES_CSTRIDE], r_local);
       }
     }
     __syncthreads();
