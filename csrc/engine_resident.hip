@@ -369,6 +369,7 @@ struct ServerState {
   unsigned long long dis_wbuf[RES_MAXP];  // batched dispatch fan-out
   int dis_w[RES_MAXP];
   int ndis;
+  unsigned long long sweep_c;  // loop-top clock, reused across the sweep
   long long k;
   int clock_;
   long long applied, rejected, max_stale;
@@ -489,6 +490,10 @@ __device__ void server_block(const ResidentArgs& a) {
     if (tid == 0) {
       s_done = 0;
       const unsigned long long now_c = realtime();
+      st.sweep_c = now_c;  // ONE clock read per sweep: s_memrealtime is a
+                           // slow memory-mapped SALU read, and per-worker
+                           // reads (submit/finish/due) were ~40% of the
+                           // server's serial batch time at P=32
       if (st.k >= a.iters) s_done = 1;
       if (now_c > a.deadline_cycles) {
         a.out[OUT_ABORT] = 1;
@@ -522,7 +527,7 @@ __device__ void server_block(const ResidentArgs& a) {
       for (int w = 0; w < a.P; ++w) {
         if (!st.busy[w]) continue;
         if (st.due_c[w] != 0) {
-          if (realtime() >= st.due_c[w]) {
+          if (st.sweep_c >= st.due_c[w]) {
             // straggler release: back to pending for immediate dispatch
             st.due_c[w] = 0;
             st.busy[w] = 0;
@@ -532,9 +537,8 @@ __device__ void server_block(const ResidentArgs& a) {
           continue;
         }
         if (st.done_snap[w] != st.round_no[w] || kv >= a.iters) continue;
-        const unsigned long long now_c = realtime();
         st.busy[w] = 0;
-        st.finish_c[w] = now_c;
+        st.finish_c[w] = st.sweep_c;
         const int staleness = st.clock_ - st.ts[w];
         st.clock_ += 1;
         if (staleness > st.max_stale) st.max_stale = staleness;
@@ -544,7 +548,8 @@ __device__ void server_block(const ResidentArgs& a) {
         st.pq_n += 1;
         if (accept) {
           if (kv < a.calib_window) {
-            st.cul_ms += (double)(now_c - st.submit_c[w]) / a.cycles_per_ms;
+            st.cul_ms +=
+                (double)(st.sweep_c - st.submit_c[w]) / a.cycles_per_ms;
             st.cul_n += 1;
           }
           const double gamma_k = a.gamma / sqrt((double)(kv / a.P + 1));
@@ -586,7 +591,7 @@ __device__ void server_block(const ResidentArgs& a) {
         st.nseg += 1;
       }
       st.k = kv;
-      if (st.nacc + st.nrej > 0) a.out[OUT_LAST_PROG_C] = realtime();
+      if (st.nacc + st.nrej > 0) a.out[OUT_LAST_PROG_C] = st.sweep_c;
     }
     __syncthreads();
     // ONE acquire for all of this sweep's completions, by every thread
@@ -676,7 +681,7 @@ __device__ void server_block(const ResidentArgs& a) {
         if (dly > 0) {
           st.busy[w] = 1;
           st.due_c[w] =
-              realtime() + (unsigned long long)(dly * a.cycles_per_ms);
+              st.sweep_c + (unsigned long long)(dly * a.cycles_per_ms);
         } else {
           st.dis_w[st.ndis] = w;
           st.dis_wbuf[st.ndis] = a.desc[(size_t)w * 8 + 2];
@@ -703,13 +708,14 @@ __device__ void server_block(const ResidentArgs& a) {
       __threadfence();  // publish this thread's snapshots + g zeroing
       __syncthreads();
       if (tid == 0) {
+        const unsigned long long sub_c = realtime();
         for (int u = 0; u < ndis; ++u) {
           const int w = st.dis_w[u];
           st.ts[w] = st.clock_;
           st.ksub[w] = st.k;
           st.busy[w] = 1;
           st.due_c[w] = 0;
-          st.submit_c[w] = realtime();
+          st.submit_c[w] = sub_c;
           a.go_key[w * RES_CSTRIDE] = (unsigned int)(st.ksub[w] + 1);
           // SAGA: bit0 = commit previous accepted round's scalars;
           // bits 1..31 carry the previous key
@@ -786,7 +792,9 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
           acq_fence();  // one invalidate, only on a real transition
           break;
         }
-        if (realtime() > a.deadline_cycles) {
+        // deadline check amortized: s_memrealtime is slow, and checking
+        // it every poll added its latency to the go-reaction time
+        if ((spins & 15) == 15 && realtime() > a.deadline_cycles) {
           r = 0xFFFFFFFEu;  // distinct from the server's stop sentinel:
           break;            // a deadline exit must NOT run the final
         }                   // commit (go_flags may be stale)
