@@ -81,7 +81,7 @@ class ResidentEngine:
             for name in ("go_round", "go_key", "go_flags", "done_round",
                          "done_ctr")
         }
-        # 17 counters + RES_MAXP packed per-worker abort-diagnostic slots
+        # 26 counters + RES_MAXP packed per-worker abort-diagnostic slots
         # (OUT_N in csrc/engine_resident.hip; 96 leaves headroom)
         self._out = torch.zeros(96, dtype=torch.int64, device=device)
 
@@ -147,7 +147,7 @@ class ResidentEngine:
             if res["last_progress_ms"] >= 0 and gap_ms < 500.0:
                 res["wall_exhausted"] = True
             else:
-                dump = self._out[17:17 + cfg.num_workers].cpu().tolist()
+                dump = self._out[26:26 + cfg.num_workers].cpu().tolist()
                 state = [
                     dict(w=i, busy=v & 1, hold=(v >> 1) & 1,
                          round_no=(v >> 8) & 0xFFFFFFFF,

@@ -99,6 +99,16 @@ enum {
   // everything — the host engines return partial results there too) from a
   // protocol wedge (no completions at all for the final stretch)
   OUT_LAST_PROG_C,
+  // fine-grained latency breakdown (cycles / counts)
+  OUT_B_C,           // phase-B classify time, work-bearing sweeps only
+  OUT_C_C,           // phase-C apply time, work-bearing sweeps only
+  OUT_NSWEEP_WORK,   // sweeps that processed >=1 completion
+  OUT_W0_WAKE_C,     // go release -> worker-0 round start
+  OUT_W0_WAKE_N,
+  OUT_W0_DET_C,      // worker-0 done publish -> server classify
+  OUT_W0_DET_N,
+  OUT_W0_SUB_LAST,   // scratch: server stamps w0 dispatch / w0 stamps done
+  OUT_W0_DONE_LAST,
   // abort diagnostics: per-worker server state, packed (see server_block's
   // abort dump); only written when the deadline watchdog fires
   OUT_DUMP, OUT_N = OUT_DUMP + RES_MAXP
@@ -370,6 +380,7 @@ struct ServerState {
   int dis_w[RES_MAXP];
   int ndis;
   unsigned long long sweep_c;  // loop-top clock, reused across the sweep
+  unsigned long long prof_b;   // phase-B/C profiling scratch (tid 0)
   long long k;
   int clock_;
   long long applied, rejected, max_stale;
@@ -520,6 +531,7 @@ __device__ void server_block(const ResidentArgs& a) {
 
     // ---- phase B: scalar classify (thread 0; builds the op lists) ----
     if (tid == 0) {
+      st.prof_b = realtime();
       st.nacc = st.nrej = st.nseg = 0;
       long long kv = st.k;
       long long snapn = (long long)a.out[OUT_SNAPN];
@@ -537,6 +549,11 @@ __device__ void server_block(const ResidentArgs& a) {
           continue;
         }
         if (st.done_snap[w] != st.round_no[w] || kv >= a.iters) continue;
+        if (w == 0) {
+          a.out[OUT_W0_DET_C] +=
+              realtime() - a.out[OUT_W0_DONE_LAST];
+          a.out[OUT_W0_DET_N] += 1;
+        }
         st.busy[w] = 0;
         st.finish_c[w] = st.sweep_c;
         const int staleness = st.clock_ - st.ts[w];
@@ -591,7 +608,12 @@ __device__ void server_block(const ResidentArgs& a) {
         st.nseg += 1;
       }
       st.k = kv;
-      if (st.nacc + st.nrej > 0) a.out[OUT_LAST_PROG_C] = st.sweep_c;
+      if (st.nacc + st.nrej > 0) {
+        a.out[OUT_LAST_PROG_C] = st.sweep_c;
+        a.out[OUT_B_C] += realtime() - st.prof_b;
+        a.out[OUT_NSWEEP_WORK] += 1;
+        st.prof_b = realtime();  // reused as phase-C start below
+      }
     }
     __syncthreads();
     // ONE acquire for all of this sweep's completions, by every thread
@@ -661,6 +683,7 @@ __device__ void server_block(const ResidentArgs& a) {
     // ---- phase D: scalar redispatch decisions (quorum gate + straggler
     // model; thread 0 only — the block is still finishing phase C) ----
     if (tid == 0) {
+      if (st.nacc + st.nrej > 0) a.out[OUT_C_C] += realtime() - st.prof_b;
       a.out[OUT_SRV_LOOPS] += 1;
       a.out[OUT_SRV_SWEEP_C] += realtime() - prof_t0;
       st.ndis = 0;
@@ -726,9 +749,11 @@ __device__ void server_block(const ResidentArgs& a) {
           st.round_no[w] += 1;
         }
         __threadfence();  // publish go_key/go_flags before the go stores
-        for (int u = 0; u < ndis; ++u)
+        for (int u = 0; u < ndis; ++u) {
+          if (st.dis_w[u] == 0) a.out[OUT_W0_SUB_LAST] = realtime();
           store_rel(&a.go_round[st.dis_w[u] * RES_CSTRIDE],
                     st.round_no[st.dis_w[u]]);
+        }
       }
     }
     if (tid == 0) a.out[OUT_SRV_DISPATCH_C] += realtime() - prof_t1;
@@ -827,6 +852,10 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
     __syncthreads();
     const unsigned long long grad_t0 =
         (tid == 0 && w == 0 && b == 0) ? realtime() : 0;
+    if (tid == 0 && w == 0 && b == 0) {
+      a.out[OUT_W0_WAKE_C] += grad_t0 - a.out[OUT_W0_SUB_LAST];
+      a.out[OUT_W0_WAKE_N] += 1;
+    }
     if (a.algo == 1 && (flags & 1u)) commit_pass(a, w, b, flags >> 1);
     worker_round<XT>(a, w, b, key, flags, lds_g);
     if (tid == 0 && w == 0 && b == 0) {
@@ -847,6 +876,7 @@ __global__ __launch_bounds__(RES_BLOCK) void resident_engine_kernel(
         __hip_atomic_store(&a.done_ctr[w * RES_CSTRIDE], 0u,
                            __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
+        if (w == 0) a.out[OUT_W0_DONE_LAST] = realtime();
         store_rel(&a.done_round[w * RES_CSTRIDE], r_local);
       }
     }
@@ -1035,6 +1065,13 @@ void register_resident_engine(py::module_& m) {
         out_h[OUT_LAST_PROG_C]
             ? ((double)out_h[OUT_LAST_PROG_C] - t0c) / cpm
             : -1.0;
+    r["srv_classify_ms"] = (double)out_h[OUT_B_C] / cpm;
+    r["srv_apply_ms"] = (double)out_h[OUT_C_C] / cpm;
+    r["srv_work_sweeps"] = (long long)out_h[OUT_NSWEEP_WORK];
+    r["w0_wake_ms"] = (double)out_h[OUT_W0_WAKE_C] / cpm;
+    r["w0_wake_n"] = (long long)out_h[OUT_W0_WAKE_N];
+    r["w0_detect_ms"] = (double)out_h[OUT_W0_DET_C] / cpm;
+    r["w0_detect_n"] = (long long)out_h[OUT_W0_DET_N];
     return r;
   });
 }

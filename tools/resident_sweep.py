@@ -40,7 +40,12 @@ def run(P, rows, d, dtype, iters, G, tag, rate=0.01, algo="asgd"):
           f"spin_ms/round={res['w0_spin_ms'] / rounds:.3f} "
           f"srv_sweep={res['srv_sweep_ms']:.0f} "
           f"srv_disp={res['srv_dispatch_ms']:.0f} el={el:.0f} "
-          f"rej={res['rejected']} stale={res['max_staleness']}",
+          f"rej={res['rejected']} stale={res['max_staleness']} "
+          f"clas={res['srv_classify_ms']:.1f} "
+          f"apply={res['srv_apply_ms']:.1f} "
+          f"wsw={res['srv_work_sweeps']} "
+          f"wake_us={res['w0_wake_ms'] / max(1, res['w0_wake_n']) * 1e3:.1f} "
+          f"det_us={res['w0_detect_ms'] / max(1, res['w0_detect_n']) * 1e3:.1f}",
           flush=True)
     del eng, X, y
     torch.cuda.empty_cache()
