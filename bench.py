@@ -45,13 +45,16 @@ MODELS = {
                           objective="logistic"),  # BASELINE config 1 names
                                                   # logistic regression
     # flagship: 32 async workers (the reference's fixed partitions=32,
-    # README.md) driven by the device-RESIDENT persistent-kernel engine —
-    # the whole bounded-staleness loop runs in one HIP kernel (~70k
-    # updates/s vs ~40k for the host-driven C++ event loop, which remains
-    # available via --engine native; --engine graph = 1-worker hipGraph)
+    # README.md) on the native C++ event loop with batched fused updates
+    # (87.2k updates/s measured; whole-chip 16.6us gradient rounds keep the
+    # quorum gate cheap). --engine resident runs the same config inside ONE
+    # persistent HIP kernel (77.2k: zero host API calls, but the static
+    # G-blocks-per-worker partition makes rounds ~190us and the reference's
+    # avail>=gate quorum turns that latency into idle — analysis in
+    # profiles/r02_resident_profile.md); --engine graph = 1-worker hipGraph
     "asgd-mnist8m": dict(rows=8_100_000, cols=784, rate=0.01, algo="asgd",
                          sync=False, dtype="bf16", sparse=False,
-                         engine="resident", workers=32),
+                         engine="native", workers=32),
     "asaga-rcv1": dict(rows=697_641, cols=47_236, rate=0.02, algo="asaga",
                        sync=False, dtype="fp32", sparse=True,
                        engine="native", workers=32),

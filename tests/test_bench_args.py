@@ -28,7 +28,7 @@ def test_default_model_is_flagship():
     assert a.model == "asgd-mnist8m"
     assert a.rows == 8_100_000 and a.cols == 784
     assert a.dtype == "bf16" and a.algo == "asgd"
-    assert a.engine == "resident" and a.preset_workers == 32
+    assert a.engine == "native" and a.preset_workers == 32
     assert a.objective == "lsq"
 
 
