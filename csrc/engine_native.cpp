@@ -57,6 +57,28 @@ void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
                           const float*, float*, float*, int*, int*, float*,
                           int*, const int*, int, long, uint64_t, uint32_t,
                           uint64_t, double, int, int, hipStream_t);
+void launch_grad_dense_flag(const void*, const float*, const float*,
+                            float*, float*, int*, const int*, long, int,
+                            uint64_t, uint32_t, uint64_t, double, int, int,
+                            hipStream_t, unsigned long long*,
+                            unsigned long long, unsigned int*);
+void launch_saga_grad_dense_flag(const void*, const float*, const float*,
+                                 float*, float*, float*, int*, int*, float*,
+                                 int*, const int*, int, long, int, uint64_t,
+                                 uint32_t, uint64_t, double, int, int,
+                                 hipStream_t, unsigned long long*,
+                                 unsigned long long, unsigned int*);
+void launch_grad_csr_flag(const int*, const int*, const void*, const float*,
+                          const float*, float*, int*, const int*, long,
+                          uint64_t, uint32_t, uint64_t, double, int, int,
+                          hipStream_t, unsigned long long*,
+                          unsigned long long, unsigned int*);
+void launch_saga_grad_csr_flag(const int*, const int*, const void*,
+                               const float*, const float*, float*, float*,
+                               int*, int*, float*, int*, const int*, int,
+                               long, uint64_t, uint32_t, uint64_t, double,
+                               int, int, hipStream_t, unsigned long long*,
+                               unsigned long long, unsigned int*);
 void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
 void launch_saga_update(float*, const float*, float*, float, float, float,
                         int, hipStream_t);
@@ -126,8 +148,19 @@ struct WorkerBuf {
   int x_is_bf16 = 0;
   // runtime
   hipStream_t stream = nullptr;
-  hipEvent_t done = nullptr;
+  // event-free completion: the grad kernel's last block release-stores
+  // round_serial into pinned host memory (publish_done in kernels.hip);
+  // the host event loop polls plain memory instead of hipEventQuery
+  volatile unsigned long long* done_flag = nullptr;  // pinned host, 1 line
+  unsigned int* done_arr = nullptr;                  // device arrival ctr
+  unsigned long long round_serial = 0;
   bool busy = false;
+  // a round is actually executing on the GPU. Distinct from busy: a
+  // straggler-DELAYED worker is busy (occupied) but not in flight, and its
+  // done flag still holds the previous round's serial — polling it would
+  // double-book that round (latent in the event-based loop too, where the
+  // completed event also stayed signaled; made explicit here)
+  bool in_flight = false;
   int ts = 0;        // arrival clock at dispatch
   long k_submit = 0; // round index at dispatch
   bool pending_commit = false;
@@ -188,6 +221,8 @@ struct NativeEngine {
   double run_t0 = 0;
   // batched-update machinery
   float** g_tab_dev = nullptr;     // device table: worker id -> g pointer
+  volatile unsigned long long* flags_host = nullptr;
+  unsigned int* arr_dev = nullptr;
   float** wbuf_tab_dev = nullptr;  // device table: worker id -> wbuf pointer
   double inv_batch = 0, inv_N = 0;
 
@@ -257,37 +292,43 @@ struct NativeEngine {
     }
     if (cfg.algo == 1) {
       if (wk.sparse)
-        launch_saga_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
-                             (const void*)wk.values, (const float*)wk.y,
-                             (const float*)wk.wbuf, (float*)wk.alpha,
-                             (float*)wk.g, (int*)wk.ctr, (int*)wk.idx_out,
-                             (float*)wk.e_out, (int*)(wk.ctr + 4), nullptr,
-                             0, wk.n_rows, cfg.seed, (uint32_t)round_key,
-                             (uint64_t)wk.row_start, cfg.rate, cfg.objective,
-                             wk.x_is_bf16, wk.stream);
+        launch_saga_grad_csr_flag(
+            (const int*)wk.indptr, (const int*)wk.indices,
+            (const void*)wk.values, (const float*)wk.y,
+            (const float*)wk.wbuf, (float*)wk.alpha, (float*)wk.g,
+            (int*)wk.ctr, (int*)wk.idx_out, (float*)wk.e_out,
+            (int*)(wk.ctr + 4), nullptr, 0, wk.n_rows, cfg.seed,
+            (uint32_t)round_key, (uint64_t)wk.row_start, cfg.rate,
+            cfg.objective, wk.x_is_bf16, wk.stream,
+            (unsigned long long*)wk.done_flag, wk.round_serial,
+            wk.done_arr);
       else
-        launch_saga_grad_dense((const void*)wk.X, (const float*)wk.y,
-                               (const float*)wk.wbuf, (float*)wk.alpha,
-                               (float*)wk.g, nullptr, (int*)wk.ctr,
-                               (int*)wk.idx_out, (float*)wk.e_out,
-                               (int*)(wk.ctr + 4), nullptr, 0, wk.n_rows,
-                               cfg.d, cfg.seed, (uint32_t)round_key,
-                               (uint64_t)wk.row_start, cfg.rate,
-                               cfg.objective, wk.x_is_bf16, wk.stream);
+        launch_saga_grad_dense_flag(
+            (const void*)wk.X, (const float*)wk.y, (const float*)wk.wbuf,
+            (float*)wk.alpha, (float*)wk.g, nullptr, (int*)wk.ctr,
+            (int*)wk.idx_out, (float*)wk.e_out, (int*)(wk.ctr + 4), nullptr,
+            0, wk.n_rows, cfg.d, cfg.seed, (uint32_t)round_key,
+            (uint64_t)wk.row_start, cfg.rate, cfg.objective, wk.x_is_bf16,
+            wk.stream, (unsigned long long*)wk.done_flag, wk.round_serial,
+            wk.done_arr);
     } else {
       if (wk.sparse)
-        launch_grad_csr((const int*)wk.indptr, (const int*)wk.indices,
-                        (const void*)wk.values, (const float*)wk.y,
-                        (const float*)wk.wbuf, (float*)wk.g, (int*)wk.ctr,
-                        nullptr, wk.n_rows, cfg.seed, (uint32_t)round_key,
-                        (uint64_t)wk.row_start, cfg.rate, cfg.objective,
-                        wk.x_is_bf16, wk.stream);
+        launch_grad_csr_flag(
+            (const int*)wk.indptr, (const int*)wk.indices,
+            (const void*)wk.values, (const float*)wk.y,
+            (const float*)wk.wbuf, (float*)wk.g, (int*)wk.ctr, nullptr,
+            wk.n_rows, cfg.seed, (uint32_t)round_key,
+            (uint64_t)wk.row_start, cfg.rate, cfg.objective, wk.x_is_bf16,
+            wk.stream, (unsigned long long*)wk.done_flag, wk.round_serial,
+            wk.done_arr);
       else
-        launch_grad_dense((const void*)wk.X, (const float*)wk.y,
-                          (const float*)wk.wbuf, (float*)wk.g, nullptr,
-                          (int*)wk.ctr, nullptr, wk.n_rows, cfg.d, cfg.seed,
-                          (uint32_t)round_key, (uint64_t)wk.row_start,
-                          cfg.rate, cfg.objective, wk.x_is_bf16, wk.stream);
+        launch_grad_dense_flag(
+            (const void*)wk.X, (const float*)wk.y, (const float*)wk.wbuf,
+            (float*)wk.g, nullptr, (int*)wk.ctr, nullptr, wk.n_rows, cfg.d,
+            cfg.seed, (uint32_t)round_key, (uint64_t)wk.row_start, cfg.rate,
+            cfg.objective, wk.x_is_bf16, wk.stream,
+            (unsigned long long*)wk.done_flag, wk.round_serial,
+            wk.done_arr);
     }
     HIP_CHECK(hipGetLastError());
   }
@@ -325,11 +366,15 @@ struct NativeEngine {
       HIP_CHECK(hipMemcpyAsync((void*)wk.wbuf, (const void*)w,
                                (size_t)cfg.d * 4, hipMemcpyDeviceToDevice,
                                wk.stream));
+    if (wk.in_flight)
+      throw std::runtime_error(
+          "native engine invariant: dispatch while a round is in flight "
+          "(worker " + std::to_string(wid) + ")");
+    wk.round_serial += 1;  // the grad kernel's last block publishes this
+    wk.in_flight = true;
     launch_grad(wk, wk.k_submit + 1);  // reference seed+k+1
-    HIP_CHECK(hipEventRecord(wk.done, wk.stream));
-    // poll gating: querying every busy event each pass costs ~1-2 us per
-    // query; skip until ~60% of the EWMA round time has elapsed
-    wk.poll_after = t_now + 0.3 * ewma_round_s;
+    // no hipEventRecord: completion = *done_flag == round_serial (plain
+    // pinned-memory read, ~0 host cost vs ~1-2 us per hipEventQuery)
   }
 
   void dispatch(int wid, double t_now) { dispatch_impl(wid, t_now, true); }
@@ -365,6 +410,7 @@ struct NativeEngine {
   bool book_completion(int wid, double t_now) {
     WorkerBuf& wk = ws[wid];
     wk.busy = false;
+    wk.in_flight = false;
     wk.finish_t = t_now;
     wk.tasks += 1;
     const double rt = t_now - wk.submit_t;
@@ -471,9 +517,26 @@ struct NativeEngine {
     HIP_CHECK(hipStreamCreateWithFlags(&sstream, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&update_ev, hipEventDisableTiming));
     HIP_CHECK(hipEventRecord(update_ev, sstream));
+    // completion flags: one pinned-host 128-B line per worker (the grad
+    // kernel writes it with a system-scope release; default hipHostMalloc
+    // memory is fine-grained coherent on MI355X) + one padded device
+    // arrival counter per worker
+    HIP_CHECK(hipHostMalloc((void**)&flags_host,
+                            (size_t)cfg.P * 16 * sizeof(unsigned long long),
+                            hipHostMallocDefault));
+    memset((void*)flags_host, 0,
+           (size_t)cfg.P * 16 * sizeof(unsigned long long));
+    HIP_CHECK(hipMalloc((void**)&arr_dev,
+                        (size_t)cfg.P * 32 * sizeof(unsigned int)));
+    HIP_CHECK(hipMemset(arr_dev, 0,
+                        (size_t)cfg.P * 32 * sizeof(unsigned int)));
+    for (int i = 0; i < cfg.P; ++i) {
+      ws[i].done_flag = flags_host + (size_t)i * 16;
+      ws[i].done_arr = arr_dev + (size_t)i * 32;
+      ws[i].round_serial = 0;
+    }
     for (auto& wk : ws) {
       HIP_CHECK(hipStreamCreateWithFlags(&wk.stream, hipStreamNonBlocking));
-      HIP_CHECK(hipEventCreateWithFlags(&wk.done, hipEventDisableTiming));
     }
     inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
     inv_N = 1.0 / (double)cfg.N;
@@ -546,17 +609,14 @@ struct NativeEngine {
         // permanently false after dispatch() seeded finish_t = t_now on the
         // first round (BENCH_r01 30-min 0%-GPU hang). !busy alone gates
         // polling; busy is only set between dispatch and completion.
-        if (!wk.busy) continue;
-        if (wk.done) {
-          if (t_now < wk.poll_after) continue;
-          const hipError_t q = hipEventQuery(wk.done);
-          if (q == hipSuccess) {
-            const double tc = now_s();
-            if (book_completion(i, tc)) append_accepted(i, tc);
-            any = true;
-          } else if (q != hipErrorNotReady) {
-            HIP_CHECK(q);
-          }
+        if (!wk.busy || !wk.in_flight) continue;
+        // event-free completion: the grad kernel's last block published
+        // round_serial to this pinned line (system-scope release store);
+        // a plain volatile read replaces hipEventQuery
+        if (*wk.done_flag == wk.round_serial) {
+          const double tc = now_s();
+          if (book_completion(i, tc)) append_accepted(i, tc);
+          any = true;
         }
       }
       if (any) {
@@ -587,9 +647,14 @@ struct NativeEngine {
     out.snap_ms = snap_ms;
     for (auto& wk : ws) out.waits.push_back(wk.waiting_ms);
     for (auto& wk : ws) {
-      HIP_CHECK(hipEventDestroy(wk.done));
       HIP_CHECK(hipStreamDestroy(wk.stream));
+      wk.done_flag = nullptr;
+      wk.done_arr = nullptr;
     }
+    HIP_CHECK(hipHostFree((void*)flags_host));
+    flags_host = nullptr;
+    HIP_CHECK(hipFree(arr_dev));
+    arr_dev = nullptr;
     HIP_CHECK(hipEventDestroy(update_ev));
     HIP_CHECK(hipStreamDestroy(sstream));
     HIP_CHECK(hipFree(g_tab_dev));

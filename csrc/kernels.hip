@@ -50,6 +50,29 @@
 
 // ---------------------------------------------------------------- helpers
 
+// Host-visible completion flag: the LAST arriving block publishes
+// `done_val` to fine-grained pinned host memory with system-scope release,
+// replacing the native engine's hipEventRecord + hipEventQuery pair
+// (~2-4 us of host API per round). All prior global writes (g atomics,
+// SAGA staging) are ordered before the flag by the fence chain.
+__device__ __forceinline__ void publish_done(
+    unsigned long long* done_flag, unsigned long long done_val,
+    unsigned int* done_arr) {
+  if (!done_flag) return;  // uniform per launch: no divergence
+  __threadfence();
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned int arrived = __hip_atomic_fetch_add(
+        done_arr, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived + 1 == gridDim.x) {
+      __hip_atomic_store(done_arr, 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(done_flag, done_val, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+  }
+}
+
 __device__ __forceinline__ float link_residual(float z, float yv, int obj) {
   if (obj == 1) return 1.0f / (1.0f + __expf(-z)) - yv;  // logistic
   return z - yv;                                         // lsq
@@ -98,7 +121,8 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     float* __restrict__ e_out, int* __restrict__ pos_ctr,
     const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
-    int take_all, int objective) {
+    int take_all, int objective, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;  // graph mode: round = k+1
   constexpr int NSUB = WAVE / LPR;
   constexpr int NSLAB = WAVES_PER_BLOCK * NSUB;
@@ -209,6 +233,7 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
+  publish_done(done_flag, done_val, done_arr);
 }
 
 // ------------------------------------------------- K1 pipelined (queue)
@@ -470,7 +495,8 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     float* __restrict__ e_out, int* __restrict__ pos_ctr,
     const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
-    int take_all, int objective) {
+    int take_all, int objective, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NW = PBLOCK / WAVE;
   constexpr int RPB = NW * ROWS_PER_WAVE;
@@ -638,6 +664,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     }
   }
   if (lane == 0 && local_count) atomicAdd(n_out, local_count);
+  publish_done(done_flag, done_val, done_arr);
 }
 
 // Sums the per-block partial slabs into g (layout g_part[j][G], contiguous
@@ -684,7 +711,9 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     int* __restrict__ idx_out, float* __restrict__ e_out,
     int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
-    uint64_t row_start, uint32_t threshold, int take_all, int objective) {
+    uint64_t row_start, uint32_t threshold, int take_all, int objective,
+    unsigned long long* done_flag, unsigned long long done_val,
+    unsigned int* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NSUB = WAVE / LPR;
   const int wave = threadIdx.x >> 6;
@@ -745,6 +774,7 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
+  publish_done(done_flag, done_val, done_arr);
 }
 
 // ---------------------------------------------------------------- K5/K6
@@ -990,7 +1020,10 @@ static void launch_dense(const XT* X, const float* y, const float* w,
                          int* pos_ctr, const int* k_dev, int commit_now,
                          long n_rows, int d, uint64_t seed, uint32_t round_k,
                          uint64_t row_start, double rate, int objective,
-                         hipStream_t stream) {
+                         hipStream_t stream,
+                         unsigned long long* done_flag = nullptr,
+                         unsigned long long done_val = 0,
+                         unsigned int* done_arr = nullptr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -1008,7 +1041,8 @@ static void launch_dense(const XT* X, const float* y, const float* w,
                        dim3(grid), dim3(256), smem, stream, X, y, w, g_out,  \
                        g_part, n_out, alpha, idx_out, e_out, pos_ctr,        \
                        k_dev, commit_now, n_rows, d, seed, round_k,          \
-                       row_start, thr, take_all, objective)
+                       row_start, thr, take_all, objective, done_flag,       \
+                       done_val, done_arr)
 #define LAUNCH_PIPE(IT)                                                      \
     do {                                                                     \
       if (depth == 1) LAUNCH_PIPE_D(IT, 1);                                  \
@@ -1038,7 +1072,7 @@ static void launch_dense(const XT* X, const float* y, const float* w,
                      dim3(BLOCK), smem, stream, X, y, w, g_out, g_part,      \
                      n_out, alpha, idx_out, e_out, pos_ctr, k_dev,           \
                      commit_now, n_rows, d, seed, round_k, row_start, thr,   \
-                     take_all, objective)
+                     take_all, objective, done_flag, done_val, done_arr)
   if (lpr == 16) DISPATCH_LPR(16);
   else if (lpr == 32) DISPATCH_LPR(32);
   else DISPATCH_LPR(64);
@@ -1144,6 +1178,90 @@ void launch_saga_grad_dense(const void* X, const float* y, const float* w,
                               rate, objective, stream);
 }
 
+void launch_grad_dense_flag(
+    const void* X, const float* y, const float* w, float* g_out,
+    float* g_part, int* n_out, const int* k_dev, long n_rows, int d,
+    uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
+    int objective, int x_is_bf16, hipStream_t stream,
+    unsigned long long* done_flag, unsigned long long done_val,
+    unsigned int* done_arr) {
+  if (x_is_bf16)
+    launch_dense<__hip_bfloat16, false>(
+        (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, nullptr,
+        nullptr, nullptr, nullptr, k_dev, 0, n_rows, d, seed, round_k,
+        row_start, rate, objective, stream, done_flag, done_val, done_arr);
+  else
+    launch_dense<float, false>((const float*)X, y, w, g_out, g_part, n_out,
+                               nullptr, nullptr, nullptr, nullptr, k_dev, 0,
+                               n_rows, d, seed, round_k, row_start, rate,
+                               objective, stream, done_flag, done_val,
+                               done_arr);
+}
+
+void launch_saga_grad_dense_flag(
+    const void* X, const float* y, const float* w, float* alpha,
+    float* g_out, float* g_part, int* n_out, int* idx_out, float* e_out,
+    int* pos_ctr, const int* k_dev, int commit_now, long n_rows, int d,
+    uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
+    int objective, int x_is_bf16, hipStream_t stream,
+    unsigned long long* done_flag, unsigned long long done_val,
+    unsigned int* done_arr) {
+  if (x_is_bf16)
+    launch_dense<__hip_bfloat16, true>(
+        (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, alpha, idx_out,
+        e_out, pos_ctr, k_dev, commit_now, n_rows, d, seed, round_k,
+        row_start, rate, objective, stream, done_flag, done_val, done_arr);
+  else
+    launch_dense<float, true>((const float*)X, y, w, g_out, g_part, n_out,
+                              alpha, idx_out, e_out, pos_ctr, k_dev,
+                              commit_now, n_rows, d, seed, round_k, row_start,
+                              rate, objective, stream, done_flag, done_val,
+                              done_arr);
+}
+
+// event-free CSR launchers keep their original names as null-flag wrappers
+void launch_grad_csr_flag(
+    const int* indptr, const int* indices, const void* values,
+    const float* y, const float* w, float* g_out, int* n_out,
+    const int* k_dev, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, double rate, int objective, int v_is_bf16,
+    hipStream_t stream, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr);
+void launch_saga_grad_csr_flag(
+    const int* indptr, const int* indices, const void* values,
+    const float* y, const float* w, float* alpha, float* g_out, int* n_out,
+    int* idx_out, float* e_out, int* pos_ctr, const int* k_dev,
+    int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, double rate, int objective, int v_is_bf16,
+    hipStream_t stream, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr);
+
+void launch_grad_csr(const int* indptr, const int* indices,
+                     const void* values, const float* y, const float* w,
+                     float* g_out, int* n_out, const int* k_dev, long n_rows,
+                     uint64_t seed, uint32_t round_k, uint64_t row_start,
+                     double rate, int objective, int v_is_bf16,
+                     hipStream_t stream) {
+  launch_grad_csr_flag(indptr, indices, values, y, w, g_out, n_out, k_dev,
+                       n_rows, seed, round_k, row_start, rate, objective,
+                       v_is_bf16, stream, nullptr, 0, nullptr);
+}
+
+void launch_saga_grad_csr(const int* indptr, const int* indices,
+                          const void* values, const float* y, const float* w,
+                          float* alpha, float* g_out, int* n_out,
+                          int* idx_out, float* e_out, int* pos_ctr,
+                          const int* k_dev, int commit_now, long n_rows,
+                          uint64_t seed, uint32_t round_k, uint64_t row_start,
+                          double rate, int objective, int v_is_bf16,
+                          hipStream_t stream) {
+  launch_saga_grad_csr_flag(indptr, indices, values, y, w, alpha, g_out,
+                            n_out, idx_out, e_out, pos_ctr, k_dev,
+                            commit_now, n_rows, seed, round_k, row_start,
+                            rate, objective, v_is_bf16, stream, nullptr, 0,
+                            nullptr);
+}
+
 void launch_reduce_partials(const float* g_part, float* g_out, int d, int G,
                             int splits, hipStream_t stream) {
   const int njc = (d + BLOCK - 1) / BLOCK;
@@ -1157,11 +1275,13 @@ static inline int csr_lpr() {
   return (v == 8 || v == 16 || v == 32 || v == 64) ? v : 32;
 }
 
-void launch_grad_csr(const int* indptr, const int* indices, const void* values,
-                     const float* y, const float* w, float* g_out, int* n_out,
-                     const int* k_dev, long n_rows, uint64_t seed,
-                     uint32_t round_k, uint64_t row_start, double rate,
-                     int objective, int v_is_bf16, hipStream_t stream) {
+void launch_grad_csr_flag(
+    const int* indptr, const int* indices, const void* values,
+    const float* y, const float* w, float* g_out, int* n_out,
+    const int* k_dev, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, double rate, int objective, int v_is_bf16,
+    hipStream_t stream, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -1171,7 +1291,7 @@ void launch_grad_csr(const int* indptr, const int* indices, const void* values,
                      dim3(BLOCK), 0, stream, indptr, indices, (CAST)values,  \
                      y, w, g_out, n_out, nullptr, nullptr, nullptr, nullptr, \
                      k_dev, 0, n_rows, seed, round_k, row_start, thr,        \
-                     take_all, objective)
+                     take_all, objective, done_flag, done_val, done_arr)
 #define CSR_DISPATCH(VT, CAST)                                               \
   do {                                                                       \
     if (lpr == 8) CSR_LAUNCH(VT, CAST, 8);                                   \
@@ -1185,14 +1305,14 @@ void launch_grad_csr(const int* indptr, const int* indices, const void* values,
 #undef CSR_LAUNCH
 }
 
-void launch_saga_grad_csr(const int* indptr, const int* indices,
-                          const void* values, const float* y, const float* w,
-                          float* alpha, float* g_out, int* n_out,
-                          int* idx_out, float* e_out, int* pos_ctr,
-                          const int* k_dev, int commit_now, long n_rows,
-                          uint64_t seed, uint32_t round_k, uint64_t row_start,
-                          double rate, int objective, int v_is_bf16,
-                          hipStream_t stream) {
+void launch_saga_grad_csr_flag(
+    const int* indptr, const int* indices, const void* values,
+    const float* y, const float* w, float* alpha, float* g_out, int* n_out,
+    int* idx_out, float* e_out, int* pos_ctr, const int* k_dev,
+    int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, double rate, int objective, int v_is_bf16,
+    hipStream_t stream, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned int* done_arr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -1202,7 +1322,8 @@ void launch_saga_grad_csr(const int* indptr, const int* indices,
                      dim3(BLOCK), 0, stream, indptr, indices, (CAST)values,  \
                      y, w, g_out, n_out, alpha, idx_out, e_out, pos_ctr,     \
                      k_dev, commit_now, n_rows, seed, round_k, row_start,    \
-                     thr, take_all, objective)
+                     thr, take_all, objective, done_flag, done_val,          \
+                     done_arr)
 #define CSR_SDISPATCH(VT, CAST)                                              \
   do {                                                                       \
     if (lpr == 8) CSR_SLAUNCH(VT, CAST, 8);                                  \
