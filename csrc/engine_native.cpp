@@ -61,24 +61,24 @@ void launch_grad_dense_flag(const void*, const float*, const float*,
                             float*, float*, int*, const int*, long, int,
                             uint64_t, uint32_t, uint64_t, double, int, int,
                             hipStream_t, unsigned long long*,
-                            unsigned long long, unsigned int*);
+                            unsigned long long, unsigned long long*);
 void launch_saga_grad_dense_flag(const void*, const float*, const float*,
                                  float*, float*, float*, int*, int*, float*,
                                  int*, const int*, int, long, int, uint64_t,
                                  uint32_t, uint64_t, double, int, int,
                                  hipStream_t, unsigned long long*,
-                                 unsigned long long, unsigned int*);
+                                 unsigned long long, unsigned long long*);
 void launch_grad_csr_flag(const int*, const int*, const void*, const float*,
                           const float*, float*, int*, const int*, long,
                           uint64_t, uint32_t, uint64_t, double, int, int,
                           hipStream_t, unsigned long long*,
-                          unsigned long long, unsigned int*);
+                          unsigned long long, unsigned long long*);
 void launch_saga_grad_csr_flag(const int*, const int*, const void*,
                                const float*, const float*, float*, float*,
                                int*, int*, float*, int*, const int*, int,
                                long, uint64_t, uint32_t, uint64_t, double,
                                int, int, hipStream_t, unsigned long long*,
-                               unsigned long long, unsigned int*);
+                               unsigned long long, unsigned long long*);
 void launch_sgd_update(float*, const float*, float, float, int, hipStream_t);
 void launch_saga_update(float*, const float*, float*, float, float, float,
                         int, hipStream_t);
@@ -152,7 +152,7 @@ struct WorkerBuf {
   // round_serial into pinned host memory (publish_done in kernels.hip);
   // the host event loop polls plain memory instead of hipEventQuery
   volatile unsigned long long* done_flag = nullptr;  // pinned host, 1 line
-  unsigned int* done_arr = nullptr;                  // device arrival ctr
+  unsigned long long* done_arr = nullptr;  // device arrival ctr (monotonic)
   unsigned long long round_serial = 0;
   bool busy = false;
   // a round is actually executing on the GPU. Distinct from busy: a
@@ -222,7 +222,7 @@ struct NativeEngine {
   // batched-update machinery
   float** g_tab_dev = nullptr;     // device table: worker id -> g pointer
   volatile unsigned long long* flags_host = nullptr;
-  unsigned int* arr_dev = nullptr;
+  unsigned long long* arr_dev = nullptr;
   float** wbuf_tab_dev = nullptr;  // device table: worker id -> wbuf pointer
   double inv_batch = 0, inv_N = 0;
 
@@ -527,12 +527,12 @@ struct NativeEngine {
     memset((void*)flags_host, 0,
            (size_t)cfg.P * 16 * sizeof(unsigned long long));
     HIP_CHECK(hipMalloc((void**)&arr_dev,
-                        (size_t)cfg.P * 32 * sizeof(unsigned int)));
+                        (size_t)cfg.P * 16 * sizeof(unsigned long long)));
     HIP_CHECK(hipMemset(arr_dev, 0,
-                        (size_t)cfg.P * 32 * sizeof(unsigned int)));
+                        (size_t)cfg.P * 16 * sizeof(unsigned long long)));
     for (int i = 0; i < cfg.P; ++i) {
       ws[i].done_flag = flags_host + (size_t)i * 16;
-      ws[i].done_arr = arr_dev + (size_t)i * 32;
+      ws[i].done_arr = arr_dev + (size_t)i * 16;
       ws[i].round_serial = 0;
     }
     for (auto& wk : ws) {

@@ -57,17 +57,23 @@
 // SAGA staging) are ordered before the flag by the fence chain.
 __device__ __forceinline__ void publish_done(
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned int* done_arr) {
+    unsigned long long* done_arr) {
   if (!done_flag) return;  // uniform per launch: no divergence
-  __threadfence();
+  __threadfence();  // agent-scope release of this block's g/staging writes
   __syncthreads();
   if (threadIdx.x == 0) {
-    const unsigned int arrived = __hip_atomic_fetch_add(
-        done_arr, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
-    if (arrived + 1 == gridDim.x) {
-      __hip_atomic_store(done_arr, 0u, __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      __hip_atomic_store(done_flag, done_val, __ATOMIC_RELEASE,
+    // monotonic arrival counter (no per-round reset => no reset/flag
+    // ordering race): round `done_val` is complete when the counter
+    // reaches done_val * gridDim.x (grid size is fixed per worker).
+    const unsigned long long arrived = __hip_atomic_fetch_add(
+        done_arr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived + 1 == done_val * gridDim.x) {
+      // RELAXED system store: a release here compiles to a full L2
+      // writeback (buffer_wbl2) per round — measured 4x flagship loss.
+      // The host only uses the VALUE as a signal; device-side readers of
+      // g are ordered by the threadfence above, which precedes the
+      // arrival RMW that gates this store.
+      __hip_atomic_store(done_flag, done_val, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_SYSTEM);
     }
   }
@@ -122,7 +128,7 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
     int take_all, int objective, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr) {
+    unsigned long long done_val, unsigned long long* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;  // graph mode: round = k+1
   constexpr int NSUB = WAVE / LPR;
   constexpr int NSLAB = WAVES_PER_BLOCK * NSUB;
@@ -496,7 +502,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
     int take_all, int objective, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr) {
+    unsigned long long done_val, unsigned long long* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NW = PBLOCK / WAVE;
   constexpr int RPB = NW * ROWS_PER_WAVE;
@@ -713,7 +719,7 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, uint32_t threshold, int take_all, int objective,
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned int* done_arr) {
+    unsigned long long* done_arr) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NSUB = WAVE / LPR;
   const int wave = threadIdx.x >> 6;
@@ -1023,7 +1029,7 @@ static void launch_dense(const XT* X, const float* y, const float* w,
                          hipStream_t stream,
                          unsigned long long* done_flag = nullptr,
                          unsigned long long done_val = 0,
-                         unsigned int* done_arr = nullptr) {
+                         unsigned long long* done_arr = nullptr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -1184,7 +1190,7 @@ void launch_grad_dense_flag(
     uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
     int objective, int x_is_bf16, hipStream_t stream,
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned int* done_arr) {
+    unsigned long long* done_arr) {
   if (x_is_bf16)
     launch_dense<__hip_bfloat16, false>(
         (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, nullptr,
@@ -1205,7 +1211,7 @@ void launch_saga_grad_dense_flag(
     uint64_t seed, uint32_t round_k, uint64_t row_start, double rate,
     int objective, int x_is_bf16, hipStream_t stream,
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned int* done_arr) {
+    unsigned long long* done_arr) {
   if (x_is_bf16)
     launch_dense<__hip_bfloat16, true>(
         (const __hip_bfloat16*)X, y, w, g_out, g_part, n_out, alpha, idx_out,
@@ -1226,7 +1232,7 @@ void launch_grad_csr_flag(
     const int* k_dev, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, double rate, int objective, int v_is_bf16,
     hipStream_t stream, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr);
+    unsigned long long done_val, unsigned long long* done_arr);
 void launch_saga_grad_csr_flag(
     const int* indptr, const int* indices, const void* values,
     const float* y, const float* w, float* alpha, float* g_out, int* n_out,
@@ -1234,7 +1240,7 @@ void launch_saga_grad_csr_flag(
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, double rate, int objective, int v_is_bf16,
     hipStream_t stream, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr);
+    unsigned long long done_val, unsigned long long* done_arr);
 
 void launch_grad_csr(const int* indptr, const int* indices,
                      const void* values, const float* y, const float* w,
@@ -1281,7 +1287,7 @@ void launch_grad_csr_flag(
     const int* k_dev, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, double rate, int objective, int v_is_bf16,
     hipStream_t stream, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr) {
+    unsigned long long done_val, unsigned long long* done_arr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
@@ -1312,7 +1318,7 @@ void launch_saga_grad_csr_flag(
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, double rate, int objective, int v_is_bf16,
     hipStream_t stream, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned int* done_arr) {
+    unsigned long long done_val, unsigned long long* done_arr) {
   const uint32_t thr = philox_threshold(rate);
   const int take_all = rate >= 1.0;
   const int grid = grad_grid(n_rows);
