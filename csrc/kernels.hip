@@ -59,23 +59,24 @@ __device__ __forceinline__ void publish_done(
     unsigned long long* done_flag, unsigned long long done_val,
     unsigned long long* done_arr) {
   if (!done_flag) return;  // uniform per launch: no divergence
-  __threadfence();  // agent-scope release of this block's g/staging writes
+  // NO fences here — fence instructions at agent scope on a multi-XCD
+  // chip emit per-block L2 writeback/invalidate (buffer_wbl2/inv), which
+  // poisoned every concurrent kernel's L2 (measured: grad 16.6 -> 117 us
+  // with a __threadfence + acq_rel arrival). The gradient's cross-kernel
+  // outputs (g, n_out) are written with ATOMICS, which execute at the
+  // agent coherence point; all that is needed before arrival is to DRAIN
+  // this thread's outstanding vmem ops, then count with a relaxed RMW.
+  __builtin_amdgcn_s_waitcnt(0);
   __syncthreads();
   if (threadIdx.x == 0) {
     // monotonic arrival counter (no per-round reset => no reset/flag
     // ordering race): round `done_val` is complete when the counter
     // reaches done_val * gridDim.x (grid size is fixed per worker).
     const unsigned long long arrived = __hip_atomic_fetch_add(
-        done_arr, 1ull, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
-    if (arrived + 1 == done_val * gridDim.x) {
-      // RELAXED system store: a release here compiles to a full L2
-      // writeback (buffer_wbl2) per round — measured 4x flagship loss.
-      // The host only uses the VALUE as a signal; device-side readers of
-      // g are ordered by the threadfence above, which precedes the
-      // arrival RMW that gates this store.
+        done_arr, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (arrived + 1 == done_val * gridDim.x)
       __hip_atomic_store(done_flag, done_val, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_SYSTEM);
-    }
   }
 }
 
