@@ -29,6 +29,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include "grad_wave.h"
 #include "multi_update.h"
 
 #include <algorithm>
@@ -57,6 +58,9 @@ void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
                           const float*, float*, float*, int*, int*, float*,
                           int*, const int*, int, long, uint64_t, uint32_t,
                           uint64_t, double, int, int, hipStream_t);
+int query_grad_grid(long);
+void launch_grad_dense_wave(const void*, const void*, long, int, uint64_t,
+                            double, int, int, hipStream_t);
 void launch_grad_dense_flag(const void*, const float*, const float*,
                             float*, float*, int*, const int*, long, int,
                             uint64_t, uint32_t, uint64_t, double, int, int,
@@ -223,6 +227,12 @@ struct NativeEngine {
   float** g_tab_dev = nullptr;     // device table: worker id -> g pointer
   volatile unsigned long long* flags_host = nullptr;
   unsigned long long* arr_dev = nullptr;
+  // wave dispatch (dense ASGD): per-worker invariant table + stream
+  GradWaveSlot* slots_dev = nullptr;
+  hipStream_t wstream = nullptr;
+  int wave_bper = 0;
+  long wave_max_rows = 0;
+  bool wave_ok = false;
   float** wbuf_tab_dev = nullptr;  // device table: worker id -> wbuf pointer
   double inv_batch = 0, inv_N = 0;
 
@@ -378,6 +388,50 @@ struct NativeEngine {
   }
 
   void dispatch(int wid, double t_now) { dispatch_impl(wid, t_now, true); }
+
+  // Wave-path bookkeeping: everything dispatch_impl does EXCEPT the grad
+  // launch (the whole quorum wave launches as ONE grad_dense_wave_kernel).
+  // ASGD-only: no SAGA commit to chain.
+  void dispatch_book(int wid, double t_now) {
+    WorkerBuf& wk = ws[wid];
+    if (wk.finish_t == 0) wk.finish_t = t_now;  // first dispatch: no wait
+    wk.waiting_ms += (t_now - wk.finish_t) * 1000.0;
+    wk.submit_t = t_now;
+    wk.busy = true;
+    wk.ts = clock;
+    wk.k_submit = k;
+    if (wk.g_dirty) {  // rejected round left sums; zero on the wave stream
+      HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wstream));
+      wk.g_dirty = false;
+    }
+    if (wk.in_flight)
+      throw std::runtime_error(
+          "native engine invariant: dispatch while a round is in flight "
+          "(worker " + std::to_string(wid) + ")");
+    wk.round_serial += 1;
+    wk.in_flight = true;
+  }
+
+  void dispatch_wave(const std::vector<int>& ready, double t_now) {
+    HIP_CHECK(hipStreamWaitEvent(wstream, update_ev, 0));
+    GradWaveCmd cmd;
+    cmd.n = 0;
+    cmd.bper = wave_bper;
+    for (int wid : ready) {
+      dispatch_book(wid, t_now);
+      WorkerBuf& wk = ws[wid];
+      cmd.wid[cmd.n] = wid;
+      cmd.round_k[cmd.n] = (unsigned int)(wk.k_submit + 1);
+      cmd.done_val[cmd.n] = wk.round_serial;
+      cmd.n += 1;
+    }
+    if (cmd.n > 0) {
+      launch_grad_dense_wave(slots_dev, &cmd, wave_max_rows, cfg.d,
+                             cfg.seed, cfg.rate, cfg.objective,
+                             ws[0].x_is_bf16, wstream);
+      HIP_CHECK(hipGetLastError());
+    }
+  }
 
   // Pop every pending worker that may dispatch now: stragglers move to the
   // delayed queue with a due time, the rest land in ``ready`` (dispatched
@@ -538,6 +592,33 @@ struct NativeEngine {
     for (auto& wk : ws) {
       HIP_CHECK(hipStreamCreateWithFlags(&wk.stream, hipStreamNonBlocking));
     }
+    // wave dispatch table (dense ASGD, pipe-kernel shapes only)
+    wave_ok = cfg.algo == 0 && !ws[0].sparse && cfg.d % 4 == 0 &&
+              cfg.d <= 2048 && cfg.P <= GRAD_WAVE_MAXP &&
+              std::getenv("ASYNCAMD_NO_WAVE") == nullptr;
+    if (wave_ok) {
+      std::vector<GradWaveSlot> hs(cfg.P);
+      wave_max_rows = 0;
+      for (int i = 0; i < cfg.P; ++i) {
+        hs[i].X = (const void*)ws[i].X;
+        hs[i].y = (const float*)ws[i].y;
+        hs[i].wbuf = (const float*)ws[i].wbuf;
+        hs[i].g = (float*)ws[i].g;
+        hs[i].n_out = (int*)ws[i].ctr;
+        hs[i].n_rows = ws[i].n_rows;
+        hs[i].row_start = ws[i].row_start;
+        hs[i].done_flag = (unsigned long long*)ws[i].done_flag;
+        hs[i].done_arr = ws[i].done_arr;
+        wave_max_rows = std::max(wave_max_rows, ws[i].n_rows);
+      }
+      HIP_CHECK(hipMalloc((void**)&slots_dev,
+                          sizeof(GradWaveSlot) * cfg.P));
+      HIP_CHECK(hipMemcpy(slots_dev, hs.data(),
+                          sizeof(GradWaveSlot) * cfg.P,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipStreamCreateWithFlags(&wstream, hipStreamNonBlocking));
+      wave_bper = query_grad_grid(wave_max_rows);
+    }
     inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
     inv_N = 1.0 / (double)cfg.N;
     {  // device pointer tables for the batched update kernel
@@ -628,7 +709,10 @@ struct NativeEngine {
         // buffers (replaces a 3 KB hipMemcpyAsync per dispatch)
         flush_batch(&ready);
         const double td = now_s();
-        for (int wid : ready) dispatch_impl(wid, td, false);
+        if (wave_ok)
+          dispatch_wave(ready, td);
+        else
+          for (int wid : ready) dispatch_impl(wid, td, false);
       }
     }
     HIP_CHECK(hipStreamSynchronize(sstream));
@@ -655,6 +739,14 @@ struct NativeEngine {
     flags_host = nullptr;
     HIP_CHECK(hipFree(arr_dev));
     arr_dev = nullptr;
+    if (slots_dev) {
+      HIP_CHECK(hipFree(slots_dev));
+      slots_dev = nullptr;
+    }
+    if (wstream) {
+      HIP_CHECK(hipStreamDestroy(wstream));
+      wstream = nullptr;
+    }
     HIP_CHECK(hipEventDestroy(update_ev));
     HIP_CHECK(hipStreamDestroy(sstream));
     HIP_CHECK(hipFree(g_tab_dev));

@@ -1,0 +1,29 @@
+// Shared wave-dispatch descriptors between kernels.hip and the native
+// engine (csrc/engine_native.cpp). One grad_dense_wave_kernel launch runs
+// a whole quorum wave of dense-ASGD rounds: per-worker invariants live in
+// a GradWaveSlot device table built once at engine init; per-round
+// variables travel by value in GradWaveCmd (~1 KB, under the 4 KB kernel
+// arg limit).
+#pragma once
+
+#define GRAD_WAVE_MAXP 64
+
+struct GradWaveSlot {  // fixed per worker for the engine's lifetime
+  const void* X;
+  const float* y;
+  const float* wbuf;
+  float* g;
+  int* n_out;
+  long n_rows;
+  long row_start;
+  unsigned long long* done_flag;  // pinned host line (publish_done)
+  unsigned long long* done_arr;   // device arrival counter
+};
+
+struct GradWaveCmd {  // per launch, by value
+  int n;     // active slots in this wave
+  int bper;  // blocks per worker (uniform across the wave)
+  int wid[GRAD_WAVE_MAXP];                  // slot index per wave member
+  unsigned int round_k[GRAD_WAVE_MAXP];     // Philox round key (k_submit+1)
+  unsigned long long done_val[GRAD_WAVE_MAXP];  // completion serial
+};

@@ -41,6 +41,7 @@
 #include <cstdint>
 #include <cstdlib>
 #include "philox.h"
+#include "grad_wave.h"
 
 #define WAVE 64
 #define BLOCK 256
@@ -57,7 +58,7 @@
 // SAGA staging) are ordered before the flag by the fence chain.
 __device__ __forceinline__ void publish_done(
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned long long* done_arr) {
+    unsigned long long* done_arr, unsigned int nblk) {
   if (!done_flag) return;  // uniform per launch: no divergence
   // NO fences here — fence instructions at agent scope on a multi-XCD
   // chip emit per-block L2 writeback/invalidate (buffer_wbl2/inv), which
@@ -69,14 +70,18 @@ __device__ __forceinline__ void publish_done(
   __builtin_amdgcn_s_waitcnt(0);
   __syncthreads();
   if (threadIdx.x == 0) {
-    // monotonic arrival counter (no per-round reset => no reset/flag
-    // ordering race): round `done_val` is complete when the counter
-    // reaches done_val * gridDim.x (grid size is fixed per worker).
     const unsigned long long arrived = __hip_atomic_fetch_add(
         done_arr, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    if (arrived + 1 == done_val * gridDim.x)
+    if (arrived + 1 == (unsigned long long)nblk) {
+      // reset for the next round with a RETURNING exchange, then drain it
+      // before the flag store: guarantees the reset is globally ordered
+      // before the host can observe this round's completion
+      (void)__hip_atomic_exchange(done_arr, 0ull, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      __builtin_amdgcn_s_waitcnt(0);
       __hip_atomic_store(done_flag, done_val, __ATOMIC_RELAXED,
                          __HIP_MEMORY_SCOPE_SYSTEM);
+    }
   }
 }
 
@@ -240,7 +245,7 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
-  publish_done(done_flag, done_val, done_arr);
+  publish_done(done_flag, done_val, done_arr, gridDim.x);
 }
 
 // ------------------------------------------------- K1 pipelined (queue)
@@ -494,7 +499,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_list_kernel(
 }
 
 template <typename XT, bool SAGA, int PBLOCK, int DEPTH, int ITERS>
-__global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
+__device__ void grad_dense_pipe_body(
     const XT* __restrict__ X, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
     float* __restrict__ g_part, int* __restrict__ n_out,
@@ -503,7 +508,8 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
     uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
     int take_all, int objective, unsigned long long* done_flag,
-    unsigned long long done_val, unsigned long long* done_arr) {
+    unsigned long long done_val, unsigned long long* done_arr, int bid,
+    int nblk) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NW = PBLOCK / WAVE;
   constexpr int RPB = NW * ROWS_PER_WAVE;
@@ -532,7 +538,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
   using RV = RowVec<XT>;
 
   const long ngroups = (n_rows + RPB - 1) / RPB;
-  long gi = blockIdx.x;
+  long gi = bid;
   bool done = false;
   while (!done) {
     // ---- scan phase (also prefetches y / alpha into the queue)
@@ -557,7 +563,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
           }
         }
       }
-      gi += gridDim.x;
+      gi += nblk;
     }
     __syncthreads();
     const int nq = min(*qn, QCAP);
@@ -655,12 +661,12 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
   }
   __syncthreads();
   if (g_part != nullptr) {
-    const size_t G = gridDim.x;
+    const size_t G = nblk;
     for (int j = threadIdx.x; j < d; j += PBLOCK) {
       float s = 0.f;
 #pragma unroll
       for (int s2 = 0; s2 < NW; ++s2) s += gacc[(size_t)s2 * DPAD + j];
-      g_part[(size_t)j * G + blockIdx.x] = s;
+      g_part[(size_t)j * G + bid] = s;
     }
   } else {
     for (int j = threadIdx.x; j < d; j += PBLOCK) {
@@ -671,8 +677,49 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
     }
   }
   if (lane == 0 && local_count) atomicAdd(n_out, local_count);
-  publish_done(done_flag, done_val, done_arr);
+  publish_done(done_flag, done_val, done_arr, nblk);
 }
+
+template <typename XT, bool SAGA, int PBLOCK, int DEPTH, int ITERS>
+__global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
+    const XT* __restrict__ X, const float* __restrict__ y,
+    const float* __restrict__ w, float* __restrict__ g_out,
+    float* __restrict__ g_part, int* __restrict__ n_out,
+    float* __restrict__ alpha, int* __restrict__ idx_out,
+    float* __restrict__ e_out, int* __restrict__ pos_ctr,
+    const int* __restrict__ k_dev, int commit_now, long n_rows, int d,
+    uint64_t seed, uint32_t round_k, uint64_t row_start, uint32_t threshold,
+    int take_all, int objective, unsigned long long* done_flag,
+    unsigned long long done_val, unsigned long long* done_arr) {
+  grad_dense_pipe_body<XT, SAGA, PBLOCK, DEPTH, ITERS>(
+      X, y, w, g_out, g_part, n_out, alpha, idx_out, e_out, pos_ctr, k_dev,
+      commit_now, n_rows, d, seed, round_k, row_start, threshold, take_all,
+      objective, done_flag, done_val, done_arr, blockIdx.x, gridDim.x);
+}
+
+// ---- wave launch: ONE kernel runs a whole quorum wave of dense-ASGD
+// rounds (the native engine previously paid ~5 us of launch latency per
+// worker per wave). Block b serves wave slot b / bper with intra-worker
+// block id b % bper; each worker's sub-grid publishes its OWN done flag
+// as it drains, so completions still stagger exactly as with per-worker
+// kernels. Per-worker invariants (X, y, wbuf, g, n_rows, flags) live in a
+// device table built once at engine init; the per-round variables (slot
+// list, Philox round keys, serials) travel by value in the launch args.
+template <typename XT, int PBLOCK, int DEPTH, int ITERS>
+__global__ __launch_bounds__(PBLOCK) void grad_dense_wave_kernel(
+    const GradWaveSlot* __restrict__ slots, GradWaveCmd cmd, int d,
+    uint64_t seed, uint32_t threshold, int take_all, int objective) {
+  const int si = blockIdx.x / cmd.bper;
+  const int bid = blockIdx.x % cmd.bper;
+  const GradWaveSlot sl = slots[cmd.wid[si]];
+  grad_dense_pipe_body<XT, false, PBLOCK, DEPTH, ITERS>(
+      (const XT*)sl.X, sl.y, sl.wbuf, sl.g, nullptr, sl.n_out, nullptr,
+      nullptr, nullptr, nullptr, nullptr, 0, sl.n_rows, d, seed,
+      cmd.round_k[si], (uint64_t)sl.row_start, threshold, take_all,
+      objective, sl.done_flag, cmd.done_val[si], sl.done_arr, bid,
+      cmd.bper);
+}
+
 
 // Sums the per-block partial slabs into g (layout g_part[j][G], contiguous
 // per column). Grid = ceil(d/BLOCK) * SPLITS.
@@ -781,7 +828,7 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
-  publish_done(done_flag, done_val, done_arr);
+  publish_done(done_flag, done_val, done_arr, gridDim.x);
 }
 
 // ---------------------------------------------------------------- K5/K6
@@ -1203,6 +1250,54 @@ void launch_grad_dense_flag(
                                n_rows, d, seed, round_k, row_start, rate,
                                objective, stream, done_flag, done_val,
                                done_arr);
+}
+
+void launch_grad_dense_wave(const void* slots_dev, const void* cmd_host,
+                            long max_rows, int d, uint64_t seed, double rate,
+                            int objective, int x_is_bf16,
+                            hipStream_t stream) {
+  const GradWaveCmd* cmd = (const GradWaveCmd*)cmd_host;
+  const uint32_t thr = philox_threshold(rate);
+  const int take_all = rate >= 1.0;
+  const int grid = cmd->n * cmd->bper;
+  const int iters = (d + 255) / 256;
+  const size_t smem = (size_t)(1 + 4) * (iters * 256) * sizeof(float) +
+                      (size_t)QCAP * sizeof(float) +
+                      (QCAP + 1) * sizeof(int);
+  const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
+  const int depth = dps ? std::atoi(dps) : 4;
+  (void)max_rows;
+#define WAVE_LAUNCH_D(XT, IT, DP)                                            \
+  hipLaunchKernelGGL((grad_dense_wave_kernel<XT, 256, DP, IT>), dim3(grid),  \
+                     dim3(256), smem, stream,                                \
+                     (const GradWaveSlot*)slots_dev, *cmd, d, seed, thr,     \
+                     take_all, objective)
+#define WAVE_LAUNCH(XT, IT)                                                  \
+  do {                                                                       \
+    if (depth == 1) WAVE_LAUNCH_D(XT, IT, 1);                                \
+    else if (depth == 2) WAVE_LAUNCH_D(XT, IT, 2);                           \
+    else if (depth == 6) WAVE_LAUNCH_D(XT, IT, 6);                           \
+    else if (depth == 8) WAVE_LAUNCH_D(XT, IT, 8);                           \
+    else WAVE_LAUNCH_D(XT, IT, 4);                                           \
+  } while (0)
+#define WAVE_DISPATCH(XT)                                                    \
+  do {                                                                       \
+    switch (iters) {                                                         \
+      case 1: WAVE_LAUNCH(XT, 1); break;                                     \
+      case 2: WAVE_LAUNCH(XT, 2); break;                                     \
+      case 3: WAVE_LAUNCH(XT, 3); break;                                     \
+      case 4: WAVE_LAUNCH(XT, 4); break;                                     \
+      case 5: WAVE_LAUNCH(XT, 5); break;                                     \
+      case 6: WAVE_LAUNCH(XT, 6); break;                                     \
+      case 7: WAVE_LAUNCH(XT, 7); break;                                     \
+      default: WAVE_LAUNCH(XT, 8); break;                                    \
+    }                                                                        \
+  } while (0)
+  if (x_is_bf16) WAVE_DISPATCH(__hip_bfloat16);
+  else WAVE_DISPATCH(float);
+#undef WAVE_DISPATCH
+#undef WAVE_LAUNCH
+#undef WAVE_LAUNCH_D
 }
 
 void launch_saga_grad_dense_flag(
