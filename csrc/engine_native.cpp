@@ -229,7 +229,12 @@ struct NativeEngine {
   unsigned long long* arr_dev = nullptr;
   // wave dispatch (dense ASGD): per-worker invariant table + stream
   GradWaveSlot* slots_dev = nullptr;
-  hipStream_t wstream = nullptr;
+  // pool of wave streams: consecutive waves overlap like the per-worker
+  // streams did (same-worker overlap is impossible — in_flight guard)
+  static constexpr int NWSTREAM = 8;
+  hipStream_t wstreams[NWSTREAM] = {};
+  unsigned wave_rr = 0;
+  hipStream_t cur_wst = nullptr;
   int wave_bper = 0;
   long wave_max_rows = 0;
   bool wave_ok = false;
@@ -401,7 +406,8 @@ struct NativeEngine {
     wk.ts = clock;
     wk.k_submit = k;
     if (wk.g_dirty) {  // rejected round left sums; zero on the wave stream
-      HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4, wstream));
+      HIP_CHECK(hipMemsetAsync((void*)wk.g, 0, (size_t)cfg.d * 4,
+                               cur_wst));
       wk.g_dirty = false;
     }
     if (wk.in_flight)
@@ -413,7 +419,9 @@ struct NativeEngine {
   }
 
   void dispatch_wave(const std::vector<int>& ready, double t_now) {
-    HIP_CHECK(hipStreamWaitEvent(wstream, update_ev, 0));
+    cur_wst = wstreams[wave_rr % NWSTREAM];
+    wave_rr += 1;
+    HIP_CHECK(hipStreamWaitEvent(cur_wst, update_ev, 0));
     GradWaveCmd cmd;
     cmd.n = 0;
     cmd.bper = wave_bper;
@@ -428,7 +436,7 @@ struct NativeEngine {
     if (cmd.n > 0) {
       launch_grad_dense_wave(slots_dev, &cmd, wave_max_rows, cfg.d,
                              cfg.seed, cfg.rate, cfg.objective,
-                             ws[0].x_is_bf16, wstream);
+                             ws[0].x_is_bf16, cur_wst);
       HIP_CHECK(hipGetLastError());
     }
   }
@@ -616,7 +624,9 @@ struct NativeEngine {
       HIP_CHECK(hipMemcpy(slots_dev, hs.data(),
                           sizeof(GradWaveSlot) * cfg.P,
                           hipMemcpyHostToDevice));
-      HIP_CHECK(hipStreamCreateWithFlags(&wstream, hipStreamNonBlocking));
+      for (int i = 0; i < NWSTREAM; ++i)
+        HIP_CHECK(hipStreamCreateWithFlags(&wstreams[i],
+                                           hipStreamNonBlocking));
       wave_bper = query_grad_grid(wave_max_rows);
     }
     inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
@@ -743,10 +753,11 @@ struct NativeEngine {
       HIP_CHECK(hipFree(slots_dev));
       slots_dev = nullptr;
     }
-    if (wstream) {
-      HIP_CHECK(hipStreamDestroy(wstream));
-      wstream = nullptr;
-    }
+    for (int i = 0; i < NWSTREAM; ++i)
+      if (wstreams[i]) {
+        HIP_CHECK(hipStreamDestroy(wstreams[i]));
+        wstreams[i] = nullptr;
+      }
     HIP_CHECK(hipEventDestroy(update_ev));
     HIP_CHECK(hipStreamDestroy(sstream));
     HIP_CHECK(hipFree(g_tab_dev));
