@@ -234,6 +234,7 @@ struct NativeEngine {
   static constexpr int NWSTREAM = 8;
   hipStream_t wstreams[NWSTREAM] = {};
   unsigned wave_rr = 0;
+  int wave_interleave = 0;
   hipStream_t cur_wst = nullptr;
   int wave_bper = 0;
   long wave_max_rows = 0;
@@ -425,6 +426,7 @@ struct NativeEngine {
     GradWaveCmd cmd;
     cmd.n = 0;
     cmd.bper = wave_bper;
+    cmd.interleave = wave_interleave;
     for (int wid : ready) {
       dispatch_book(wid, t_now);
       WorkerBuf& wk = ws[wid];
@@ -628,6 +630,8 @@ struct NativeEngine {
         HIP_CHECK(hipStreamCreateWithFlags(&wstreams[i],
                                            hipStreamNonBlocking));
       wave_bper = query_grad_grid(wave_max_rows);
+      const char* wi = std::getenv("ASYNCAMD_WAVE_INTERLEAVE");
+      wave_interleave = wi ? std::atoi(wi) : 0;
     }
     inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
     inv_N = 1.0 / (double)cfg.N;

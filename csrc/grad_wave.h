@@ -23,6 +23,8 @@ struct GradWaveSlot {  // fixed per worker for the engine's lifetime
 struct GradWaveCmd {  // per launch, by value
   int n;     // active slots in this wave
   int bper;  // blocks per worker (uniform across the wave)
+  int interleave;  // 1: block b -> slot b%n (workers progress together);
+                   // 0: block b -> slot b/bper (workers drain in order)
   int wid[GRAD_WAVE_MAXP];                  // slot index per wave member
   unsigned int round_k[GRAD_WAVE_MAXP];     // Philox round key (k_submit+1)
   unsigned long long done_val[GRAD_WAVE_MAXP];  // completion serial

@@ -709,8 +709,10 @@ template <typename XT, int PBLOCK, int DEPTH, int ITERS>
 __global__ __launch_bounds__(PBLOCK) void grad_dense_wave_kernel(
     const GradWaveSlot* __restrict__ slots, GradWaveCmd cmd, int d,
     uint64_t seed, uint32_t threshold, int take_all, int objective) {
-  const int si = blockIdx.x / cmd.bper;
-  const int bid = blockIdx.x % cmd.bper;
+  const int si = cmd.interleave ? (int)(blockIdx.x % cmd.n)
+                                : (int)(blockIdx.x / cmd.bper);
+  const int bid = cmd.interleave ? (int)(blockIdx.x / cmd.n)
+                                 : (int)(blockIdx.x % cmd.bper);
   const GradWaveSlot sl = slots[cmd.wid[si]];
   grad_dense_pipe_body<XT, false, PBLOCK, DEPTH, ITERS>(
       (const XT*)sl.X, sl.y, sl.wbuf, sl.g, nullptr, sl.n_out, nullptr,
