@@ -630,8 +630,11 @@ struct NativeEngine {
         HIP_CHECK(hipStreamCreateWithFlags(&wstreams[i],
                                            hipStreamNonBlocking));
       wave_bper = query_grad_grid(wave_max_rows);
+      // interleaved block->slot mapping measured +21% on the flagship
+      // (workers progress together => completions bunch => bigger update
+      // batches and fuller next waves)
       const char* wi = std::getenv("ASYNCAMD_WAVE_INTERLEAVE");
-      wave_interleave = wi ? std::atoi(wi) : 0;
+      wave_interleave = wi ? std::atoi(wi) : 1;
     }
     inv_batch = (double)cfg.P / (cfg.rate * (double)cfg.N);
     inv_N = 1.0 / (double)cfg.N;
