@@ -73,10 +73,10 @@ def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="asgd-mnist8m", choices=list(MODELS))
     p.add_argument("--gpus", type=int, default=1)
-    # defaults sized so the timed region is >=0.2 s at the flagship's
-    # measured updates/s (a 2000-step region was ~50 ms — too noisy)
-    p.add_argument("--steps", type=int, default=20000)
-    p.add_argument("--warmup", type=int, default=2000)
+    # defaults sized so the timed region is >=0.25 s at the flagship's
+    # measured ~113k updates/s (a 2000-step region was ~20 ms — too noisy)
+    p.add_argument("--steps", type=int, default=30000)
+    p.add_argument("--warmup", type=int, default=3000)
     p.add_argument("--rows", type=int, default=0)
     p.add_argument("--cols", type=int, default=0)
     p.add_argument("--rate", type=float, default=0.0)
