@@ -32,8 +32,10 @@ SOURCES = [ROOT / "csrc" / "kernels.hip", ROOT / "csrc" / "bindings.cpp",
            ROOT / "csrc" / "libsvm_parser.cpp",
            ROOT / "csrc" / "engine_native.cpp",
            ROOT / "csrc" / "engine_resident.hip"]
-HEADERS = [ROOT / "csrc" / "philox.h", ROOT / "csrc" / "multi_update.h"]
+HEADERS = [ROOT / "csrc" / "philox.h", ROOT / "csrc" / "multi_update.h",
+           ROOT / "csrc" / "grad_wave.h"]
 DIST_SRC = ROOT / "csrc" / "server_dist.cpp"
+DIST_HIP = ROOT / "csrc" / "dist_update.hip"
 DIST_DIR = ROOT / "asyncframework_amd" / "_dist_build"
 PROV = ROOT / "asyncframework_amd" / "_hip_core.provenance.json"
 
@@ -57,6 +59,7 @@ def dist_src_hash() -> str:
     h = hashlib.sha256()
     h.update(ARCH.encode())
     h.update(DIST_SRC.read_bytes())
+    h.update(DIST_HIP.read_bytes())
     return h.hexdigest()[:16]
 
 
@@ -115,7 +118,7 @@ def build_dist_core(force: bool = False, verbose: bool = True) -> Path:
     DIST_DIR.mkdir(parents=True, exist_ok=True)
     if force and out.exists():
         out.unlink()  # torch load() otherwise reuses by its own mtime check
-    load(name="_dist_core", sources=[str(DIST_SRC)],
+    load(name="_dist_core", sources=[str(DIST_SRC), str(DIST_HIP)],
          build_directory=str(DIST_DIR), verbose=verbose)
     _write_prov(dist_src_hash=cur)
     return out

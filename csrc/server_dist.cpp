@@ -32,6 +32,16 @@
 // dist.send on the pair group is race-free).
 
 #include <torch/extension.h>
+#ifdef __HIP_PLATFORM_AMD__
+#include <c10/hip/HIPStream.h>
+extern "C" void launch_dist_sgd_update(float*, const float*, float, float,
+                                       int, hipStream_t);
+extern "C" void launch_dist_saga_update(float*, const float*, float*, float,
+                                        float, float, int, hipStream_t);
+namespace at { namespace cuda {
+using c10::hip::getCurrentHIPStream;
+} }
+#endif
 #include <torch/csrc/distributed/c10d/ProcessGroup.hpp>
 
 #include <atomic>
@@ -486,6 +496,27 @@ class DistServer {
   void apply(const at::Tensor& g) {
     at::Tensor gg = g;
     if (gg.device() != w_.device()) gg = gg.to(w_.device());
+#ifdef __HIP_PLATFORM_AMD__
+    if (w_.is_cuda()) {
+      // one fused HIP launch on the current stream instead of 1-3 aten
+      // dispatcher round-trips per update (same arithmetic as the
+      // local-engine kernels; aten path below stays the CPU/gloo oracle)
+      auto stream = at::cuda::getCurrentHIPStream().stream();
+      if (cfg_.asaga)
+        launch_dist_saga_update(
+            w_.data_ptr<float>(), gg.data_ptr<float>(),
+            alpha_bar_.data_ptr<float>(), (float)cfg_.gamma,
+            (float)(1.0 / cfg_.par_recs()), (float)(1.0 / (double)cfg_.N),
+            (int)cfg_.d, stream);
+      else
+        launch_dist_sgd_update(
+            w_.data_ptr<float>(), gg.data_ptr<float>(),
+            (float)(cfg_.gamma /
+                    std::sqrt((double)(k_ / cfg_.P + 1))),
+            (float)(1.0 / cfg_.par_recs()), (int)cfg_.d, stream);
+      return;
+    }
+#endif
     if (cfg_.asaga) {
       // w -= gamma*(g/parRecs); w -= gamma*alphaBar; alphaBar += g/N
       // (SparkASAGAThread.scala:217-220)
