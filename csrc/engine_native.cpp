@@ -59,6 +59,9 @@ void launch_saga_grad_csr(const int*, const int*, const void*, const float*,
                           int*, const int*, int, long, uint64_t, uint32_t,
                           uint64_t, double, int, int, hipStream_t);
 int query_grad_grid(long);
+void launch_grad_csr_wave(const void*, const void*, uint64_t, double, int,
+                          int, int, hipStream_t);
+void launch_saga_commit_wave(const void*, const void*, hipStream_t);
 void launch_grad_dense_wave(const void*, const void*, long, int, uint64_t,
                             double, int, int, hipStream_t);
 void launch_grad_dense_flag(const void*, const float*, const float*,
@@ -229,6 +232,7 @@ struct NativeEngine {
   unsigned long long* arr_dev = nullptr;
   // wave dispatch (dense ASGD): per-worker invariant table + stream
   GradWaveSlot* slots_dev = nullptr;
+  CsrWaveSlot* csr_slots_dev = nullptr;
   // pool of wave streams: consecutive waves overlap like the per-worker
   // streams did (same-worker overlap is impossible — in_flight guard)
   static constexpr int NWSTREAM = 8;
@@ -360,11 +364,23 @@ struct NativeEngine {
     // accept-gated SAGA history commit from the worker's previous round
     // (spill mode: scatter straight into the pinned master table)
     if (cfg.algo == 1 && wk.pending_commit) {
-      float* dst = (float*)(wk.alpha_host ? wk.alpha_host : wk.alpha);
-      launch_saga_commit_devn(dst, (const int*)wk.idx_out,
-                              (const float*)wk.e_out,
-                              (const int*)(wk.ctr + 4), wk.saga_cap,
-                              wk.stream);
+      if (csr_slots_dev) {
+        // wave mode: the previous grad ran on a wave stream, so the
+        // staging must be read with the RMW-load commit kernel (normal
+        // loads could hit a stale L2 line across the stream boundary)
+        CsrCommitCmd cc;
+        cc.n = 1;
+        cc.bper = 16;
+        cc.wid[0] = wid;
+        cc.do_commit[0] = 1;
+        launch_saga_commit_wave(csr_slots_dev, &cc, wk.stream);
+      } else {
+        float* dst = (float*)(wk.alpha_host ? wk.alpha_host : wk.alpha);
+        launch_saga_commit_devn(dst, (const int*)wk.idx_out,
+                                (const float*)wk.e_out,
+                                (const int*)(wk.ctr + 4), wk.saga_cap,
+                                wk.stream);
+      }
       HIP_CHECK(hipGetLastError());
     }
     wk.pending_commit = false;
@@ -423,6 +439,23 @@ struct NativeEngine {
     cur_wst = wstreams[wave_rr % NWSTREAM];
     wave_rr += 1;
     HIP_CHECK(hipStreamWaitEvent(cur_wst, update_ev, 0));
+    if (csr_slots_dev && cfg.algo == 1) {
+      // SAGA: one commit + staging-reset kernel for the whole wave,
+      // stream-ordered before the wave's gradient kernel
+      CsrCommitCmd cc;
+      cc.n = 0;
+      cc.bper = 16;
+      for (int wid : ready) {
+        cc.wid[cc.n] = wid;
+        cc.do_commit[cc.n] = ws[wid].pending_commit ? 1 : 0;
+        ws[wid].pending_commit = false;
+        cc.n += 1;
+      }
+      if (cc.n > 0) {
+        launch_saga_commit_wave(csr_slots_dev, &cc, cur_wst);
+        HIP_CHECK(hipGetLastError());
+      }
+    }
     GradWaveCmd cmd;
     cmd.n = 0;
     cmd.bper = wave_bper;
@@ -436,9 +469,14 @@ struct NativeEngine {
       cmd.n += 1;
     }
     if (cmd.n > 0) {
-      launch_grad_dense_wave(slots_dev, &cmd, wave_max_rows, cfg.d,
-                             cfg.seed, cfg.rate, cfg.objective,
-                             ws[0].x_is_bf16, cur_wst);
+      if (slots_dev)
+        launch_grad_dense_wave(slots_dev, &cmd, wave_max_rows, cfg.d,
+                               cfg.seed, cfg.rate, cfg.objective,
+                               ws[0].x_is_bf16, cur_wst);
+      else
+        launch_grad_csr_wave(csr_slots_dev, &cmd, cfg.seed, cfg.rate,
+                             cfg.objective, ws[0].x_is_bf16,
+                             cfg.algo == 1 ? 1 : 0, cur_wst);
       HIP_CHECK(hipGetLastError());
     }
   }
@@ -602,30 +640,63 @@ struct NativeEngine {
     for (auto& wk : ws) {
       HIP_CHECK(hipStreamCreateWithFlags(&wk.stream, hipStreamNonBlocking));
     }
-    // wave dispatch table (dense ASGD, pipe-kernel shapes only)
-    wave_ok = cfg.algo == 0 && !ws[0].sparse && cfg.d % 4 == 0 &&
-              cfg.d <= 2048 && cfg.P <= GRAD_WAVE_MAXP &&
-              std::getenv("ASYNCAMD_NO_WAVE") == nullptr;
+    // wave dispatch tables. Dense: ASGD pipe-kernel shapes. CSR: ASGD and
+    // SAGA with device-resident history (spill mode keeps per-worker
+    // launches — its per-round scan/gather chain doesn't batch).
+    const bool no_wave = std::getenv("ASYNCAMD_NO_WAVE") != nullptr;
+    const bool dense_wave = cfg.algo == 0 && !ws[0].sparse &&
+                            cfg.d % 4 == 0 && cfg.d <= 2048;
+    const bool csr_wave =
+        ws[0].sparse && (cfg.algo == 0 || ws[0].alpha_host == 0);
+    wave_ok = (dense_wave || csr_wave) && cfg.P <= GRAD_WAVE_MAXP &&
+              !no_wave;
     if (wave_ok) {
-      std::vector<GradWaveSlot> hs(cfg.P);
       wave_max_rows = 0;
-      for (int i = 0; i < cfg.P; ++i) {
-        hs[i].X = (const void*)ws[i].X;
-        hs[i].y = (const float*)ws[i].y;
-        hs[i].wbuf = (const float*)ws[i].wbuf;
-        hs[i].g = (float*)ws[i].g;
-        hs[i].n_out = (int*)ws[i].ctr;
-        hs[i].n_rows = ws[i].n_rows;
-        hs[i].row_start = ws[i].row_start;
-        hs[i].done_flag = (unsigned long long*)ws[i].done_flag;
-        hs[i].done_arr = ws[i].done_arr;
+      for (int i = 0; i < cfg.P; ++i)
         wave_max_rows = std::max(wave_max_rows, ws[i].n_rows);
+      if (dense_wave) {
+        std::vector<GradWaveSlot> hs(cfg.P);
+        for (int i = 0; i < cfg.P; ++i) {
+          hs[i].X = (const void*)ws[i].X;
+          hs[i].y = (const float*)ws[i].y;
+          hs[i].wbuf = (const float*)ws[i].wbuf;
+          hs[i].g = (float*)ws[i].g;
+          hs[i].n_out = (int*)ws[i].ctr;
+          hs[i].n_rows = ws[i].n_rows;
+          hs[i].row_start = ws[i].row_start;
+          hs[i].done_flag = (unsigned long long*)ws[i].done_flag;
+          hs[i].done_arr = ws[i].done_arr;
+        }
+        HIP_CHECK(hipMalloc((void**)&slots_dev,
+                            sizeof(GradWaveSlot) * cfg.P));
+        HIP_CHECK(hipMemcpy(slots_dev, hs.data(),
+                            sizeof(GradWaveSlot) * cfg.P,
+                            hipMemcpyHostToDevice));
+      } else {
+        std::vector<CsrWaveSlot> hs(cfg.P);
+        for (int i = 0; i < cfg.P; ++i) {
+          hs[i].indptr = (const int*)ws[i].indptr;
+          hs[i].indices = (const int*)ws[i].indices;
+          hs[i].values = (const void*)ws[i].values;
+          hs[i].y = (const float*)ws[i].y;
+          hs[i].wbuf = (const float*)ws[i].wbuf;
+          hs[i].alpha = (float*)ws[i].alpha;
+          hs[i].g = (float*)ws[i].g;
+          hs[i].n_out = (int*)ws[i].ctr;
+          hs[i].idx_out = (int*)ws[i].idx_out;
+          hs[i].e_out = (float*)ws[i].e_out;
+          hs[i].pos_ctr = (int*)(ws[i].ctr + 4);
+          hs[i].n_rows = ws[i].n_rows;
+          hs[i].row_start = ws[i].row_start;
+          hs[i].done_flag = (unsigned long long*)ws[i].done_flag;
+          hs[i].done_arr = ws[i].done_arr;
+        }
+        HIP_CHECK(hipMalloc((void**)&csr_slots_dev,
+                            sizeof(CsrWaveSlot) * cfg.P));
+        HIP_CHECK(hipMemcpy(csr_slots_dev, hs.data(),
+                            sizeof(CsrWaveSlot) * cfg.P,
+                            hipMemcpyHostToDevice));
       }
-      HIP_CHECK(hipMalloc((void**)&slots_dev,
-                          sizeof(GradWaveSlot) * cfg.P));
-      HIP_CHECK(hipMemcpy(slots_dev, hs.data(),
-                          sizeof(GradWaveSlot) * cfg.P,
-                          hipMemcpyHostToDevice));
       for (int i = 0; i < NWSTREAM; ++i)
         HIP_CHECK(hipStreamCreateWithFlags(&wstreams[i],
                                            hipStreamNonBlocking));
@@ -759,6 +830,10 @@ struct NativeEngine {
     if (slots_dev) {
       HIP_CHECK(hipFree(slots_dev));
       slots_dev = nullptr;
+    }
+    if (csr_slots_dev) {
+      HIP_CHECK(hipFree(csr_slots_dev));
+      csr_slots_dev = nullptr;
     }
     for (int i = 0; i < NWSTREAM; ++i)
       if (wstreams[i]) {

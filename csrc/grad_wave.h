@@ -29,3 +29,31 @@ struct GradWaveCmd {  // per launch, by value
   unsigned int round_k[GRAD_WAVE_MAXP];     // Philox round key (k_submit+1)
   unsigned long long done_val[GRAD_WAVE_MAXP];  // completion serial
 };
+
+// ---- CSR wave (SAGA + ASGD sparse) ----
+
+struct CsrWaveSlot {  // fixed per worker for the engine's lifetime
+  const int* indptr;
+  const int* indices;
+  const void* values;
+  const float* y;
+  const float* wbuf;
+  float* alpha;      // worker-resident SAGA history slice (null for ASGD)
+  float* g;
+  int* n_out;        // wk.ctr
+  int* idx_out;      // staging row list (SAGA)
+  float* e_out;      // staging scalars (SAGA)
+  int* pos_ctr;      // staging counter (wk.ctr + 4)
+  long n_rows;
+  long row_start;
+  unsigned long long* done_flag;
+  unsigned long long* done_arr;
+};
+
+struct CsrCommitCmd {  // per launch, by value: commit + staging reset pass
+  int n;
+  int bper;
+  int wid[GRAD_WAVE_MAXP];
+  int do_commit[GRAD_WAVE_MAXP];  // 1: scatter staged scalars, then reset;
+                                  // 0: reset only (rejected previous round)
+};

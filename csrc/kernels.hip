@@ -759,7 +759,7 @@ __global__ __launch_bounds__(BLOCK) void reduce_partials_kernel(
 // wave (the gradient scatter uses global atomics — no LDS slabs — so
 // unlike the dense kernel, sub-waves here cost no occupancy).
 template <typename VT, bool SAGA, int LPR>
-__global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
+__device__ void grad_csr_body(
     const int* __restrict__ indptr, const int* __restrict__ indices,
     const VT* __restrict__ values, const float* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ g_out,
@@ -769,7 +769,7 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
     uint64_t row_start, uint32_t threshold, int take_all, int objective,
     unsigned long long* done_flag, unsigned long long done_val,
-    unsigned long long* done_arr) {
+    unsigned long long* done_arr, int bid, int nblk) {
   if (k_dev) round_k = (uint32_t)(*k_dev) + 1u;
   constexpr int NSUB = WAVE / LPR;
   const int wave = threadIdx.x >> 6;
@@ -777,8 +777,8 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
   const int sub = lane / LPR;
   const int sl = lane % LPR;
   int local_count = 0;
-  const long gstride = (long)gridDim.x * ROWS_PER_BLOCK_ITER;
-  for (long bb = (long)blockIdx.x * ROWS_PER_BLOCK_ITER; bb < n_rows;
+  const long gstride = (long)nblk * ROWS_PER_BLOCK_ITER;
+  for (long bb = (long)bid * ROWS_PER_BLOCK_ITER; bb < n_rows;
        bb += gstride) {
     const long base = bb + (long)wave * ROWS_PER_WAVE;
     if (base >= n_rows) continue;
@@ -816,9 +816,16 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
             if (commit_now) {
               alpha[rr] = e;
             } else {
+              // ATOMIC staging: the wave-mode commit kernel reads these
+              // from another kernel with no stream-order edge, so the
+              // writes must land at the agent coherence point (plain
+              // stores can sit in one XCD's L2)
               const int pos = atomicAdd(pos_ctr, 1);
-              idx_out[pos] = (int)rr;
-              e_out[pos] = e;
+              __hip_atomic_exchange(&idx_out[pos], (int)rr,
+                                    __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+              __hip_atomic_exchange(&e_out[pos], e, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
             }
           }
         }
@@ -830,8 +837,96 @@ __global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
     }
   }
   if (sl == 0 && local_count) atomicAdd(n_out, local_count);
-  publish_done(done_flag, done_val, done_arr, gridDim.x);
+  publish_done(done_flag, done_val, done_arr, nblk);
 }
+
+template <typename VT, bool SAGA, int LPR>
+__global__ __launch_bounds__(BLOCK) void grad_csr_kernel(
+    const int* __restrict__ indptr, const int* __restrict__ indices,
+    const VT* __restrict__ values, const float* __restrict__ y,
+    const float* __restrict__ w, float* __restrict__ g_out,
+    int* __restrict__ n_out, float* __restrict__ alpha,
+    int* __restrict__ idx_out, float* __restrict__ e_out,
+    int* __restrict__ pos_ctr, const int* __restrict__ k_dev,
+    int commit_now, long n_rows, uint64_t seed, uint32_t round_k,
+    uint64_t row_start, uint32_t threshold, int take_all, int objective,
+    unsigned long long* done_flag, unsigned long long done_val,
+    unsigned long long* done_arr) {
+  grad_csr_body<VT, SAGA, LPR>(indptr, indices, values, y, w, g_out, n_out,
+                               alpha, idx_out, e_out, pos_ctr, k_dev,
+                               commit_now, n_rows, seed, round_k, row_start,
+                               threshold, take_all, objective, done_flag,
+                               done_val, done_arr, blockIdx.x, gridDim.x);
+}
+
+// CSR wave: one kernel per quorum wave (see grad_dense_wave_kernel)
+template <typename VT, bool SAGA, int LPR>
+__global__ __launch_bounds__(BLOCK) void grad_csr_wave_kernel(
+    const CsrWaveSlot* __restrict__ slots, GradWaveCmd cmd, uint64_t seed,
+    uint32_t threshold, int take_all, int objective) {
+  const int si = cmd.interleave ? (int)(blockIdx.x % cmd.n)
+                                : (int)(blockIdx.x / cmd.bper);
+  const int bid = cmd.interleave ? (int)(blockIdx.x / cmd.n)
+                                 : (int)(blockIdx.x % cmd.bper);
+  const CsrWaveSlot sl = slots[cmd.wid[si]];
+  grad_csr_body<VT, SAGA, LPR>(
+      sl.indptr, sl.indices, (const VT*)sl.values, sl.y, sl.wbuf, sl.g,
+      sl.n_out, sl.alpha, sl.idx_out, sl.e_out, sl.pos_ctr, nullptr, 0,
+      sl.n_rows, seed, cmd.round_k[si], (uint64_t)sl.row_start, threshold,
+      take_all, objective, sl.done_flag, cmd.done_val[si], sl.done_arr, bid,
+      cmd.bper);
+}
+
+// SAGA commit + staging-reset pass for a whole wave, launched before the
+// wave's grad kernel on the same stream. Scatters the previous accepted
+// round's staged scalars into the worker-resident alpha slice
+// (commit-gated) and resets the staging counter unconditionally (a
+// REJECTED previous round must also restart its staging list). Reads use
+// RMW loads: the staging was written by another kernel with no
+// stream-order edge, so normal loads could hit a stale L2 line.
+__device__ __forceinline__ int atomic_load_int(int* p) {
+  return __hip_atomic_fetch_add(p, 0, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ float atomic_load_f32(float* p) {
+  return __hip_atomic_fetch_add(p, 0.f, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__global__ __launch_bounds__(BLOCK) void saga_commit_wave_kernel(
+    const CsrWaveSlot* __restrict__ slots, CsrCommitCmd cmd) {
+  const int si = (int)(blockIdx.x % cmd.n);
+  const int bid = (int)(blockIdx.x / cmd.n);
+  const CsrWaveSlot sl = slots[cmd.wid[si]];
+  const int cnt = atomic_load_int(sl.pos_ctr);
+  if (cmd.do_commit[si]) {
+    for (int i = bid * BLOCK + threadIdx.x; i < cnt;
+         i += cmd.bper * BLOCK) {
+      const int r = atomic_load_int(&sl.idx_out[i]);
+      const float e = atomic_load_f32(&sl.e_out[i]);
+      sl.alpha[r] = e;  // only this worker's rounds touch its slice
+    }
+  }
+  // last block of the slot resets the staging + sample counters (reuses
+  // done_arr, which is 0 between rounds; the wave's grad kernel is
+  // stream-ordered after this kernel)
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const unsigned long long a = __hip_atomic_fetch_add(
+        sl.done_arr, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (a + 1 == (unsigned long long)cmd.bper) {
+      (void)__hip_atomic_exchange(sl.done_arr, 0ull, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      (void)__hip_atomic_exchange(sl.pos_ctr, 0, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      (void)__hip_atomic_exchange(sl.n_out, 0, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      __builtin_amdgcn_s_waitcnt(0);
+    }
+  }
+}
+
 
 // ---------------------------------------------------------------- K5/K6
 
@@ -1377,6 +1472,46 @@ static inline int csr_lpr() {
   const char* s = std::getenv("ASYNCAMD_CSR_LPR");
   const int v = s ? std::atoi(s) : 32;  // measured best on rcv1 shape
   return (v == 8 || v == 16 || v == 32 || v == 64) ? v : 32;
+}
+
+void launch_grad_csr_wave(const void* slots_dev, const void* cmd_host,
+                          uint64_t seed, double rate, int objective,
+                          int v_is_bf16, int saga, hipStream_t stream) {
+  const GradWaveCmd* cmd = (const GradWaveCmd*)cmd_host;
+  const uint32_t thr = philox_threshold(rate);
+  const int take_all = rate >= 1.0;
+  const int grid = cmd->n * cmd->bper;
+  const int lpr = csr_lpr();
+#define CSRW_LAUNCH(VT, SG, L)                                               \
+  hipLaunchKernelGGL((grad_csr_wave_kernel<VT, SG, L>), dim3(grid),          \
+                     dim3(BLOCK), 0, stream,                                 \
+                     (const CsrWaveSlot*)slots_dev, *cmd, seed, thr,         \
+                     take_all, objective)
+#define CSRW_LPR(VT, SG)                                                     \
+  do {                                                                       \
+    if (lpr == 8) CSRW_LAUNCH(VT, SG, 8);                                    \
+    else if (lpr == 32) CSRW_LAUNCH(VT, SG, 32);                             \
+    else if (lpr == 64) CSRW_LAUNCH(VT, SG, 64);                             \
+    else CSRW_LAUNCH(VT, SG, 16);                                            \
+  } while (0)
+#define CSRW_DISPATCH(VT)                                                    \
+  do {                                                                       \
+    if (saga) CSRW_LPR(VT, true);                                            \
+    else CSRW_LPR(VT, false);                                                \
+  } while (0)
+  if (v_is_bf16) CSRW_DISPATCH(__hip_bfloat16);
+  else CSRW_DISPATCH(float);
+#undef CSRW_DISPATCH
+#undef CSRW_LPR
+#undef CSRW_LAUNCH
+}
+
+void launch_saga_commit_wave(const void* slots_dev, const void* cmd_host,
+                             hipStream_t stream) {
+  const CsrCommitCmd* cmd = (const CsrCommitCmd*)cmd_host;
+  const int grid = cmd->n * cmd->bper;
+  hipLaunchKernelGGL(saga_commit_wave_kernel, dim3(grid), dim3(BLOCK), 0,
+                     stream, (const CsrWaveSlot*)slots_dev, *cmd);
 }
 
 void launch_grad_csr_flag(
