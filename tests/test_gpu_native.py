@@ -158,3 +158,40 @@ def test_native_host_spill_multiworker():
     assert res["k"] >= cfg.num_iterations
     assert all(t.is_pinned() for t in eng.alpha_tables)
     assert any(int((t != 0).sum()) > 0 for t in eng.alpha_tables)
+
+
+def test_native_csr_saga_wave_matches_nowave_p1():
+    """CSR SAGA at P=1 is a deterministic sequential iteration: the wave
+    dispatch path (grad_csr_wave_kernel + saga_commit_wave_kernel) must
+    reproduce the per-worker singleton path (same kernels, same Philox
+    keys; only atomic accumulation order may differ)."""
+    import os
+    from asyncframework_amd.data.synthetic import synthetic_csr
+
+    def run(no_wave):
+        if no_wave:
+            os.environ["ASYNCAMD_NO_WAVE"] = "1"
+        else:
+            os.environ.pop("ASYNCAMD_NO_WAVE", None)
+        cfg = _cfg(algo="asaga", gamma=0.05, num_workers=1,
+                   num_iterations=120, N=20_000, d=512, batch_rate=0.02)
+        indptr, indices, values, y = synthetic_csr(
+            cfg.N, cfg.d, seed=11, device="cuda:0")
+        sh = Shard(row_start=0, n_rows=cfg.N, indptr=indptr,
+                   indices=indices, values=values, y=y)
+        eng = NativeLocalEngine(cfg, [sh], torch.device("cuda:0"))
+        res = eng.run(max_wall_s=120)
+        assert res["k"] == cfg.num_iterations
+        return eng.w.clone(), eng.alpha_tables[0].clone()
+
+    try:
+        w_wave, a_wave = run(no_wave=False)
+        w_ref, a_ref = run(no_wave=True)
+    finally:
+        os.environ.pop("ASYNCAMD_NO_WAVE", None)
+    # identical sampled-row sets => identical alpha nonzero patterns
+    assert torch.equal(a_wave != 0, a_ref != 0)
+    arel = float((a_wave - a_ref).norm() / (a_ref.norm() + 1e-12))
+    assert arel < 2e-3, arel
+    rel = float((w_wave - w_ref).norm() / (w_ref.norm() + 1e-12))
+    assert rel < 1e-3, rel
