@@ -62,8 +62,11 @@ int query_grad_grid(long);
 void launch_grad_csr_wave(const void*, const void*, uint64_t, double, int,
                           int, int, hipStream_t);
 void launch_saga_commit_wave(const void*, const void*, hipStream_t);
+void launch_scan_rows_wave(const void*, const void*, uint64_t, double,
+                           hipStream_t);
+void launch_alpha_gather_wave(const void*, const void*, hipStream_t);
 void launch_grad_dense_wave(const void*, const void*, long, int, uint64_t,
-                            double, int, int, hipStream_t);
+                            double, int, int, int, hipStream_t);
 void launch_grad_dense_flag(const void*, const float*, const float*,
                             float*, float*, int*, const int*, long, int,
                             uint64_t, uint32_t, uint64_t, double, int, int,
@@ -233,6 +236,8 @@ struct NativeEngine {
   // wave dispatch (dense ASGD): per-worker invariant table + stream
   GradWaveSlot* slots_dev = nullptr;
   CsrWaveSlot* csr_slots_dev = nullptr;
+  CommitSlot* commit_slots_dev = nullptr;
+  bool wave_spill = false;
   // pool of wave streams: consecutive waves overlap like the per-worker
   // streams did (same-worker overlap is impossible — in_flight guard)
   static constexpr int NWSTREAM = 8;
@@ -364,7 +369,7 @@ struct NativeEngine {
     // accept-gated SAGA history commit from the worker's previous round
     // (spill mode: scatter straight into the pinned master table)
     if (cfg.algo == 1 && wk.pending_commit) {
-      if (csr_slots_dev) {
+      if (commit_slots_dev) {
         // wave mode: the previous grad ran on a wave stream, so the
         // staging must be read with the RMW-load commit kernel (normal
         // loads could hit a stale L2 line across the stream boundary)
@@ -373,7 +378,7 @@ struct NativeEngine {
         cc.bper = 16;
         cc.wid[0] = wid;
         cc.do_commit[0] = 1;
-        launch_saga_commit_wave(csr_slots_dev, &cc, wk.stream);
+        launch_saga_commit_wave(commit_slots_dev, &cc, wk.stream);
       } else {
         float* dst = (float*)(wk.alpha_host ? wk.alpha_host : wk.alpha);
         launch_saga_commit_devn(dst, (const int*)wk.idx_out,
@@ -439,7 +444,7 @@ struct NativeEngine {
     cur_wst = wstreams[wave_rr % NWSTREAM];
     wave_rr += 1;
     HIP_CHECK(hipStreamWaitEvent(cur_wst, update_ev, 0));
-    if (csr_slots_dev && cfg.algo == 1) {
+    if (commit_slots_dev) {
       // SAGA: one commit + staging-reset kernel for the whole wave,
       // stream-ordered before the wave's gradient kernel
       CsrCommitCmd cc;
@@ -452,7 +457,7 @@ struct NativeEngine {
         cc.n += 1;
       }
       if (cc.n > 0) {
-        launch_saga_commit_wave(csr_slots_dev, &cc, cur_wst);
+        launch_saga_commit_wave(commit_slots_dev, &cc, cur_wst);
         HIP_CHECK(hipGetLastError());
       }
     }
@@ -469,14 +474,24 @@ struct NativeEngine {
       cmd.n += 1;
     }
     if (cmd.n > 0) {
-      if (slots_dev)
+      if (slots_dev) {
+        if (wave_spill) {
+          // spill refresh, stream-ordered: recompute the round's Philox
+          // row sets, then gather those master entries into the staging
+          // tables the gradient reads
+          launch_scan_rows_wave(slots_dev, &cmd, cfg.seed, cfg.rate,
+                                cur_wst);
+          launch_alpha_gather_wave(slots_dev, &cmd, cur_wst);
+        }
         launch_grad_dense_wave(slots_dev, &cmd, wave_max_rows, cfg.d,
                                cfg.seed, cfg.rate, cfg.objective,
-                               ws[0].x_is_bf16, cur_wst);
-      else
+                               ws[0].x_is_bf16, cfg.algo == 1 ? 1 : 0,
+                               cur_wst);
+      } else {
         launch_grad_csr_wave(csr_slots_dev, &cmd, cfg.seed, cfg.rate,
                              cfg.objective, ws[0].x_is_bf16,
                              cfg.algo == 1 ? 1 : 0, cur_wst);
+      }
       HIP_CHECK(hipGetLastError());
     }
   }
@@ -644,12 +659,12 @@ struct NativeEngine {
     // SAGA with device-resident history (spill mode keeps per-worker
     // launches — its per-round scan/gather chain doesn't batch).
     const bool no_wave = std::getenv("ASYNCAMD_NO_WAVE") != nullptr;
-    const bool dense_wave = cfg.algo == 0 && !ws[0].sparse &&
-                            cfg.d % 4 == 0 && cfg.d <= 2048;
-    const bool csr_wave =
-        ws[0].sparse && (cfg.algo == 0 || ws[0].alpha_host == 0);
+    const bool dense_wave =
+        !ws[0].sparse && cfg.d % 4 == 0 && cfg.d <= 2048;
+    const bool csr_wave = ws[0].sparse;
     wave_ok = (dense_wave || csr_wave) && cfg.P <= GRAD_WAVE_MAXP &&
               !no_wave;
+    wave_spill = cfg.algo == 1 && ws[0].alpha_host != 0;
     if (wave_ok) {
       wave_max_rows = 0;
       for (int i = 0; i < cfg.P; ++i)
@@ -666,6 +681,14 @@ struct NativeEngine {
           hs[i].row_start = ws[i].row_start;
           hs[i].done_flag = (unsigned long long*)ws[i].done_flag;
           hs[i].done_arr = ws[i].done_arr;
+          hs[i].alpha = (float*)ws[i].alpha;
+          hs[i].idx_out = (int*)ws[i].idx_out;
+          hs[i].e_out = (float*)ws[i].e_out;
+          hs[i].pos_ctr = (int*)(ws[i].ctr + 4);
+          hs[i].alpha_host = (const float*)ws[i].alpha_host;
+          hs[i].srows = (int*)ws[i].srows;
+          hs[i].sylist = (float*)ws[i].sylist;
+          hs[i].scnt = (int*)ws[i].scnt;
         }
         HIP_CHECK(hipMalloc((void**)&slots_dev,
                             sizeof(GradWaveSlot) * cfg.P));
@@ -695,6 +718,24 @@ struct NativeEngine {
                             sizeof(CsrWaveSlot) * cfg.P));
         HIP_CHECK(hipMemcpy(csr_slots_dev, hs.data(),
                             sizeof(CsrWaveSlot) * cfg.P,
+                            hipMemcpyHostToDevice));
+      }
+      if (cfg.algo == 1) {  // unified commit view (dense + CSR SAGA)
+        std::vector<CommitSlot> cs(cfg.P);
+        for (int i = 0; i < cfg.P; ++i) {
+          cs[i].dst = (float*)(ws[i].alpha_host ? ws[i].alpha_host
+                                                : ws[i].alpha);
+          cs[i].idx = (int*)ws[i].idx_out;
+          cs[i].e = (float*)ws[i].e_out;
+          cs[i].pos_ctr = (int*)(ws[i].ctr + 4);
+          cs[i].scnt = ws[i].alpha_host ? (int*)ws[i].scnt : nullptr;
+          cs[i].n_out = (int*)ws[i].ctr;
+          cs[i].arr = ws[i].done_arr;
+        }
+        HIP_CHECK(hipMalloc((void**)&commit_slots_dev,
+                            sizeof(CommitSlot) * cfg.P));
+        HIP_CHECK(hipMemcpy(commit_slots_dev, cs.data(),
+                            sizeof(CommitSlot) * cfg.P,
                             hipMemcpyHostToDevice));
       }
       for (int i = 0; i < NWSTREAM; ++i)
@@ -834,6 +875,10 @@ struct NativeEngine {
     if (csr_slots_dev) {
       HIP_CHECK(hipFree(csr_slots_dev));
       csr_slots_dev = nullptr;
+    }
+    if (commit_slots_dev) {
+      HIP_CHECK(hipFree(commit_slots_dev));
+      commit_slots_dev = nullptr;
     }
     for (int i = 0; i < NWSTREAM; ++i)
       if (wstreams[i]) {

@@ -18,6 +18,15 @@ struct GradWaveSlot {  // fixed per worker for the engine's lifetime
   long row_start;
   unsigned long long* done_flag;  // pinned host line (publish_done)
   unsigned long long* done_arr;   // device arrival counter
+  // SAGA fields (null for ASGD)
+  float* alpha;            // device table (resident) or staging (spill)
+  int* idx_out;            // staged row list
+  float* e_out;            // staged scalars
+  int* pos_ctr;            // staging counter (wk.ctr + 4)
+  const float* alpha_host; // pinned master table (spill mode) or null
+  int* srows;              // spill refresh: sampled-row list
+  float* sylist;           //   (y values — unused by the wave path)
+  int* scnt;               //   sampled-row count
 };
 
 struct GradWaveCmd {  // per launch, by value
@@ -56,4 +65,15 @@ struct CsrCommitCmd {  // per launch, by value: commit + staging reset pass
   int wid[GRAD_WAVE_MAXP];
   int do_commit[GRAD_WAVE_MAXP];  // 1: scatter staged scalars, then reset;
                                   // 0: reset only (rejected previous round)
+};
+
+// unified commit view (dense SAGA — resident or host-spill — and CSR SAGA)
+struct CommitSlot {
+  float* dst;     // alpha_host (spill) or the worker's alpha table
+  int* idx;       // staged row list
+  float* e;       // staged scalars
+  int* pos_ctr;   // staging counter to reset
+  int* scnt;      // spill refresh counter to reset (null otherwise)
+  int* n_out;     // sampled-count accumulator to reset
+  unsigned long long* arr;  // per-slot arrival counter (reused done_arr)
 };

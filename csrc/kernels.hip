@@ -206,9 +206,13 @@ __global__ __launch_bounds__(BLOCK) void grad_dense_kernel(
               // place (each row sampled once per round, read-before-write)
               alpha[rr] = e;
             } else {
+              // atomic staging: read cross-stream in wave mode
               const int pos = atomicAdd(pos_ctr, 1);
-              idx_out[pos] = (int)rr;
-              e_out[pos] = e;
+              __hip_atomic_exchange(&idx_out[pos], (int)rr,
+                                    __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
+              __hip_atomic_exchange(&e_out[pos], e, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
             }
           }
         }
@@ -337,11 +341,12 @@ __device__ __forceinline__ __amdgpu_buffer_rsrc_t row_rsrc(
 // with round r's gradient/update. The round key comes from a dedicated
 // scan_round counter bumped stream-order by bump_counter_kernel (reading
 // k_dev here would race the concurrent update's k++).
-__global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
+__device__ void scan_rows_body(
     const float* __restrict__ y, int* __restrict__ rowlist,
     float* __restrict__ ylist, int* __restrict__ count_dev,
     const int* __restrict__ scan_round_dev, long n_rows, uint64_t seed,
-    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all) {
+    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all,
+    int bid, int nblk) {
   if (scan_round_dev) round_k = (uint32_t)(*scan_round_dev);
   constexpr int SCAP = 2048;
   __shared__ int rows_s[SCAP];
@@ -350,7 +355,7 @@ __global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const long ngroups = (n_rows + ROWS_PER_BLOCK_ITER - 1) / ROWS_PER_BLOCK_ITER;
-  long gi = blockIdx.x;
+  long gi = bid;
   bool done = false;
   while (!done) {
     while (true) {
@@ -372,7 +377,7 @@ __global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
           }
         }
       }
-      gi += gridDim.x;
+      gi += nblk;
     }
     __syncthreads();
     const int nq = min(qn_s, SCAP);
@@ -387,6 +392,16 @@ __global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
     if (threadIdx.x == 0) qn_s = 0;
     done = gi >= ngroups;
   }
+}
+
+__global__ __launch_bounds__(BLOCK) void scan_rows_kernel(
+    const float* __restrict__ y, int* __restrict__ rowlist,
+    float* __restrict__ ylist, int* __restrict__ count_dev,
+    const int* __restrict__ scan_round_dev, long n_rows, uint64_t seed,
+    uint32_t round_k, uint64_t row_start, uint32_t threshold, int take_all) {
+  scan_rows_body(y, rowlist, ylist, count_dev, scan_round_dev, n_rows, seed,
+                 round_k, row_start, threshold, take_all, blockIdx.x,
+                 gridDim.x);
 }
 
 __global__ void bump_counter_kernel(int* __restrict__ p) { *p += 1; }
@@ -649,9 +664,12 @@ __device__ void grad_dense_pipe_body(
         if (commit_now) {
           alpha[row] = e;
         } else {
+          // atomic staging: read cross-stream in wave mode
           const int gpos = atomicAdd(pos_ctr, 1);
-          idx_out[gpos] = row;
-          e_out[gpos] = e;
+          __hip_atomic_exchange(&idx_out[gpos], row, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
+          __hip_atomic_exchange(&e_out[gpos], e, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT);
         }
       }
       __syncthreads();
@@ -705,7 +723,7 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_pipe_kernel(
 // kernels. Per-worker invariants (X, y, wbuf, g, n_rows, flags) live in a
 // device table built once at engine init; the per-round variables (slot
 // list, Philox round keys, serials) travel by value in the launch args.
-template <typename XT, int PBLOCK, int DEPTH, int ITERS>
+template <typename XT, bool SAGA, int PBLOCK, int DEPTH, int ITERS>
 __global__ __launch_bounds__(PBLOCK) void grad_dense_wave_kernel(
     const GradWaveSlot* __restrict__ slots, GradWaveCmd cmd, int d,
     uint64_t seed, uint32_t threshold, int take_all, int objective) {
@@ -714,9 +732,9 @@ __global__ __launch_bounds__(PBLOCK) void grad_dense_wave_kernel(
   const int bid = cmd.interleave ? (int)(blockIdx.x / cmd.n)
                                  : (int)(blockIdx.x % cmd.bper);
   const GradWaveSlot sl = slots[cmd.wid[si]];
-  grad_dense_pipe_body<XT, false, PBLOCK, DEPTH, ITERS>(
-      (const XT*)sl.X, sl.y, sl.wbuf, sl.g, nullptr, sl.n_out, nullptr,
-      nullptr, nullptr, nullptr, nullptr, 0, sl.n_rows, d, seed,
+  grad_dense_pipe_body<XT, SAGA, PBLOCK, DEPTH, ITERS>(
+      (const XT*)sl.X, sl.y, sl.wbuf, sl.g, nullptr, sl.n_out, sl.alpha,
+      sl.idx_out, sl.e_out, sl.pos_ctr, nullptr, 0, sl.n_rows, d, seed,
       cmd.round_k[si], (uint64_t)sl.row_start, threshold, take_all,
       objective, sl.done_flag, cmd.done_val[si], sl.done_arr, bid,
       cmd.bper);
@@ -894,17 +912,17 @@ __device__ __forceinline__ float atomic_load_f32(float* p) {
 }
 
 __global__ __launch_bounds__(BLOCK) void saga_commit_wave_kernel(
-    const CsrWaveSlot* __restrict__ slots, CsrCommitCmd cmd) {
+    const CommitSlot* __restrict__ slots, CsrCommitCmd cmd) {
   const int si = (int)(blockIdx.x % cmd.n);
   const int bid = (int)(blockIdx.x / cmd.n);
-  const CsrWaveSlot sl = slots[cmd.wid[si]];
+  const CommitSlot sl = slots[cmd.wid[si]];
   const int cnt = atomic_load_int(sl.pos_ctr);
   if (cmd.do_commit[si]) {
     for (int i = bid * BLOCK + threadIdx.x; i < cnt;
          i += cmd.bper * BLOCK) {
-      const int r = atomic_load_int(&sl.idx_out[i]);
-      const float e = atomic_load_f32(&sl.e_out[i]);
-      sl.alpha[r] = e;  // only this worker's rounds touch its slice
+      const int r = atomic_load_int(&sl.idx[i]);
+      const float e = atomic_load_f32(&sl.e[i]);
+      sl.dst[r] = e;  // only this worker's rounds touch its slice
     }
   }
   // last block of the slot resets the staging + sample counters (reuses
@@ -914,16 +932,46 @@ __global__ __launch_bounds__(BLOCK) void saga_commit_wave_kernel(
   __syncthreads();
   if (threadIdx.x == 0) {
     const unsigned long long a = __hip_atomic_fetch_add(
-        sl.done_arr, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        sl.arr, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     if (a + 1 == (unsigned long long)cmd.bper) {
-      (void)__hip_atomic_exchange(sl.done_arr, 0ull, __ATOMIC_RELAXED,
+      (void)__hip_atomic_exchange(sl.arr, 0ull, __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
       (void)__hip_atomic_exchange(sl.pos_ctr, 0, __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
       (void)__hip_atomic_exchange(sl.n_out, 0, __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
+      if (sl.scnt)
+        (void)__hip_atomic_exchange(sl.scnt, 0, __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT);
       __builtin_amdgcn_s_waitcnt(0);
     }
+  }
+}
+
+// spill-refresh wave: per slot, recompute the round's Philox row set into
+// (srows, scnt), then gather ONLY those entries from the pinned master
+// into the device staging table (both stream-ordered before the wave's
+// gradient). Slot-mapped wrappers over the singleton kernels' logic.
+__global__ __launch_bounds__(BLOCK) void scan_rows_wave_kernel(
+    const GradWaveSlot* __restrict__ slots, GradWaveCmd cmd, uint64_t seed,
+    uint32_t threshold, int take_all) {
+  const int si = (int)(blockIdx.x % cmd.n);
+  const int bid = (int)(blockIdx.x / cmd.n);
+  const GradWaveSlot sl = slots[cmd.wid[si]];
+  scan_rows_body(sl.y, sl.srows, sl.sylist, sl.scnt, nullptr, sl.n_rows,
+                 seed, cmd.round_k[si], (uint64_t)sl.row_start, threshold,
+                 take_all, bid, cmd.bper);
+}
+
+__global__ void alpha_gather_wave_kernel(
+    const GradWaveSlot* __restrict__ slots, GradWaveCmd cmd) {
+  const int si = (int)(blockIdx.x % cmd.n);
+  const int bid = (int)(blockIdx.x / cmd.n);
+  const GradWaveSlot sl = slots[cmd.wid[si]];
+  const int n = *sl.scnt;  // same stream as the scan: plain read is fine
+  for (int i = bid * BLOCK + threadIdx.x; i < n; i += cmd.bper * BLOCK) {
+    const int r = sl.srows[i];
+    sl.alpha[r] = sl.alpha_host[r];
   }
 }
 
@@ -1351,7 +1399,7 @@ void launch_grad_dense_flag(
 
 void launch_grad_dense_wave(const void* slots_dev, const void* cmd_host,
                             long max_rows, int d, uint64_t seed, double rate,
-                            int objective, int x_is_bf16,
+                            int objective, int x_is_bf16, int saga,
                             hipStream_t stream) {
   const GradWaveCmd* cmd = (const GradWaveCmd*)cmd_host;
   const uint32_t thr = philox_threshold(rate);
@@ -1359,42 +1407,67 @@ void launch_grad_dense_wave(const void* slots_dev, const void* cmd_host,
   const int grid = cmd->n * cmd->bper;
   const int iters = (d + 255) / 256;
   const size_t smem = (size_t)(1 + 4) * (iters * 256) * sizeof(float) +
-                      (size_t)QCAP * sizeof(float) +
+                      (size_t)QCAP * sizeof(float) * (saga ? 3 : 1) +
                       (QCAP + 1) * sizeof(int);
   const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
   const int depth = dps ? std::atoi(dps) : 4;
   (void)max_rows;
-#define WAVE_LAUNCH_D(XT, IT, DP)                                            \
-  hipLaunchKernelGGL((grad_dense_wave_kernel<XT, 256, DP, IT>), dim3(grid),  \
-                     dim3(256), smem, stream,                                \
+#define WAVE_LAUNCH_D(XT, SG, IT, DP)                                        \
+  hipLaunchKernelGGL((grad_dense_wave_kernel<XT, SG, 256, DP, IT>),          \
+                     dim3(grid), dim3(256), smem, stream,                    \
                      (const GradWaveSlot*)slots_dev, *cmd, d, seed, thr,     \
                      take_all, objective)
-#define WAVE_LAUNCH(XT, IT)                                                  \
+#define WAVE_LAUNCH(XT, SG, IT)                                              \
   do {                                                                       \
-    if (depth == 1) WAVE_LAUNCH_D(XT, IT, 1);                                \
-    else if (depth == 2) WAVE_LAUNCH_D(XT, IT, 2);                           \
-    else if (depth == 6) WAVE_LAUNCH_D(XT, IT, 6);                           \
-    else if (depth == 8) WAVE_LAUNCH_D(XT, IT, 8);                           \
-    else WAVE_LAUNCH_D(XT, IT, 4);                                           \
+    if (depth == 1) WAVE_LAUNCH_D(XT, SG, IT, 1);                            \
+    else if (depth == 2) WAVE_LAUNCH_D(XT, SG, IT, 2);                       \
+    else if (depth == 6) WAVE_LAUNCH_D(XT, SG, IT, 6);                       \
+    else if (depth == 8) WAVE_LAUNCH_D(XT, SG, IT, 8);                       \
+    else WAVE_LAUNCH_D(XT, SG, IT, 4);                                       \
+  } while (0)
+#define WAVE_ITERS(XT, SG)                                                   \
+  do {                                                                       \
+    switch (iters) {                                                         \
+      case 1: WAVE_LAUNCH(XT, SG, 1); break;                                 \
+      case 2: WAVE_LAUNCH(XT, SG, 2); break;                                 \
+      case 3: WAVE_LAUNCH(XT, SG, 3); break;                                 \
+      case 4: WAVE_LAUNCH(XT, SG, 4); break;                                 \
+      case 5: WAVE_LAUNCH(XT, SG, 5); break;                                 \
+      case 6: WAVE_LAUNCH(XT, SG, 6); break;                                 \
+      case 7: WAVE_LAUNCH(XT, SG, 7); break;                                 \
+      default: WAVE_LAUNCH(XT, SG, 8); break;                                \
+    }                                                                        \
   } while (0)
 #define WAVE_DISPATCH(XT)                                                    \
   do {                                                                       \
-    switch (iters) {                                                         \
-      case 1: WAVE_LAUNCH(XT, 1); break;                                     \
-      case 2: WAVE_LAUNCH(XT, 2); break;                                     \
-      case 3: WAVE_LAUNCH(XT, 3); break;                                     \
-      case 4: WAVE_LAUNCH(XT, 4); break;                                     \
-      case 5: WAVE_LAUNCH(XT, 5); break;                                     \
-      case 6: WAVE_LAUNCH(XT, 6); break;                                     \
-      case 7: WAVE_LAUNCH(XT, 7); break;                                     \
-      default: WAVE_LAUNCH(XT, 8); break;                                    \
-    }                                                                        \
+    if (saga) WAVE_ITERS(XT, true);                                          \
+    else WAVE_ITERS(XT, false);                                              \
   } while (0)
   if (x_is_bf16) WAVE_DISPATCH(__hip_bfloat16);
   else WAVE_DISPATCH(float);
 #undef WAVE_DISPATCH
+#undef WAVE_ITERS
 #undef WAVE_LAUNCH
 #undef WAVE_LAUNCH_D
+}
+
+void launch_scan_rows_wave(const void* slots_dev, const void* cmd_host,
+                           uint64_t seed, double rate, hipStream_t stream) {
+  const GradWaveCmd* cmd = (const GradWaveCmd*)cmd_host;
+  const uint32_t thr = philox_threshold(rate);
+  const int take_all = rate >= 1.0;
+  hipLaunchKernelGGL(scan_rows_wave_kernel, dim3(cmd->n * cmd->bper),
+                     dim3(BLOCK), 0, stream,
+                     (const GradWaveSlot*)slots_dev, *cmd, seed, thr,
+                     take_all);
+}
+
+void launch_alpha_gather_wave(const void* slots_dev, const void* cmd_host,
+                              hipStream_t stream) {
+  const GradWaveCmd* cmd = (const GradWaveCmd*)cmd_host;
+  hipLaunchKernelGGL(alpha_gather_wave_kernel, dim3(cmd->n * cmd->bper),
+                     dim3(BLOCK), 0, stream,
+                     (const GradWaveSlot*)slots_dev, *cmd);
 }
 
 void launch_saga_grad_dense_flag(
@@ -1511,7 +1584,7 @@ void launch_saga_commit_wave(const void* slots_dev, const void* cmd_host,
   const CsrCommitCmd* cmd = (const CsrCommitCmd*)cmd_host;
   const int grid = cmd->n * cmd->bper;
   hipLaunchKernelGGL(saga_commit_wave_kernel, dim3(grid), dim3(BLOCK), 0,
-                     stream, (const CsrWaveSlot*)slots_dev, *cmd);
+                     stream, (const CommitSlot*)slots_dev, *cmd);
 }
 
 void launch_grad_csr_flag(
