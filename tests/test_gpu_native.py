@@ -195,3 +195,32 @@ def test_native_csr_saga_wave_matches_nowave_p1():
     assert arel < 2e-3, arel
     rel = float((w_wave - w_ref).norm() / (w_ref.norm() + 1e-12))
     assert rel < 1e-3, rel
+
+
+def test_native_dense_wave_matches_nowave_p1():
+    """Dense ASGD at P=1 is deterministic and sequential: the wave path
+    (grad_dense_wave_kernel) must reproduce the singleton path bit-for-bit
+    up to atomic accumulation order."""
+    import os
+
+    def run(no_wave):
+        if no_wave:
+            os.environ["ASYNCAMD_NO_WAVE"] = "1"
+        else:
+            os.environ.pop("ASYNCAMD_NO_WAVE", None)
+        cfg = _cfg(num_iterations=150, N=30_000, d=256, batch_rate=0.02,
+                   gamma=0.2)
+        X, y = synthetic_dense(cfg.N, cfg.d, seed=13, device="cuda:0")
+        eng = NativeLocalEngine(cfg, _shards(cfg, X, y),
+                                torch.device("cuda:0"))
+        res = eng.run(max_wall_s=120)
+        assert res["k"] == cfg.num_iterations
+        return eng.w.clone()
+
+    try:
+        w_wave = run(no_wave=False)
+        w_ref = run(no_wave=True)
+    finally:
+        os.environ.pop("ASYNCAMD_NO_WAVE", None)
+    rel = float((w_wave - w_ref).norm() / (w_ref.norm() + 1e-12))
+    assert rel < 1e-4, rel
