@@ -742,6 +742,14 @@ struct NativeEngine {
         HIP_CHECK(hipStreamCreateWithFlags(&wstreams[i],
                                            hipStreamNonBlocking));
       wave_bper = query_grad_grid(wave_max_rows);
+      if (!std::getenv("ASYNCAMD_GRAD_GRID")) {
+        // measured (flagship, P=32): with interleaved full-P waves the
+        // chip is best fed by ~512-768 TOTAL blocks, each carrying many
+        // sampled rows (deep pipelines) — per-worker grids sized as if
+        // launched alone left 5 rows/block and 114k updates/s vs 186k
+        // at 16-24 blocks/worker (same-box sweep, ASYNCAMD_GRAD_GRID)
+        wave_bper = std::min(wave_bper, std::max(8, 640 / cfg.P));
+      }
       // interleaved block->slot mapping measured +21% on the flagship
       // (workers progress together => completions bunch => bigger update
       // batches and fuller next waves)
