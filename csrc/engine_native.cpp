@@ -176,7 +176,6 @@ struct WorkerBuf {
   bool pending_commit = false;
   int saga_cap = 0;  // staging capacity (commit grid bound, SAGA only)
   double submit_t = 0, finish_t = 0, waiting_ms = 0;
-  double poll_after = 0;  // don't hipEventQuery before this time
   bool g_dirty = false;   // gradient buffer holds a rejected round's sums
   long tasks = 0;
 };
@@ -226,7 +225,6 @@ struct NativeEngine {
   // straggler sets (reference :124-141)
   std::vector<int> straggler_kind;  // 0 none, 1 normal, 2 longtail
   double mark_lo_t = 0, mark_hi_t = 0;
-  double ewma_round_s = 0;  // poll gating
   std::vector<double> snap_ms;  // host stamps for the optVars snapshots
   double run_t0 = 0;
   // batched-update machinery
@@ -531,7 +529,6 @@ struct NativeEngine {
     wk.finish_t = t_now;
     wk.tasks += 1;
     const double rt = t_now - wk.submit_t;
-    ewma_round_s = ewma_round_s == 0 ? rt : 0.9 * ewma_round_s + 0.1 * rt;
     const int staleness = clock - wk.ts;  // arrival-clock staleness
     clock += 1;
     max_staleness_seen = std::max<long>(max_staleness_seen, staleness);

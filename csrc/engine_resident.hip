@@ -1024,7 +1024,7 @@ void register_resident_engine(py::module_& m) {
       if (q == hipSuccess) break;
       if (q != hipErrorNotReady) RHIP_CHECK(q);
       if (wall_s() > host_deadline) {
-        hipStreamDestroy(stream);
+        (void)hipStreamDestroy(stream);
         throw std::runtime_error(
             "resident engine: kernel exceeded host deadline (device "
             "watchdog failed to fire)");
