@@ -745,7 +745,7 @@ struct NativeEngine {
         // sampled rows (deep pipelines) — per-worker grids sized as if
         // launched alone left 5 rows/block and 114k updates/s vs 186k
         // at 16-24 blocks/worker (same-box sweep, ASYNCAMD_GRAD_GRID)
-        wave_bper = std::min(wave_bper, std::max(8, 640 / cfg.P));
+        wave_bper = std::min(wave_bper, std::max(8, 512 / cfg.P));
       }
       // interleaved block->slot mapping measured +21% on the flagship
       // (workers progress together => completions bunch => bigger update

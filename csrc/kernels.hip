@@ -1410,8 +1410,8 @@ void launch_grad_dense_wave(const void* slots_dev, const void* cmd_host,
                       (size_t)QCAP * sizeof(float) * (saga ? 3 : 1) +
                       (QCAP + 1) * sizeof(int);
   const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
-  const int depth = dps ? std::atoi(dps) : 4;
-  (void)max_rows;
+  const int depth = dps ? std::atoi(dps) : 6;  // wave-measured optimum
+  (void)max_rows;                              // (solo kernels keep 4)
 #define WAVE_LAUNCH_D(XT, SG, IT, DP)                                        \
   hipLaunchKernelGGL((grad_dense_wave_kernel<XT, SG, 256, DP, IT>),          \
                      dim3(grid), dim3(256), smem, stream,                    \
