@@ -1410,8 +1410,11 @@ void launch_grad_dense_wave(const void* slots_dev, const void* cmd_host,
                       (size_t)QCAP * sizeof(float) * (saga ? 3 : 1) +
                       (QCAP + 1) * sizeof(int);
   const char* dps = std::getenv("ASYNCAMD_PIPE_DEPTH");
-  const int depth = dps ? std::atoi(dps) : 6;  // wave-measured optimum
-  (void)max_rows;                              // (solo kernels keep 4)
+  // wave-measured: depth 6 wins at small ITERS (d<=1024, flagship +8%);
+  // keep 4 at larger feature dims where the deeper pipeline's register
+  // pressure bites (epsilon d=2000). Solo kernels keep 4 everywhere.
+  const int depth = dps ? std::atoi(dps) : (iters <= 4 ? 6 : 4);
+  (void)max_rows;
 #define WAVE_LAUNCH_D(XT, SG, IT, DP)                                        \
   hipLaunchKernelGGL((grad_dense_wave_kernel<XT, SG, 256, DP, IT>),          \
                      dim3(grid), dim3(256), smem, stream,                    \
