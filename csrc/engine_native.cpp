@@ -5,24 +5,34 @@
 // Executor.scala TaskRunner) for co-located workers: ONE host thread drives
 // an event loop over per-worker HIP streams —
 //
-//   dispatch(i): stream_i waits the last update event, copies w into the
-//                worker's snapshot buffer (the versioned-broadcast semantic),
-//                zeroes g_i, launches the fused gradient kernel with the
-//                round's Philox key, records done_i;
-//   poll:        hipEventQuery(done_i) == success -> completion bookkeeping
-//                (arrival clock, staleness, availability) exactly as the
-//                Python engine does;
-//   accept:      tau filter; SGD/SAGA update on the server stream (+ SAGA
-//                commit on the worker stream, ordered before its next
-//                round); requeue; quorum-gated redispatch;
+//   dispatch:    each quorum WAVE launches as ONE kernel
+//                (grad_dense_wave_kernel / grad_csr_wave_kernel over a
+//                device slot table, SAGA commits batched into one
+//                commit+staging-reset kernel, spill refresh as
+//                scan/gather wave kernels) on a rotating stream pool; the
+//                batched update kernel writes the redispatching workers'
+//                weight snapshots (versioned-broadcast semantic), so the
+//                whole wave costs a handful of launches instead of 2-5
+//                per worker. First dispatches and straggler releases use
+//                the per-worker singleton path;
+//   poll:        EVENT-FREE — the grad kernel's last block publishes the
+//                round serial to a pinned-host line (publish_done,
+//                kernels.hip) and the host polls plain memory;
+//   accept:      tau filter; accepted gradients accumulate into ONE
+//                elementwise-sequential multi-update kernel per sweep;
+//                requeue; quorum-gated redispatch;
 //   delay:       the reference's straggler model (cloud long-tail / coeff),
 //                implemented as host-side due-times on the dispatch queue —
 //                no thread sleeps, no locks, no GIL (released for the whole
 //                run).
 //
-// This is the same control plane as engine/local.py with the thread handoffs
-// (~100 us each under the GIL) replaced by ~1 us event polls. The RCCL
-// multi-GPU server reuses this loop shape with peers instead of streams.
+// This is the same control plane as engine/local.py with the thread
+// handoffs (~100 us each under the GIL) replaced by sub-us flag reads.
+// Measured evolution on the flagship config (profiles/r02_wave_dispatch.md):
+// 39.9k updates/s (r01, per-round launches + events) -> 87.2k (batched
+// updates) -> 103k (event-free) -> 188k (wave dispatch, interleaved
+// mapping, wave-aware grid sizing). The RCCL multi-GPU server reuses this
+// loop shape with peers instead of streams (csrc/server_dist.cpp).
 
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
