@@ -2,12 +2,14 @@
 (csrc/engine_native.cpp): multi-worker bounded-staleness async on one GPU
 with zero Python in the round loop.
 
-Each worker owns a shard, a HIP stream (created natively), a weight-snapshot
-buffer (the versioned-broadcast semantic) and a gradient accumulator; the
-C++ loop does dispatch -> event poll -> tau filter -> fused update ->
+Each worker owns a shard, a weight-snapshot buffer (the
+versioned-broadcast semantic), a gradient accumulator and a pinned-host
+completion flag; the C++ loop does wave dispatch (one kernel per quorum
+wave) -> pinned-flag poll -> tau filter -> batched fused update ->
 quorum-gated redispatch, including the reference's straggler model and
-calibration. Use for GPU multi-worker configs (the threaded Python engine
-stays for CPU tests and as the semantics oracle)."""
+calibration (csrc/engine_native.cpp header for the measured evolution).
+Use for GPU multi-worker configs (the threaded Python engine stays for
+CPU tests and as the semantics oracle)."""
 
 from __future__ import annotations
 
