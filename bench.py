@@ -94,6 +94,12 @@ def parse_args():
                    help="N>1 control plane: python (default) or the C++ "
                         "server (also via ASYNCAMD_DIST_ENGINE=native)")
     args = p.parse_args()
+    # marks record at exact post-increment update counts, so k=0 is
+    # unreachable: clamp to >=1 warmup step (negligible) rather than fail
+    # if a caller passes --warmup 0; likewise a 1-step timed region is
+    # meaningless — keep at least 2
+    args.warmup = max(1, args.warmup)
+    args.steps = max(2, args.steps)
     preset = MODELS[args.model]
     args.rows = args.rows or preset["rows"]
     args.cols = args.cols or preset["cols"]
