@@ -45,12 +45,12 @@ MODELS = {
                           objective="logistic"),  # BASELINE config 1 names
                                                   # logistic regression
     # flagship: 32 async workers (the reference's fixed partitions=32,
-    # README.md) on the native C++ event loop with batched fused updates
-    # (87.2k updates/s measured; whole-chip 16.6us gradient rounds keep the
-    # quorum gate cheap). --engine resident runs the same config inside ONE
-    # persistent HIP kernel (77.2k: zero host API calls, but the static
-    # G-blocks-per-worker partition makes rounds ~190us and the reference's
-    # avail>=gate quorum turns that latency into idle — analysis in
+    # README.md) on the native C++ engine: wave dispatch + event-free
+    # pinned-host completion + batched fused updates — 188-191k updates/s
+    # fresh-box validated (profiles/r02_wave_dispatch.md). --engine
+    # resident runs the same config inside ONE persistent HIP kernel
+    # (77.2k: zero host API calls, but the reference's avail>=gate quorum
+    # turns its static-partition round latency into idle —
     # profiles/r02_resident_profile.md); --engine graph = 1-worker hipGraph
     "asgd-mnist8m": dict(rows=8_100_000, cols=784, rate=0.01, algo="asgd",
                          sync=False, dtype="bf16", sparse=False,
@@ -73,8 +73,8 @@ def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="asgd-mnist8m", choices=list(MODELS))
     p.add_argument("--gpus", type=int, default=1)
-    # defaults sized so the timed region is >=0.25 s at the flagship's
-    # measured ~113k updates/s (a 2000-step region was ~20 ms — too noisy)
+    # defaults sized so the timed region is >=0.15 s at the flagship's
+    # measured ~190k updates/s (a 2000-step region was ~10 ms — too noisy)
     p.add_argument("--steps", type=int, default=30000)
     p.add_argument("--warmup", type=int, default=3000)
     p.add_argument("--rows", type=int, default=0)
