@@ -861,8 +861,13 @@ struct NativeEngine {
     }
     HIP_CHECK(hipStreamSynchronize(sstream));
     const double t1 = now_s();
-    // drain in-flight rounds so buffers are quiescent before Python resumes
+    // drain in-flight rounds so buffers are quiescent before Python
+    // resumes — INCLUDING the wave streams: the final sweep may have
+    // dispatched a wave whose kernels are still writing g/wbuf/alpha and
+    // publishing to the pinned flags this teardown is about to free
     for (auto& wk : ws) HIP_CHECK(hipStreamSynchronize(wk.stream));
+    for (int i = 0; i < NWSTREAM; ++i)
+      if (wstreams[i]) HIP_CHECK(hipStreamSynchronize(wstreams[i]));
     Result out;
     out.k = k;
     out.elapsed_ms = (t1 - t0) * 1000.0;
