@@ -62,3 +62,14 @@ def test_make_cfg_iteration_budget():
     cfg = bench.make_cfg(a, 2, "cpu")
     assert cfg.num_iterations == 111
     assert cfg.snapshot_weights is False
+
+
+def test_warmup_and_steps_clamped():
+    """marks record at exact post-increment update counts, so k=0 is
+    unreachable: bench must clamp --warmup 0 (and degenerate --steps)
+    instead of failing with 'marks missing'."""
+    a = _parse(["--warmup", "0", "--steps", "1"])
+    assert a.warmup == 1
+    assert a.steps == 2
+    a = _parse(["--warmup", "3000", "--steps", "30000"])
+    assert a.warmup == 3000 and a.steps == 30000
