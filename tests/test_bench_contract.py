@@ -64,3 +64,33 @@ def test_bench_torchrun_dist_cpu_contract(dist_engine):
     assert j["n_gpus"] == 2
     assert j["value"] > 0
     assert j["steps"] == 150
+
+
+def test_to_target_summary_dense_cpu():
+    """The BASELINE metric's second clause (wall-clock-to-target-loss) is
+    computed from the optVars ring: verify target detection, the
+    always-defined 95%-of-achieved-progress clause, and the None guard."""
+    import torch
+
+    import bench
+
+    class A:
+        sparse = False
+        objective = "lsq"
+
+    torch.manual_seed(0)
+    X = torch.randn(500, 16)
+    w_true = torch.randn(16)
+    y = X @ w_true
+    # opt_vars: w converging toward w_true -> objective decreasing
+    opt_vars = [(0, torch.zeros(16))] + [
+        (100 * (i + 1), w_true * (i + 1) / 5.0) for i in range(5)]
+    out = bench.to_target_summary(A(), (X, y), opt_vars, target_frac=0.5)
+    assert out is not None
+    assert out["obj_initial"] > out["obj_final"]
+    assert out["ms_to_target"] is not None  # exact solve crosses 0.5*obj0
+    assert out["ms_to_95pct_of_achieved_progress"] is not None
+    assert out["ms_to_95pct_of_achieved_progress"] <= 500
+    # guard: too-short rings yield None, not a crash
+    assert bench.to_target_summary(A(), (X, y), [(0, torch.zeros(16))]) is None
+    assert bench.to_target_summary(A(), (X, y), None) is None
